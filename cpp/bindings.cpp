@@ -1,0 +1,346 @@
+// grapehip — Python bindings (pybind11).
+//
+// Engine = one rank of the distributed runtime (TCP control plane; RCCL data
+// plane on GPU). Graph = one edge-cut fragment (host CSR + optional device
+// mirror). App entry points return (oid array, value array) for this rank's
+// inner vertices plus timing metadata.
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <chrono>
+#include <memory>
+#include <optional>
+
+#include "apps/bfs.hpp"
+#include "apps/cdlp.hpp"
+#include "apps/lcc.hpp"
+#include "apps/pagerank.hpp"
+#include "apps/sssp.hpp"
+#include "apps/wcc.hpp"
+#include "core/fragment.hpp"
+#include "core/message_manager.hpp"
+#include "core/net.hpp"
+
+#ifdef GRAPEHIP_WITH_HIP
+#include "hip/gpu_engine.hpp"
+#endif
+
+namespace py = pybind11;
+using namespace grapehip;
+
+namespace {
+
+double now_s() {
+  return std::chrono::duration<double>(
+             std::chrono::steady_clock::now().time_since_epoch())
+      .count();
+}
+
+struct PyGraph {
+  std::shared_ptr<VertexMap> vm;
+  std::unique_ptr<Fragment> frag;
+#ifdef GRAPEHIP_WITH_HIP
+  std::unique_ptr<DeviceGraph> dev;
+#endif
+};
+
+struct PyEngine {
+  TcpComm comm;
+  int rank = 0, world = 1, n_threads = 0;
+  bool use_gpu = false;
+#ifdef GRAPEHIP_WITH_HIP
+  std::unique_ptr<GpuContext> gpu;
+#endif
+
+  PyEngine(int rank_, int world_, const std::string& addr, int port,
+           int threads, bool gpu_) {
+    rank = rank_;
+    world = world_;
+    use_gpu = gpu_;
+    if (threads > 0) ThreadPool::Get().set_num_threads(threads);
+    n_threads = ThreadPool::Get().num_threads();
+    comm.init(rank, world, addr, port);
+#ifdef GRAPEHIP_WITH_HIP
+    if (use_gpu) gpu = std::make_unique<GpuContext>(&comm, rank, world);
+#else
+    if (use_gpu)
+      throw std::runtime_error(
+          "grapehip was built without HIP support but gpu=True was "
+          "requested — rebuild with GRAPEHIP_WITH_HIP");
+#endif
+  }
+
+  TcpComm* c() { return world > 1 ? &comm : nullptr; }
+};
+
+// Build fragment from this rank's slice of the edge list.
+using arr_i64 = py::array_t<int64_t, py::array::c_style | py::array::forcecast>;
+using arr_f32 = py::array_t<float, py::array::c_style | py::array::forcecast>;
+
+std::shared_ptr<PyGraph> load_edges(
+    PyEngine& eng, arr_i64 src, arr_i64 dst,
+    std::optional<arr_f32> weights, bool directed,
+    int64_t num_vertices, std::optional<arr_i64> vertex_oids,
+    bool build_in_csr, const std::string& partitioner) {
+  size_t n = src.size();
+  if (static_cast<size_t>(dst.size()) != n)
+    throw std::runtime_error("src/dst size mismatch");
+  bool weighted = weights.has_value();
+  std::vector<EdgeTriple> edges(n);
+  {
+    auto s = src.unchecked<1>();
+    auto d = dst.unchecked<1>();
+    const float* w = weighted ? weights->data() : nullptr;
+    for (size_t i = 0; i < n; ++i)
+      edges[i] = {s(i), d(i), w ? w[i] : 1.0f};
+  }
+
+  auto g = std::make_shared<PyGraph>();
+  g->vm = std::make_shared<VertexMap>();
+  py::gil_scoped_release rel;
+
+  if (vertex_oids.has_value()) {
+    // hashmap idxer: route each oid to its owner, owners dedup + index
+    PartitionerKind pk = partitioner == "hash" ? PartitionerKind::kHash
+                                               : PartitionerKind::kHash;
+    // (segmented needs dense oids; with explicit oids use hash)
+    const int64_t* po = vertex_oids->data();
+    size_t nv = vertex_oids->size();
+    std::vector<std::vector<oid_t>> bins(eng.world);
+    for (size_t i = 0; i < nv; ++i) {
+      oid_t o = po[i];
+      bins[hash_oid(o) % eng.world].push_back(o);
+    }
+    std::vector<std::string> send(eng.world);
+    for (int f = 0; f < eng.world; ++f)
+      send[f].assign(reinterpret_cast<const char*>(bins[f].data()),
+                     bins[f].size() * sizeof(oid_t));
+    std::vector<std::string> recv =
+        eng.c() ? eng.c()->exchange_all(send) : std::move(send);
+    std::vector<oid_t> owned;
+    for (auto& blob : recv) {
+      size_t m = blob.size() / sizeof(oid_t);
+      const oid_t* p = reinterpret_cast<const oid_t*>(blob.data());
+      owned.insert(owned.end(), p, p + m);
+    }
+    std::sort(owned.begin(), owned.end());
+    owned.erase(std::unique(owned.begin(), owned.end()), owned.end());
+    g->vm->init_hashmap(eng.world, pk, eng.c(), std::move(owned));
+  } else {
+    if (num_vertices <= 0)
+      throw std::runtime_error("num_vertices required for identity mapping");
+    g->vm->init_identity(eng.world, static_cast<uint64_t>(num_vertices));
+  }
+
+  g->frag = Fragment::Build(g->vm, eng.c(), eng.rank, eng.world,
+                            std::move(edges), directed, weighted,
+                            build_in_csr, n);
+#ifdef GRAPEHIP_WITH_HIP
+  if (eng.use_gpu) g->dev = eng.gpu->upload(*g->frag);
+#endif
+  return g;
+}
+
+template <typename T>
+py::array_t<T> to_np(const std::vector<T>& v) {
+  py::array_t<T> a(v.size());
+  std::memcpy(a.mutable_data(), v.data(), v.size() * sizeof(T));
+  return a;
+}
+
+py::array_t<int64_t> inner_oids(const Fragment& f) {
+  py::array_t<int64_t> a(f.ivnum());
+  auto* p = a.mutable_data();
+  for (vid_t v = 0; v < f.ivnum(); ++v) p[v] = f.lid2oid(v);
+  return a;
+}
+
+template <typename RunFn>
+py::dict run_timed(PyEngine& eng, const Fragment& frag, RunFn&& run) {
+  py::dict out;
+  double t0, t1;
+  int rounds;
+  {
+    py::gil_scoped_release rel;
+    if (eng.c()) eng.c()->barrier();
+    t0 = now_s();
+    rounds = run();
+    if (eng.c()) eng.c()->barrier();
+    t1 = now_s();
+  }
+  out["rounds"] = rounds;
+  out["seconds"] = t1 - t0;
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "grapehip core engine";
+
+  py::class_<PyGraph, std::shared_ptr<PyGraph>>(m, "Graph")
+      .def_property_readonly(
+          "num_vertices",
+          [](const PyGraph& g) { return g.frag->total_vertices(); })
+      .def_property_readonly(
+          "num_edges", [](const PyGraph& g) { return g.frag->total_edges(); })
+      .def_property_readonly(
+          "input_edges",
+          [](const PyGraph& g) { return g.frag->input_edges(); })
+      .def_property_readonly(
+          "ivnum", [](const PyGraph& g) { return g.frag->ivnum(); })
+      .def_property_readonly(
+          "ovnum", [](const PyGraph& g) { return g.frag->ovnum(); })
+      .def_property_readonly("directed",
+                             [](const PyGraph& g) { return g.frag->directed(); })
+      .def("inner_oids",
+           [](const PyGraph& g) { return inner_oids(*g.frag); });
+
+  py::class_<PyEngine>(m, "Engine")
+      .def(py::init<int, int, const std::string&, int, int, bool>(),
+           py::arg("rank") = 0, py::arg("world") = 1,
+           py::arg("master_addr") = "127.0.0.1",
+           py::arg("master_port") = 29517, py::arg("n_threads") = 0,
+           py::arg("gpu") = false)
+      .def_readonly("rank", &PyEngine::rank)
+      .def_readonly("world", &PyEngine::world)
+      .def_property_readonly("gpu", [](const PyEngine& e) { return e.use_gpu; })
+      .def("barrier",
+           [](PyEngine& e) {
+             py::gil_scoped_release rel;
+             if (e.c()) e.c()->barrier();
+           })
+      .def("load_edges", &load_edges, py::arg("src"), py::arg("dst"),
+           py::arg("weights") = std::nullopt, py::arg("directed") = false,
+           py::arg("num_vertices") = -1,
+           py::arg("vertex_oids") = std::nullopt,
+           py::arg("build_in_csr") = false,
+           py::arg("partitioner") = "segmented")
+      .def("bfs",
+           [](PyEngine& eng, PyGraph& g, int64_t source) {
+#ifdef GRAPEHIP_WITH_HIP
+             if (eng.use_gpu) return eng.gpu->bfs(*g.dev, source);
+#endif
+             BFSApp app;
+             BFSContext ctx;
+             MessageManager mm;
+             mm.init(eng.c(), g.frag.get(), eng.n_threads);
+             ctx.init(*g.frag, source);
+             py::dict meta = run_timed(eng, *g.frag, [&] {
+               return RunWorker(app, ctx, *g.frag, mm);
+             });
+             std::vector<int64_t> vals(g.frag->ivnum());
+             for (vid_t v = 0; v < g.frag->ivnum(); ++v)
+               vals[v] = ctx.depth[v].load(std::memory_order_relaxed);
+             meta["oids"] = inner_oids(*g.frag);
+             meta["values"] = to_np(vals);
+             return meta;
+           },
+           py::arg("graph"), py::arg("source") = 0)
+      .def("sssp",
+           [](PyEngine& eng, PyGraph& g, int64_t source) {
+#ifdef GRAPEHIP_WITH_HIP
+             if (eng.use_gpu) return eng.gpu->sssp(*g.dev, source);
+#endif
+             SSSPApp app;
+             SSSPContext ctx;
+             MessageManager mm;
+             mm.init(eng.c(), g.frag.get(), eng.n_threads);
+             ctx.init(*g.frag, source);
+             py::dict meta = run_timed(eng, *g.frag, [&] {
+               return RunWorker(app, ctx, *g.frag, mm);
+             });
+             std::vector<double> vals(g.frag->ivnum());
+             for (vid_t v = 0; v < g.frag->ivnum(); ++v)
+               vals[v] = ctx.dist[v].load(std::memory_order_relaxed);
+             meta["oids"] = inner_oids(*g.frag);
+             meta["values"] = to_np(vals);
+             return meta;
+           },
+           py::arg("graph"), py::arg("source") = 0)
+      .def("pagerank",
+           [](PyEngine& eng, PyGraph& g, double damping, int iters) {
+#ifdef GRAPEHIP_WITH_HIP
+             if (eng.use_gpu) return eng.gpu->pagerank(*g.dev, damping, iters);
+#endif
+             PageRankApp app;
+             PageRankContext ctx;
+             MessageManager mm;
+             mm.init(eng.c(), g.frag.get(), eng.n_threads);
+             ctx.init(*g.frag, damping, iters);
+             py::dict meta = run_timed(eng, *g.frag, [&] {
+               return RunWorker(app, ctx, *g.frag, mm);
+             });
+             meta["oids"] = inner_oids(*g.frag);
+             meta["values"] = to_np(ctx.rank);
+             return meta;
+           },
+           py::arg("graph"), py::arg("damping") = 0.85, py::arg("iters") = 10)
+      .def("wcc",
+           [](PyEngine& eng, PyGraph& g) {
+#ifdef GRAPEHIP_WITH_HIP
+             if (eng.use_gpu) return eng.gpu->wcc(*g.dev);
+#endif
+             WCCApp app;
+             WCCContext ctx;
+             MessageManager mm;
+             mm.init(eng.c(), g.frag.get(), eng.n_threads);
+             ctx.init(*g.frag);
+             py::dict meta = run_timed(eng, *g.frag, [&] {
+               return RunWorker(app, ctx, *g.frag, mm);
+             });
+             std::vector<int64_t> vals(g.frag->ivnum());
+             for (vid_t v = 0; v < g.frag->ivnum(); ++v)
+               vals[v] = ctx.label[v].load(std::memory_order_relaxed);
+             meta["oids"] = inner_oids(*g.frag);
+             meta["values"] = to_np(vals);
+             return meta;
+           },
+           py::arg("graph"))
+      .def("cdlp",
+           [](PyEngine& eng, PyGraph& g, int iters) {
+#ifdef GRAPEHIP_WITH_HIP
+             if (eng.use_gpu) return eng.gpu->cdlp(*g.dev, iters);
+#endif
+             CDLPApp app;
+             CDLPContext ctx;
+             MessageManager mm;
+             mm.init(eng.c(), g.frag.get(), eng.n_threads);
+             ctx.init(*g.frag, iters);
+             py::dict meta = run_timed(eng, *g.frag, [&] {
+               return RunWorker(app, ctx, *g.frag, mm);
+             });
+             meta["oids"] = inner_oids(*g.frag);
+             std::vector<int64_t> vals(ctx.label.begin(),
+                                       ctx.label.begin() + g.frag->ivnum());
+             meta["values"] = to_np(vals);
+             return meta;
+           },
+           py::arg("graph"), py::arg("iters") = 10)
+      .def("lcc",
+           [](PyEngine& eng, PyGraph& g) {
+#ifdef GRAPEHIP_WITH_HIP
+             if (eng.use_gpu) return eng.gpu->lcc(*g.dev);
+#endif
+             LCCApp app;
+             LCCContext ctx;
+             MessageManager mm;
+             mm.init(eng.c(), g.frag.get(), eng.n_threads);
+             ctx.init(*g.frag);
+             py::dict meta = run_timed(eng, *g.frag, [&] {
+               return RunWorker(app, ctx, *g.frag, mm);
+             });
+             meta["oids"] = inner_oids(*g.frag);
+             meta["values"] = to_np(ctx.lcc);
+             return meta;
+           },
+           py::arg("graph"));
+
+  m.attr("WITH_HIP") =
+#ifdef GRAPEHIP_WITH_HIP
+      true;
+#else
+      false;
+#endif
+}

@@ -1,0 +1,157 @@
+// grapehip — CPU message manager + BSP superstep worker.
+//
+// Reference parity: grape/parallel/{message_manager_base,
+// default_message_manager,parallel_message_manager}.h and
+// grape/worker/worker.h. Semantics kept exactly:
+//   * per-round per-destination archives of (gid, msg) pairs;
+//   * SyncStateOnOuterVertex / SendMsgThroughOEdges helpers;
+//   * termination when a round moves zero bytes globally and no rank called
+//     ForceContinue (worker.h:105-146, default_message_manager.h:324-350);
+//   * per-thread channel buffers to avoid append contention
+//     (thread_local_message_buffer.h).
+// Transport is the TCP control plane (exchange_all) instead of MPI isend
+// rings — the CPU path is the correctness oracle; the RCCL path lives in
+// cpp/hip/.
+#pragma once
+
+#include <cstring>
+#include <string>
+#include <vector>
+
+#include "archive.hpp"
+#include "fragment.hpp"
+#include "net.hpp"
+#include "parallel.hpp"
+#include "types.hpp"
+
+namespace grapehip {
+
+class MessageManager {
+ public:
+  void init(TcpComm* comm, const Fragment* frag, int n_threads) {
+    comm_ = comm;
+    frag_ = frag;
+    fnum_ = frag->fnum();
+    n_threads_ = n_threads;
+    channels_.assign(n_threads, std::vector<InArchive>(fnum_));
+    recv_.assign(fnum_, OutArchive());
+    force_continue_ = false;
+    terminated_ = false;
+    round_ = 0;
+  }
+
+  void start_round() { force_continue_ = false; }
+
+  // ---- send API (thread-safe via tid channel) --------------------------
+  template <typename MSG>
+  void sync_state_on_outer_vertex(int tid, vid_t outer_lid, const MSG& msg) {
+    vid_t gid = frag_->lid2gid(outer_lid);
+    fid_t dst = frag_->parser().fid(gid);
+    auto& ar = channels_[tid][dst];
+    ar.add(gid);
+    ar.add(msg);
+  }
+  template <typename MSG>
+  void send_to_fragment(int tid, fid_t dst, vid_t gid, const MSG& msg) {
+    auto& ar = channels_[tid][dst];
+    ar.add(gid);
+    ar.add(msg);
+  }
+
+  void force_continue() { force_continue_ = true; }
+
+  // ---- round boundary --------------------------------------------------
+  void finish_round() {
+    // concatenate per-thread channels per destination
+    std::vector<std::string> send(fnum_);
+    for (int f = 0; f < fnum_; ++f) {
+      size_t total = 0;
+      for (int t = 0; t < n_threads_; ++t) total += channels_[t][f].size();
+      send[f].reserve(total);
+      for (int t = 0; t < n_threads_; ++t) {
+        send[f].append(channels_[t][f].data(), channels_[t][f].size());
+        channels_[t][f].clear();
+      }
+    }
+    uint64_t moved = 0;
+    for (auto& s : send) moved += s.size();
+    std::vector<std::string> recv =
+        (comm_ && fnum_ > 1) ? comm_->exchange_all(send) : std::move(send);
+    for (int f = 0; f < fnum_; ++f) recv_[f].reset(std::move(recv[f]));
+
+    uint64_t global_moved =
+        comm_ ? comm_->allreduce_sum(moved + (force_continue_ ? 1 : 0))
+              : moved + (force_continue_ ? 1 : 0);
+    terminated_ = (global_moved == 0);
+    ++round_;
+  }
+
+  bool to_terminate() const { return terminated_; }
+  int round() const { return round_; }
+  int n_threads() const { return n_threads_; }
+
+  // App-level collectives (reference grape/communication/communicator.h) and
+  // dense batch-shuffle exchanges (batch_shuffle_message_manager.h) go
+  // straight over the control plane.
+  TcpComm* comm() { return comm_; }
+  double sum_double(double v) {
+    if (!comm_ || fnum_ == 1) return v;
+    std::vector<double> all(fnum_);
+    comm_->allgather(&v, sizeof(double), all.data());
+    double s = 0;
+    for (double x : all) s += x;
+    return s;
+  }
+
+  // ---- receive API -----------------------------------------------------
+  // Decode all (gid, MSG) pairs from every source fragment; f(lid, msg).
+  // Parallel over source archives and chunks within each archive.
+  template <typename MSG, typename F>
+  void process(F&& f) {
+    const IdParser& P = frag_->parser();
+    constexpr size_t unit = sizeof(vid_t) + sizeof(MSG);
+    for (int src = 0; src < fnum_; ++src) {
+      OutArchive& ar = recv_[src];
+      size_t n = ar.remaining() / unit;
+      const char* base = ar.cursor();
+      parallel_for_tid(0, n, [&](int tid, size_t i) {
+        vid_t gid;
+        MSG m;
+        std::memcpy(&gid, base + i * unit, sizeof(vid_t));
+        std::memcpy(&m, base + i * unit + sizeof(vid_t), sizeof(MSG));
+        f(tid, P.lid(gid), m);
+      }, 2048);
+      ar.skip(n * unit);
+    }
+  }
+
+ private:
+  TcpComm* comm_ = nullptr;
+  const Fragment* frag_ = nullptr;
+  int fnum_ = 1;
+  int n_threads_ = 1;
+  bool force_continue_ = false;
+  bool terminated_ = false;
+  int round_ = 0;
+  std::vector<std::vector<InArchive>> channels_;  // [tid][fid]
+  std::vector<OutArchive> recv_;                  // [src fid]
+};
+
+// BSP superstep driver (reference worker.h:105-146).
+template <typename APP, typename CTX>
+inline int RunWorker(APP& app, CTX& ctx, const Fragment& frag,
+                     MessageManager& mm) {
+  mm.start_round();
+  app.PEval(frag, ctx, mm);
+  mm.finish_round();
+  int rounds = 1;
+  while (!mm.to_terminate()) {
+    mm.start_round();
+    app.IncEval(frag, ctx, mm);
+    mm.finish_round();
+    ++rounds;
+  }
+  return rounds;
+}
+
+}  // namespace grapehip

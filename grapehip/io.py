@@ -1,0 +1,59 @@
+"""LDBC Graphalytics TSV input (.v / .e files).
+
+Reference parity: grape/io/local_io_adaptor.cc partial-read loader — each
+rank reads a byte-range of the file aligned to line boundaries and parses
+its slice; the engine's distributed builder shuffles records to owners.
+"""
+from __future__ import annotations
+
+import os
+
+import numpy as np
+
+
+def _byte_slice(path: str, rank: int, world: int):
+    size = os.path.getsize(path)
+    lo = size * rank // world
+    hi = size * (rank + 1) // world
+    with open(path, "rb") as f:
+        if lo > 0:
+            f.seek(lo - 1)
+            # advance to the start of the next full line
+            chunk = f.read(1)
+            if chunk != b"\n":
+                f.readline()
+            lo = f.tell()
+        f.seek(lo)
+        data = f.read(hi - lo)
+        if hi < size and (not data or data[-1:] != b"\n"):
+            # finish the straddling line
+            with open(path, "rb") as g:
+                g.seek(lo + len(data))
+                data += g.readline()
+    return data
+
+
+def read_ldbc_edges(efile: str, weighted: bool = False, rank: int = 0,
+                    world: int = 1):
+    """Parse this rank's slice of an LDBC .e file (src dst [weight])."""
+    data = _byte_slice(efile, rank, world)
+    if not data.strip():
+        empty = np.zeros(0, dtype=np.int64)
+        return (empty, empty.copy(),
+                np.zeros(0, dtype=np.float32) if weighted else None)
+    fields = np.array(data.split(), dtype=np.float64 if weighted else np.int64)
+    ncols = 3 if weighted else 2
+    fields = fields.reshape(-1, ncols)
+    src = fields[:, 0].astype(np.int64)
+    dst = fields[:, 1].astype(np.int64)
+    w = fields[:, 2].astype(np.float32) if weighted else None
+    return src, dst, w
+
+
+def read_ldbc_vertices(vfile: str, rank: int = 0, world: int = 1):
+    """Parse this rank's slice of an LDBC .v file (oid [vdata])."""
+    data = _byte_slice(vfile, rank, world)
+    if not data.strip():
+        return np.zeros(0, dtype=np.int64)
+    oids = [int(line.split()[0]) for line in data.splitlines() if line.strip()]
+    return np.array(oids, dtype=np.int64)

@@ -1,0 +1,69 @@
+"""Subprocess worker for multi-rank CPU tests.
+
+Launched by test_multiproc.py with RANK/WORLD_SIZE/MASTER_ADDR/MASTER_PORT
+set; loads a deterministic slice of the shared synthetic graph, runs the
+requested app, writes this rank's (oids, values) to OUT_DIR/rank<k>.npz.
+"""
+import json
+import os
+import sys
+from pathlib import Path
+
+import numpy as np
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+import grapehip  # noqa: E402
+
+
+def main():
+    cfg = json.loads(os.environ["GRAPEHIP_TEST_CFG"])
+    rank = int(os.environ["RANK"])
+    world = int(os.environ["WORLD_SIZE"])
+    eng = grapehip.engine_from_env(n_threads=2)
+
+    rng = np.random.default_rng(cfg["seed"])
+    src = rng.integers(0, cfg["num_v"], size=cfg["num_e"], dtype=np.int64)
+    dst = rng.integers(0, cfg["num_v"], size=cfg["num_e"], dtype=np.int64)
+    keep = src != dst
+    src, dst = src[keep], dst[keep]
+    w = None
+    if cfg["weighted"]:
+        w = rng.random(len(src), dtype=np.float32) * 9 + 1
+    # each rank takes a strided slice of the same global edge list
+    sl = slice(rank, None, world)
+    kw = {}
+    if cfg.get("vertex_oids"):
+        # split vertex list round-robin too
+        all_oids = np.arange(cfg["num_v"], dtype=np.int64) * 3 + 1
+        kw["vertex_oids"] = all_oids[sl]
+        src = all_oids[src]
+        dst = all_oids[dst]
+    else:
+        kw["num_vertices"] = cfg["num_v"]
+    g = eng.load_edges(src[sl], dst[sl],
+                       weights=None if w is None else w[sl],
+                       directed=cfg["directed"],
+                       build_in_csr=cfg["in_csr"], **kw)
+
+    app = cfg["app"]
+    if app == "bfs":
+        res = eng.bfs(g, cfg["source"])
+    elif app == "sssp":
+        res = eng.sssp(g, cfg["source"])
+    elif app == "pagerank":
+        res = eng.pagerank(g, 0.85, 10)
+    elif app == "wcc":
+        res = eng.wcc(g)
+    elif app == "cdlp":
+        res = eng.cdlp(g, 10)
+    elif app == "lcc":
+        res = eng.lcc(g)
+    else:
+        raise ValueError(app)
+
+    np.savez(os.path.join(cfg["out_dir"], f"rank{rank}.npz"),
+             oids=res["oids"], values=res["values"])
+
+
+if __name__ == "__main__":
+    main()
