@@ -1,0 +1,119 @@
+#!/usr/bin/env python3
+"""grapehip flagship benchmark — the BASELINE.json headline metric:
+LDBC Graphalytics makespan + TEPS for BFS / SSSP / PageRank on a
+datagen-9_0-fb-shaped graph (|V|=404,817,003, |E|=1,010,447,118; synthetic
+RMAT stand-in with random [1,100) weights — no network, no datasets), on
+1..8 MI355X GPUs, strong scaling (fixed graph partitioned across ranks).
+
+One "step" = one full suite run: BFS + SSSP + PageRank(10 iters) on the
+resident graph. value = aggregate traversed-edges/second over the timed
+steps (TEPS, whole-job across all ranks). The graph build (generation +
+CSR in HBM) is untimed setup, like LDBC's load phase.
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  # N>1: python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+  #        --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+# datagen-9_0-fb shape (LDBC Graphalytics XL dataset)
+DATAGEN_9_0_FB_V = 404_817_003
+DATAGEN_9_0_FB_E = 1_010_447_118
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--source", type=int, default=0)
+    ap.add_argument("--pr-iters", type=int, default=10)
+    ap.add_argument("--nv", type=int, default=DATAGEN_9_0_FB_V)
+    ap.add_argument("--ne", type=int, default=DATAGEN_9_0_FB_E)
+    ap.add_argument("--seed", type=int, default=42)
+    args = ap.parse_args()
+
+    import grapehip
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    if world != args.gpus and "WORLD_SIZE" not in os.environ:
+        world = args.gpus
+    eng = grapehip.engine_from_env(gpu=True) if world > 1 or "RANK" in os.environ \
+        else grapehip.Engine(rank=0, world=1, master_port=29517, gpu=True)
+
+    t_build0 = time.time()
+    g = eng.load_synthetic(num_vertices=args.nv, num_edges=args.ne,
+                           seed=args.seed, directed=False, weighted=True)
+    t_build = time.time() - t_build0
+
+    def suite():
+        r_bfs = eng.bfs(g, args.source)
+        r_sssp = eng.sssp(g, args.source)
+        r_pr = eng.pagerank(g, 0.85, args.pr_iters)
+        traversed = (g.input_edges + g.input_edges +
+                     args.pr_iters * g.num_edges)
+        return traversed, {"bfs_ms": r_bfs["seconds"] * 1e3,
+                           "sssp_ms": r_sssp["seconds"] * 1e3,
+                           "pr_ms": r_pr["seconds"] * 1e3,
+                           "bfs_rounds": r_bfs["rounds"],
+                           "sssp_rounds": r_sssp["rounds"]}
+
+    for _ in range(args.warmup):
+        suite()
+
+    eng.barrier()
+    eng.device_sync()
+    t0 = time.time()
+    traversed_total = 0
+    last = {}
+    for _ in range(args.steps):
+        t, last = suite()
+        traversed_total += t
+    eng.device_sync()
+    eng.barrier()
+    t1 = time.time()
+
+    elapsed = eng.allreduce_max(t1 - t0)
+    teps = traversed_total / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+
+    if rank == 0:
+        out = {
+            "metric": "TEPS_bfs+sssp+pagerank_datagen-9_0-fb",
+            "value": teps,
+            "unit": "TEPS",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "fp32/fp64",
+            "data": "synthetic",
+            "config": {
+                "model": "LDBC suite: BFS + SSSP + PageRank(x%d)" % args.pr_iters,
+                "graph": "datagen-9_0-fb-shaped RMAT (synthetic)",
+                "num_vertices": args.nv,
+                "num_edges": args.ne,
+                "stored_edges": g.num_edges,
+                "weights": "uniform [1,100) fp32",
+                "source": args.source,
+                "parallelism": "graph partitioned over %d GPU(s), RCCL/xGMI halo" % world,
+                "build_seconds": t_build,
+                "per_app_last_step": last,
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

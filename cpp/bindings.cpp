@@ -1,9 +1,9 @@
 // grapehip — Python bindings (pybind11).
 //
 // Engine = one rank of the distributed runtime (TCP control plane; RCCL data
-// plane on GPU). Graph = one edge-cut fragment (host CSR + optional device
-// mirror). App entry points return (oid array, value array) for this rank's
-// inner vertices plus timing metadata.
+// plane on GPU). Graph = one edge-cut fragment (host CSR and/or device
+// graph). App entry points return (oid array, value array) for this rank's
+// owned vertices plus timing metadata.
 #include <pybind11/numpy.h>
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
@@ -39,10 +39,28 @@ double now_s() {
 
 struct PyGraph {
   std::shared_ptr<VertexMap> vm;
-  std::unique_ptr<Fragment> frag;
+  std::unique_ptr<Fragment> frag;  // host CSR (may be null for device-only)
 #ifdef GRAPEHIP_WITH_HIP
   std::unique_ptr<DeviceGraph> dev;
 #endif
+  uint64_t nv() const {
+#ifdef GRAPEHIP_WITH_HIP
+    if (dev) return dev->nv_global;
+#endif
+    return frag->total_vertices();
+  }
+  uint64_t ne() const {
+#ifdef GRAPEHIP_WITH_HIP
+    if (dev) return dev->total_edges;
+#endif
+    return frag->total_edges();
+  }
+  uint64_t input_ne() const {
+#ifdef GRAPEHIP_WITH_HIP
+    if (dev) return dev->input_edges;
+#endif
+    return frag->input_edges();
+  }
 };
 
 struct PyEngine {
@@ -74,14 +92,12 @@ struct PyEngine {
   TcpComm* c() { return world > 1 ? &comm : nullptr; }
 };
 
-// Build fragment from this rank's slice of the edge list.
 using arr_i64 = py::array_t<int64_t, py::array::c_style | py::array::forcecast>;
 using arr_f32 = py::array_t<float, py::array::c_style | py::array::forcecast>;
 
 std::shared_ptr<PyGraph> load_edges(
-    PyEngine& eng, arr_i64 src, arr_i64 dst,
-    std::optional<arr_f32> weights, bool directed,
-    int64_t num_vertices, std::optional<arr_i64> vertex_oids,
+    PyEngine& eng, arr_i64 src, arr_i64 dst, std::optional<arr_f32> weights,
+    bool directed, int64_t num_vertices, std::optional<arr_i64> vertex_oids,
     bool build_in_csr, const std::string& partitioner) {
   size_t n = src.size();
   if (static_cast<size_t>(dst.size()) != n)
@@ -92,8 +108,7 @@ std::shared_ptr<PyGraph> load_edges(
     auto s = src.unchecked<1>();
     auto d = dst.unchecked<1>();
     const float* w = weighted ? weights->data() : nullptr;
-    for (size_t i = 0; i < n; ++i)
-      edges[i] = {s(i), d(i), w ? w[i] : 1.0f};
+    for (size_t i = 0; i < n; ++i) edges[i] = {s(i), d(i), w ? w[i] : 1.0f};
   }
 
   auto g = std::make_shared<PyGraph>();
@@ -101,17 +116,11 @@ std::shared_ptr<PyGraph> load_edges(
   py::gil_scoped_release rel;
 
   if (vertex_oids.has_value()) {
-    // hashmap idxer: route each oid to its owner, owners dedup + index
-    PartitionerKind pk = partitioner == "hash" ? PartitionerKind::kHash
-                                               : PartitionerKind::kHash;
-    // (segmented needs dense oids; with explicit oids use hash)
     const int64_t* po = vertex_oids->data();
     size_t nv = vertex_oids->size();
     std::vector<std::vector<oid_t>> bins(eng.world);
-    for (size_t i = 0; i < nv; ++i) {
-      oid_t o = po[i];
-      bins[hash_oid(o) % eng.world].push_back(o);
-    }
+    for (size_t i = 0; i < nv; ++i)
+      bins[hash_oid(po[i]) % eng.world].push_back(po[i]);
     std::vector<std::string> send(eng.world);
     for (int f = 0; f < eng.world; ++f)
       send[f].assign(reinterpret_cast<const char*>(bins[f].data()),
@@ -126,7 +135,9 @@ std::shared_ptr<PyGraph> load_edges(
     }
     std::sort(owned.begin(), owned.end());
     owned.erase(std::unique(owned.begin(), owned.end()), owned.end());
-    g->vm->init_hashmap(eng.world, pk, eng.c(), std::move(owned));
+    g->vm->init_hashmap(eng.world, PartitionerKind::kHash, eng.c(),
+                        std::move(owned));
+    (void)partitioner;
   } else {
     if (num_vertices <= 0)
       throw std::runtime_error("num_vertices required for identity mapping");
@@ -157,7 +168,7 @@ py::array_t<int64_t> inner_oids(const Fragment& f) {
 }
 
 template <typename RunFn>
-py::dict run_timed(PyEngine& eng, const Fragment& frag, RunFn&& run) {
+py::dict run_timed(PyEngine& eng, RunFn&& run) {
   py::dict out;
   double t0, t1;
   int rounds;
@@ -169,10 +180,29 @@ py::dict run_timed(PyEngine& eng, const Fragment& frag, RunFn&& run) {
     if (eng.c()) eng.c()->barrier();
     t1 = now_s();
   }
+  double secs = t1 - t0;
+  if (eng.c()) secs = eng.c()->allreduce_max_double(secs);
   out["rounds"] = rounds;
-  out["seconds"] = t1 - t0;
+  out["seconds"] = secs;
   return out;
 }
+
+#ifdef GRAPEHIP_WITH_HIP
+py::dict gpu_dict(const GpuRunResult& r, const DeviceGraph& g, bool is_i64) {
+  py::dict out;
+  out["rounds"] = r.rounds;
+  out["seconds"] = r.seconds;
+  out["traversed_edges"] = r.traversed_edges;
+  uint32_t owned = g.owned();
+  py::array_t<int64_t> oids(owned);
+  auto* p = oids.mutable_data();
+  for (uint32_t i = 0; i < owned; ++i)
+    p[i] = static_cast<int64_t>(g.v_begin) + i;
+  out["oids"] = oids;
+  out["values"] = is_i64 ? py::object(to_np(r.i64)) : py::object(to_np(r.f64));
+  return out;
+}
+#endif
 
 }  // namespace
 
@@ -180,22 +210,23 @@ PYBIND11_MODULE(_core, m) {
   m.doc() = "grapehip core engine";
 
   py::class_<PyGraph, std::shared_ptr<PyGraph>>(m, "Graph")
+      .def_property_readonly("num_vertices",
+                             [](const PyGraph& g) { return g.nv(); })
+      .def_property_readonly("num_edges",
+                             [](const PyGraph& g) { return g.ne(); })
+      .def_property_readonly("input_edges",
+                             [](const PyGraph& g) { return g.input_ne(); })
       .def_property_readonly(
-          "num_vertices",
-          [](const PyGraph& g) { return g.frag->total_vertices(); })
-      .def_property_readonly(
-          "num_edges", [](const PyGraph& g) { return g.frag->total_edges(); })
-      .def_property_readonly(
-          "input_edges",
-          [](const PyGraph& g) { return g.frag->input_edges(); })
-      .def_property_readonly(
-          "ivnum", [](const PyGraph& g) { return g.frag->ivnum(); })
-      .def_property_readonly(
-          "ovnum", [](const PyGraph& g) { return g.frag->ovnum(); })
-      .def_property_readonly("directed",
-                             [](const PyGraph& g) { return g.frag->directed(); })
-      .def("inner_oids",
-           [](const PyGraph& g) { return inner_oids(*g.frag); });
+          "ivnum",
+          [](const PyGraph& g) {
+#ifdef GRAPEHIP_WITH_HIP
+            if (g.dev) return static_cast<uint32_t>(g.dev->owned());
+#endif
+            return g.frag->ivnum();
+          })
+      .def_property_readonly("ovnum", [](const PyGraph& g) {
+        return g.frag ? g.frag->ovnum() : 0;
+      });
 
   py::class_<PyEngine>(m, "Engine")
       .def(py::init<int, int, const std::string&, int, int, bool>(),
@@ -205,31 +236,66 @@ PYBIND11_MODULE(_core, m) {
            py::arg("gpu") = false)
       .def_readonly("rank", &PyEngine::rank)
       .def_readonly("world", &PyEngine::world)
-      .def_property_readonly("gpu", [](const PyEngine& e) { return e.use_gpu; })
+      .def_property_readonly("gpu",
+                             [](const PyEngine& e) { return e.use_gpu; })
       .def("barrier",
            [](PyEngine& e) {
              py::gil_scoped_release rel;
              if (e.c()) e.c()->barrier();
            })
+      .def("allreduce_max",
+           [](PyEngine& e, double v) {
+             py::gil_scoped_release rel;
+             return e.c() ? e.c()->allreduce_max_double(v) : v;
+           })
+      .def("device_sync",
+           [](PyEngine& e) {
+#ifdef GRAPEHIP_WITH_HIP
+             if (e.gpu) e.gpu->device_sync();
+#endif
+           })
       .def("load_edges", &load_edges, py::arg("src"), py::arg("dst"),
            py::arg("weights") = std::nullopt, py::arg("directed") = false,
-           py::arg("num_vertices") = -1,
-           py::arg("vertex_oids") = std::nullopt,
+           py::arg("num_vertices") = -1, py::arg("vertex_oids") = std::nullopt,
            py::arg("build_in_csr") = false,
            py::arg("partitioner") = "segmented")
+      .def("load_synthetic",
+           [](PyEngine& eng, uint64_t nv, uint64_t ne, uint64_t seed,
+              bool directed, bool weighted, double a, double b, double c) {
+#ifdef GRAPEHIP_WITH_HIP
+             if (!eng.use_gpu)
+               throw std::runtime_error("load_synthetic requires gpu=True");
+             auto g = std::make_shared<PyGraph>();
+             py::gil_scoped_release rel;
+             g->dev = eng.gpu->gen_synthetic(nv, ne, seed, directed, weighted,
+                                             false, a, b, c);
+             return g;
+#else
+             throw std::runtime_error("built without HIP");
+#endif
+           },
+           py::arg("num_vertices"), py::arg("num_edges"), py::arg("seed") = 42,
+           py::arg("directed") = false, py::arg("weighted") = false,
+           py::arg("a") = 0.57, py::arg("b") = 0.19, py::arg("c") = 0.19)
       .def("bfs",
            [](PyEngine& eng, PyGraph& g, int64_t source) {
 #ifdef GRAPEHIP_WITH_HIP
-             if (eng.use_gpu) return eng.gpu->bfs(*g.dev, source);
+             if (eng.use_gpu) {
+               GpuRunResult r;
+               {
+                 py::gil_scoped_release rel;
+                 r = eng.gpu->bfs(*g.dev, source);
+               }
+               return gpu_dict(r, *g.dev, true);
+             }
 #endif
              BFSApp app;
              BFSContext ctx;
              MessageManager mm;
              mm.init(eng.c(), g.frag.get(), eng.n_threads);
              ctx.init(*g.frag, source);
-             py::dict meta = run_timed(eng, *g.frag, [&] {
-               return RunWorker(app, ctx, *g.frag, mm);
-             });
+             py::dict meta = run_timed(
+                 eng, [&] { return RunWorker(app, ctx, *g.frag, mm); });
              std::vector<int64_t> vals(g.frag->ivnum());
              for (vid_t v = 0; v < g.frag->ivnum(); ++v)
                vals[v] = ctx.depth[v].load(std::memory_order_relaxed);
@@ -239,18 +305,24 @@ PYBIND11_MODULE(_core, m) {
            },
            py::arg("graph"), py::arg("source") = 0)
       .def("sssp",
-           [](PyEngine& eng, PyGraph& g, int64_t source) {
+           [](PyEngine& eng, PyGraph& g, int64_t source, float delta) {
 #ifdef GRAPEHIP_WITH_HIP
-             if (eng.use_gpu) return eng.gpu->sssp(*g.dev, source);
+             if (eng.use_gpu) {
+               GpuRunResult r;
+               {
+                 py::gil_scoped_release rel;
+                 r = eng.gpu->sssp(*g.dev, source, delta);
+               }
+               return gpu_dict(r, *g.dev, false);
+             }
 #endif
              SSSPApp app;
              SSSPContext ctx;
              MessageManager mm;
              mm.init(eng.c(), g.frag.get(), eng.n_threads);
              ctx.init(*g.frag, source);
-             py::dict meta = run_timed(eng, *g.frag, [&] {
-               return RunWorker(app, ctx, *g.frag, mm);
-             });
+             py::dict meta = run_timed(
+                 eng, [&] { return RunWorker(app, ctx, *g.frag, mm); });
              std::vector<double> vals(g.frag->ivnum());
              for (vid_t v = 0; v < g.frag->ivnum(); ++v)
                vals[v] = ctx.dist[v].load(std::memory_order_relaxed);
@@ -258,20 +330,26 @@ PYBIND11_MODULE(_core, m) {
              meta["values"] = to_np(vals);
              return meta;
            },
-           py::arg("graph"), py::arg("source") = 0)
+           py::arg("graph"), py::arg("source") = 0, py::arg("delta") = -1.0f)
       .def("pagerank",
            [](PyEngine& eng, PyGraph& g, double damping, int iters) {
 #ifdef GRAPEHIP_WITH_HIP
-             if (eng.use_gpu) return eng.gpu->pagerank(*g.dev, damping, iters);
+             if (eng.use_gpu) {
+               GpuRunResult r;
+               {
+                 py::gil_scoped_release rel;
+                 r = eng.gpu->pagerank(*g.dev, damping, iters);
+               }
+               return gpu_dict(r, *g.dev, false);
+             }
 #endif
              PageRankApp app;
              PageRankContext ctx;
              MessageManager mm;
              mm.init(eng.c(), g.frag.get(), eng.n_threads);
              ctx.init(*g.frag, damping, iters);
-             py::dict meta = run_timed(eng, *g.frag, [&] {
-               return RunWorker(app, ctx, *g.frag, mm);
-             });
+             py::dict meta = run_timed(
+                 eng, [&] { return RunWorker(app, ctx, *g.frag, mm); });
              meta["oids"] = inner_oids(*g.frag);
              meta["values"] = to_np(ctx.rank);
              return meta;
@@ -280,16 +358,22 @@ PYBIND11_MODULE(_core, m) {
       .def("wcc",
            [](PyEngine& eng, PyGraph& g) {
 #ifdef GRAPEHIP_WITH_HIP
-             if (eng.use_gpu) return eng.gpu->wcc(*g.dev);
+             if (eng.use_gpu) {
+               GpuRunResult r;
+               {
+                 py::gil_scoped_release rel;
+                 r = eng.gpu->wcc(*g.dev);
+               }
+               return gpu_dict(r, *g.dev, true);
+             }
 #endif
              WCCApp app;
              WCCContext ctx;
              MessageManager mm;
              mm.init(eng.c(), g.frag.get(), eng.n_threads);
              ctx.init(*g.frag);
-             py::dict meta = run_timed(eng, *g.frag, [&] {
-               return RunWorker(app, ctx, *g.frag, mm);
-             });
+             py::dict meta = run_timed(
+                 eng, [&] { return RunWorker(app, ctx, *g.frag, mm); });
              std::vector<int64_t> vals(g.frag->ivnum());
              for (vid_t v = 0; v < g.frag->ivnum(); ++v)
                vals[v] = ctx.label[v].load(std::memory_order_relaxed);
@@ -301,16 +385,22 @@ PYBIND11_MODULE(_core, m) {
       .def("cdlp",
            [](PyEngine& eng, PyGraph& g, int iters) {
 #ifdef GRAPEHIP_WITH_HIP
-             if (eng.use_gpu) return eng.gpu->cdlp(*g.dev, iters);
+             if (eng.use_gpu) {
+               GpuRunResult r;
+               {
+                 py::gil_scoped_release rel;
+                 r = eng.gpu->cdlp(*g.dev, iters);
+               }
+               return gpu_dict(r, *g.dev, true);
+             }
 #endif
              CDLPApp app;
              CDLPContext ctx;
              MessageManager mm;
              mm.init(eng.c(), g.frag.get(), eng.n_threads);
              ctx.init(*g.frag, iters);
-             py::dict meta = run_timed(eng, *g.frag, [&] {
-               return RunWorker(app, ctx, *g.frag, mm);
-             });
+             py::dict meta = run_timed(
+                 eng, [&] { return RunWorker(app, ctx, *g.frag, mm); });
              meta["oids"] = inner_oids(*g.frag);
              std::vector<int64_t> vals(ctx.label.begin(),
                                        ctx.label.begin() + g.frag->ivnum());
@@ -321,16 +411,22 @@ PYBIND11_MODULE(_core, m) {
       .def("lcc",
            [](PyEngine& eng, PyGraph& g) {
 #ifdef GRAPEHIP_WITH_HIP
-             if (eng.use_gpu) return eng.gpu->lcc(*g.dev);
+             if (eng.use_gpu) {
+               GpuRunResult r;
+               {
+                 py::gil_scoped_release rel;
+                 r = eng.gpu->lcc(*g.dev);
+               }
+               return gpu_dict(r, *g.dev, false);
+             }
 #endif
              LCCApp app;
              LCCContext ctx;
              MessageManager mm;
              mm.init(eng.c(), g.frag.get(), eng.n_threads);
              ctx.init(*g.frag);
-             py::dict meta = run_timed(eng, *g.frag, [&] {
-               return RunWorker(app, ctx, *g.frag, mm);
-             });
+             py::dict meta = run_timed(
+                 eng, [&] { return RunWorker(app, ctx, *g.frag, mm); });
              meta["oids"] = inner_oids(*g.frag);
              meta["values"] = to_np(ctx.lcc);
              return meta;

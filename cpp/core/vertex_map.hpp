@@ -11,7 +11,9 @@
 //     HashMapIdxer, vertex_map.h:312).
 #pragma once
 
+#include <algorithm>
 #include <cstdint>
+#include <cstring>
 #include <unordered_map>
 #include <vector>
 
@@ -58,11 +60,12 @@ class VertexMap {
     idxer_ = IdxerKind::kIdentity;
     pkind_ = PartitionerKind::kSegmented;
     total_vertices_ = nv;
+    // uniform ceil-slices: owner(v) == min(v / slice, fnum-1) — O(1) on the
+    // GPU and reduce-scatter/allgather slices are equal-sized (last padded).
     seg_.resize(fnum + 1);
-    uint64_t per = nv / fnum, rem = nv % fnum;
-    seg_[0] = 0;
-    for (int f = 0; f < fnum; ++f)
-      seg_[f + 1] = seg_[f] + per + (static_cast<uint64_t>(f) < rem ? 1 : 0);
+    uint64_t slice = (nv + fnum - 1) / fnum;
+    for (int f = 0; f <= fnum; ++f)
+      seg_[f] = std::min<uint64_t>(static_cast<uint64_t>(f) * slice, nv);
   }
 
   // Hashmap mode: every rank supplies the oids it OWNS (dedup'd); lids are

@@ -1,0 +1,83 @@
+// grapehip — GPU engine interface (host side).
+//
+// Reference parity: grape/cuda/worker/gpu_worker.h + gpu_message_manager.h
+// reimagined: TCP control plane (lengths, termination, scalar collectives) +
+// RCCL-over-xGMI data plane, dual HIP streams. One GpuContext per process ==
+// one MI355X GPU == one fragment.
+#pragma once
+
+#include <cstdint>
+#include <memory>
+#include <vector>
+
+#include "../core/fragment.hpp"
+#include "../core/net.hpp"
+#include "../core/types.hpp"
+#include "hip_common.hpp"
+
+struct ncclComm;  // fwd (rccl.h included in the .hip TU)
+
+namespace grapehip {
+
+struct DeviceGraph {
+  uint32_t nv_global = 0;
+  uint32_t v_begin = 0, v_end = 0;
+  bool directed = false, weighted = false, has_in = false;
+  uint64_t local_edges = 0, total_edges = 0, input_edges = 0;
+  std::vector<uint32_t> seg_host;
+  DeviceBuffer<uint32_t> seg;
+  DeviceBuffer<uint64_t> oe_off;
+  DeviceBuffer<uint32_t> oe_dst;
+  DeviceBuffer<float> oe_w;
+  DeviceBuffer<uint64_t> ie_off;
+  DeviceBuffer<uint32_t> ie_dst;
+  DeviceBuffer<float> ie_w;
+  uint32_t owned() const { return v_end - v_begin; }
+};
+
+struct GpuRunResult {
+  std::vector<int64_t> i64;   // BFS depth / WCC & CDLP labels (owned range)
+  std::vector<double> f64;    // SSSP dist / PR rank / LCC coeff
+  int rounds = 0;
+  double seconds = 0;         // max over ranks, kernel-side barrier-bracketed
+  uint64_t traversed_edges = 0;  // for TEPS accounting (global)
+};
+
+class GpuContext {
+ public:
+  GpuContext(TcpComm* comm, int rank, int world);
+  ~GpuContext();
+
+  // Build a device graph from a host fragment (identity vertex map only).
+  std::unique_ptr<DeviceGraph> upload(const Fragment& frag);
+
+  // Generate an LDBC-datagen-shaped synthetic graph directly in HBM3E
+  // (RMAT skew, random [1,100) weights), one owned slice per rank.
+  std::unique_ptr<DeviceGraph> gen_synthetic(uint64_t nv, uint64_t ne,
+                                             uint64_t seed, bool directed,
+                                             bool weighted, bool build_in_csr,
+                                             double rmat_a, double rmat_b,
+                                             double rmat_c);
+
+  GpuRunResult bfs(const DeviceGraph& g, int64_t source);
+  GpuRunResult sssp(const DeviceGraph& g, int64_t source, float delta);
+  GpuRunResult pagerank(const DeviceGraph& g, double damping, int iters);
+  GpuRunResult wcc(const DeviceGraph& g);
+  GpuRunResult cdlp(const DeviceGraph& g, int iters);
+  GpuRunResult lcc(const DeviceGraph& g);
+
+  void device_sync();
+  int device_id() const { return dev_; }
+  TcpComm* comm() { return comm_; }
+  int rank() const { return rank_; }
+  int world() const { return world_; }
+
+  struct Impl;  // internal (kernels + scratch); public for free helpers
+
+ private:
+  std::unique_ptr<Impl> impl_;
+  TcpComm* comm_;
+  int rank_, world_, dev_ = 0;
+};
+
+}  // namespace grapehip
