@@ -243,6 +243,14 @@ PYBIND11_MODULE(_core, m) {
              py::gil_scoped_release rel;
              if (e.c()) e.c()->barrier();
            })
+      .def("_debug_scan",
+           [](PyEngine& e, std::vector<uint32_t> in) {
+#ifdef GRAPEHIP_WITH_HIP
+             return e.gpu->debug_scan(in);
+#else
+             throw std::runtime_error("no hip");
+#endif
+           })
       .def("allreduce_max",
            [](PyEngine& e, double v) {
              py::gil_scoped_release rel;

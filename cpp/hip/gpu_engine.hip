@@ -126,17 +126,18 @@ __global__ void scan_pass_b(const IN* __restrict__ in, size_t n,
 }
 
 // Grow-only per-level scratch so per-round frontier scans never hipMalloc.
+// Levels are FIXED-depth (2048^4 > 2^44 elements) so references returned by
+// sum_buf/off_buf stay valid while deeper recursion levels allocate.
 struct ScanTemp {
-  std::vector<DeviceBuffer<uint64_t>> sums;   // per level: block sums
-  std::vector<DeviceBuffer<uint64_t>> offs;   // per level: scanned sums
-  DeviceBuffer<uint64_t> zero;                // single 0 (spine base)
+  static constexpr size_t kMaxLevels = 4;
+  DeviceBuffer<uint64_t> sums[kMaxLevels];   // per level: block sums
+  DeviceBuffer<uint64_t> offs[kMaxLevels];   // per level: scanned sums
+  DeviceBuffer<uint64_t> zero;               // single 0 (spine base)
   DeviceBuffer<uint64_t>& sum_buf(size_t l, size_t n) {
-    if (sums.size() <= l) sums.resize(l + 1);
     if (sums[l].size() < n) sums[l].resize(n + (n >> 2) + 16);
     return sums[l];
   }
   DeviceBuffer<uint64_t>& off_buf(size_t l, size_t n) {
-    if (offs.size() <= l) offs.resize(l + 1);
     if (offs[l].size() < n) offs[l].resize(n + (n >> 2) + 16);
     return offs[l];
   }
@@ -155,6 +156,8 @@ void scan_recurse(const IN* d_in, uint64_t* d_out, size_t n, hipStream_t s,
                        tmp.zero.data(), d_out);
     return;
   }
+  if (level >= ScanTemp::kMaxLevels)
+    throw std::runtime_error("scan: input too large");
   auto& bsums = tmp.sum_buf(level, nblocks);
   auto& boffs = tmp.off_buf(level, nblocks);
   hipLaunchKernelGGL(scan_pass_a<IN>, dim3(nblocks), dim3(kBlock), 0, s, d_in,
@@ -772,6 +775,15 @@ GpuContext::~GpuContext() {
 
 void GpuContext::device_sync() { HIP_CHECK(hipDeviceSynchronize()); }
 
+std::vector<uint64_t> GpuContext::debug_scan(const std::vector<uint32_t>& in) {
+  hipStream_t s = impl_->compute;
+  DeviceBuffer<uint32_t> d_in;
+  d_in.upload(in, s);
+  DeviceBuffer<uint64_t> d_out(in.size() + 1);
+  exclusive_scan(d_in.data(), d_out.data(), in.size(), s, impl_->scan);
+  return d_out.download(s);
+}
+
 // ---------------------------------------------------------------------------
 // Graph construction
 // ---------------------------------------------------------------------------
@@ -793,6 +805,8 @@ void build_csr_from_coo(const DeviceBuffer<uint32_t>& src,
                                                     deg.data());
   out_off.resize(owned + 1);
   uint64_t total = exclusive_scan(deg.data(), out_off.data(), owned, s, scan);
+  fprintf(stderr, "[csr] scan total=%lu n=%lu\n", (unsigned long)total,
+          (unsigned long)n);
   if (total != n) throw std::runtime_error("CSR build: degree sum mismatch");
   deg.free();
   DeviceBuffer<unsigned long long> cursor(owned);
@@ -844,6 +858,8 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
   DeviceBuffer<float> e_w(weighted ? est : 0);
   DeviceBuffer<unsigned long long> cnt(1);
   cnt.zero(s);
+  fprintf(stderr, "[gen] nv=%u ne=%lu est=%lu scale=%d\n", g->nv_global,
+          (unsigned long)ne, (unsigned long)est, scale);
   gen_edges_kernel<<<kMaxGrid, kBlock, 0, s>>>(
       ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, g->v_begin, g->v_end,
       !directed, weighted, e_src.data(), e_dst.data(),
@@ -852,6 +868,7 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
   HIP_CHECK(hipMemcpyAsync(&n_local, cnt.data(), 8, hipMemcpyDeviceToHost, s));
   HIP_CHECK(hipStreamSynchronize(s));
   if (n_local > est) throw std::runtime_error("gen_synthetic: overflow");
+  fprintf(stderr, "[gen] n_local=%llu\n", n_local);
 
   build_csr_from_coo(e_src, e_dst, e_w, n_local, g->v_begin, owned, weighted,
                      g->oe_off, g->oe_dst, g->oe_w, s, impl_->scan);

@@ -67,6 +67,8 @@ class GpuContext {
   GpuRunResult lcc(const DeviceGraph& g);
 
   void device_sync();
+  // test hook: exclusive scan of host u32 data on the device
+  std::vector<uint64_t> debug_scan(const std::vector<uint32_t>& in);
   int device_id() const { return dev_; }
   TcpComm* comm() { return comm_; }
   int rank() const { return rank_; }
