@@ -32,6 +32,11 @@ struct DeviceGraph {
   DeviceBuffer<uint64_t> ie_off;
   DeviceBuffer<uint32_t> ie_dst;
   DeviceBuffer<float> ie_w;
+  // degree buckets over owned rows (built lazily; row-per-thread/wave/block
+  // scheduling for whole-graph sweeps — the CTA scheduler, preprocessed)
+  DeviceBuffer<uint32_t> rows_small, rows_mid, rows_large;
+  uint64_t n_small = 0, n_mid = 0, n_large = 0;
+  bool buckets_built = false;
   uint32_t owned() const { return v_end - v_begin; }
 };
 
@@ -59,12 +64,12 @@ class GpuContext {
                                              double rmat_a, double rmat_b,
                                              double rmat_c);
 
-  GpuRunResult bfs(const DeviceGraph& g, int64_t source);
-  GpuRunResult sssp(const DeviceGraph& g, int64_t source, float delta);
-  GpuRunResult pagerank(const DeviceGraph& g, double damping, int iters);
-  GpuRunResult wcc(const DeviceGraph& g);
-  GpuRunResult cdlp(const DeviceGraph& g, int iters);
-  GpuRunResult lcc(const DeviceGraph& g);
+  GpuRunResult bfs(DeviceGraph& g, int64_t source);
+  GpuRunResult sssp(DeviceGraph& g, int64_t source, float delta);
+  GpuRunResult pagerank(DeviceGraph& g, double damping, int iters);
+  GpuRunResult wcc(DeviceGraph& g);
+  GpuRunResult cdlp(DeviceGraph& g, int iters);
+  GpuRunResult lcc(DeviceGraph& g);
 
   void device_sync();
   // test hook: exclusive scan of host u32 data on the device
