@@ -347,13 +347,21 @@ __global__ void expand_cm_frontier(DevGraphView g, const uint32_t* __restrict__ 
     }
     __syncthreads();
     const uint64_t e0 = s_off[0], e1 = s_off[n];
-    for (uint64_t e = e0 + tid; e < e1; e += blockDim.x) {
-      // owner = last j with s_off[j] <= e
-      int lo = 0, hi = n - 1;
-      while (lo < hi) {
-        int mid = (lo + hi + 1) >> 1;
-        if (s_off[mid] <= e) lo = mid; else hi = mid - 1;
+    // a thread's edges are strided, so its owner index only moves forward:
+    // binary-search once, then advance incrementally (amortized O(1) LDS)
+    int lo = 0;
+    {
+      uint64_t e = e0 + tid;
+      if (e < e1) {
+        int hi = n - 1;
+        while (lo < hi) {
+          int mid = (lo + hi + 1) >> 1;
+          if (s_off[mid] <= e) lo = mid; else hi = mid - 1;
+        }
       }
+    }
+    for (uint64_t e = e0 + tid; e < e1; e += blockDim.x) {
+      while (lo + 1 < n && s_off[lo + 1] <= e) ++lo;
       const uint32_t u = s_v[lo];
       const uint64_t eid = s_base[lo] + (e - s_off[lo]);
       op(u, g.oe_dst[eid], g.oe_w ? g.oe_w[eid] : 1.0f);
@@ -379,12 +387,19 @@ __global__ void expand_cm_range(DevGraphView g, EdgeOp op) {
     if (tid == 0) s_off[n] = off[chunk + n];
     __syncthreads();
     const uint64_t e0 = s_off[0], e1 = s_off[n];
-    for (uint64_t e = e0 + tid; e < e1; e += blockDim.x) {
-      int lo = 0, hi = n - 1;
-      while (lo < hi) {
-        int mid = (lo + hi + 1) >> 1;
-        if (s_off[mid] <= e) lo = mid; else hi = mid - 1;
+    int lo = 0;
+    {
+      uint64_t e = e0 + tid;
+      if (e < e1) {
+        int hi = n - 1;
+        while (lo < hi) {
+          int mid = (lo + hi + 1) >> 1;
+          if (s_off[mid] <= e) lo = mid; else hi = mid - 1;
+        }
       }
+    }
+    for (uint64_t e = e0 + tid; e < e1; e += blockDim.x) {
+      while (lo + 1 < n && s_off[lo + 1] <= e) ++lo;
       op(g.v_begin + chunk + lo, dst[e], wt ? wt[e] : 1.0f);
     }
     __syncthreads();
@@ -483,13 +498,47 @@ __global__ void clear_bits_kernel(const uint32_t* __restrict__ q, uint32_t n,
 }
 
 // ===========================================================================
+// Bitmap frontier compaction: mark bits during expansion (1 atomicOr per
+// improvement instead of bitmap+queue atomics), then popc+scan+fill to
+// materialize the queue. Clears the bitmap as it fills.
+// ===========================================================================
+
+__global__ void popc_words_kernel(const uint32_t* __restrict__ words,
+                                  size_t nwords, uint32_t* __restrict__ out) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (size_t i = static_cast<size_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+       i < nwords; i += stride)
+    out[i] = __popc(words[i]);
+}
+
+__global__ void fill_frontier_kernel(uint32_t* __restrict__ words,
+                                     size_t nwords,
+                                     const uint64_t* __restrict__ woff,
+                                     uint32_t bit_base,
+                                     uint32_t* __restrict__ q) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (size_t i = static_cast<size_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+       i < nwords; i += stride) {
+    uint32_t bits = words[i];
+    if (!bits) continue;
+    uint64_t base = woff[i];
+    while (bits) {
+      int b = __builtin_ctz(bits);
+      bits &= bits - 1;
+      q[base++] = bit_base + (static_cast<uint32_t>(i) << 5) + b;
+    }
+    words[i] = 0;
+  }
+}
+
+// ===========================================================================
 // App edge-op functors
 // ===========================================================================
 
 struct BfsOp {
   uint32_t* depth;
   uint32_t next_depth;
-  DevQueue next;
+  DevBitmap next_bm;  // over owned bits
   uint32_t v_begin, v_end;
   bool multi;
   DevHalo halo;
@@ -499,7 +548,7 @@ struct BfsOp {
     if (depth[d] <= next_depth) return;
     if (atomicMin(&depth[d], next_depth) > next_depth) {
       if (d >= v_begin && d < v_end)
-        next.push(d);
+        next_bm.set_once(d - v_begin);
       else if (multi)
         halo.add(d);
     }
@@ -508,33 +557,35 @@ struct BfsOp {
 
 struct BfsRecvOp {  // incoming (v, depth) — v owned here
   uint32_t* depth;
-  DevQueue next;
+  DevBitmap next_bm;
+  uint32_t v_begin;
   __device__ __forceinline__ void operator()(uint32_t v, uint32_t dv) const {
-    if (atomicMin(&depth[v], dv) > dv) next.push(v);
+    if (atomicMin(&depth[v], dv) > dv) next_bm.set_once(v - v_begin);
   }
 };
 
-__global__ void bfs_seed_kernel(uint32_t* depth, uint32_t src, uint32_t* q,
-                                unsigned long long* cnt) {
+__global__ void bfs_seed_kernel(uint32_t* depth, uint32_t src,
+                                uint32_t v_begin, uint32_t* bm_words) {
   if (threadIdx.x == 0 && blockIdx.x == 0) {
     depth[src] = 0;
-    q[0] = src;
-    *cnt = 1;
+    uint32_t r = src - v_begin;
+    bm_words[r >> 5] |= 1u << (r & 31);
   }
 }
 
 struct SsspOp {
   float* dist;
   float prio_hi;
-  DevQueue near_q, far_q;
-  DevBitmap near_bm, far_bm;
+  DevBitmap near_bm;  // over owned bits; compacted to a queue per round
+  DevQueue far_q;
+  DevBitmap far_bm;  // over global vids (dedup across rounds)
   uint32_t v_begin, v_end;
   bool multi;
   DevHalo halo;
   __device__ __forceinline__ void bucket(uint32_t d, float nd) const {
     if (d >= v_begin && d < v_end) {
       if (nd < prio_hi) {
-        if (near_bm.set_once(d)) near_q.push(d);
+        near_bm.set_once(d - v_begin);
       } else {
         if (far_bm.set_once(d)) far_q.push(d);
       }
@@ -554,13 +605,15 @@ struct SsspOp {
 struct SsspRecvOp {
   float* dist;
   float prio_hi;
-  DevQueue near_q, far_q;
-  DevBitmap near_bm, far_bm;
+  DevBitmap near_bm;
+  DevQueue far_q;
+  DevBitmap far_bm;
+  uint32_t v_begin;
   __device__ __forceinline__ void operator()(uint32_t v, float dv) const {
     float old = atomicMinPosFloat(&dist[v], dv);
     if (dv < old) {
       if (dv < prio_hi) {
-        if (near_bm.set_once(v)) near_q.push(v);
+        near_bm.set_once(v - v_begin);
       } else {
         if (far_bm.set_once(v)) far_q.push(v);
       }
@@ -568,21 +621,20 @@ struct SsspRecvOp {
   }
 };
 
-__global__ void sssp_seed_kernel(float* dist, uint32_t src, uint32_t* q,
-                                 unsigned long long* cnt, uint32_t* bm_words) {
+__global__ void sssp_seed_kernel(float* dist, uint32_t src, uint32_t v_begin,
+                                 uint32_t* near_words) {
   if (threadIdx.x == 0 && blockIdx.x == 0) {
     dist[src] = 0.0f;
-    q[0] = src;
-    *cnt = 1;
-    atomicOr(&bm_words[src >> 5], 1u << (src & 31));
+    uint32_t r = src - v_begin;
+    near_words[r >> 5] |= 1u << (r & 31);
   }
 }
 
 // far-queue repartition after a priority advance
 __global__ void sssp_repart_kernel(const uint32_t* __restrict__ far_in,
                                    uint32_t n, const float* __restrict__ dist,
-                                   float prio_hi, DevQueue near_q,
-                                   DevQueue far_out, DevBitmap near_bm,
+                                   float prio_hi, DevBitmap near_bm,
+                                   uint32_t v_begin, DevQueue far_out,
                                    DevBitmap far_bm) {
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
@@ -590,7 +642,7 @@ __global__ void sssp_repart_kernel(const uint32_t* __restrict__ far_in,
     uint32_t v = far_in[i];
     if (dist[v] < prio_hi) {
       far_bm.clear_bit(v);
-      if (near_bm.set_once(v)) near_q.push(v);
+      near_bm.set_once(v - v_begin);
     } else {
       far_out.push(v);  // keeps its far bit
     }
@@ -707,21 +759,38 @@ namespace grapehip {
 constexpr uint64_t kSmallDeg = 64;
 constexpr uint64_t kMidDeg = 16384;
 
+// Block-aggregated: LDS tallies + one global reservation per block per
+// bucket (405M per-thread global atomics on 3 hot words took 4.8 s).
 __global__ void bucket_rows_kernel(const uint64_t* __restrict__ off,
                                    uint32_t owned, uint32_t* sm,
                                    unsigned long long* cs, uint32_t* md,
                                    unsigned long long* cm, uint32_t* lg,
                                    unsigned long long* cl) {
-  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
-  for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < owned;
-       r += stride) {
-    uint64_t deg = off[r + 1] - off[r];
-    if (deg < kSmallDeg)
-      sm[atomicAdd(cs, 1ull)] = r;
-    else if (deg < kMidDeg)
-      md[atomicAdd(cm, 1ull)] = r;
-    else
-      lg[atomicAdd(cl, 1ull)] = r;
+  __shared__ uint32_t s_cnt[3];
+  __shared__ unsigned long long s_base[3];
+  uint32_t* lists[3] = {sm, md, lg};
+  unsigned long long* gcnt[3] = {cs, cm, cl};
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t base = blockIdx.x * blockDim.x; base < owned;
+       base += stride) {
+    if (threadIdx.x < 3) s_cnt[threadIdx.x] = 0;
+    __syncthreads();
+    uint32_t r = base + threadIdx.x;
+    int b = -1;
+    uint32_t loc = 0;
+    if (r < owned) {
+      uint64_t deg = off[r + 1] - off[r];
+      b = deg < kSmallDeg ? 0 : (deg < kMidDeg ? 1 : 2);
+      loc = atomicAdd(&s_cnt[b], 1u);
+    }
+    __syncthreads();
+    if (threadIdx.x < 3 && s_cnt[threadIdx.x])
+      s_base[threadIdx.x] =
+          atomicAdd(gcnt[threadIdx.x],
+                    static_cast<unsigned long long>(s_cnt[threadIdx.x]));
+    __syncthreads();
+    if (b >= 0) lists[b][s_base[b] + loc] = r;
+    __syncthreads();
   }
 }
 
@@ -790,6 +859,96 @@ __global__ void pr_pull_large_kernel(const uint64_t* __restrict__ off,
 #pragma unroll
       for (int w = 0; w < kBlock / kWave; ++w) t += s_wave[w];
       acc[v_begin + r] = t;
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Direction-optimizing BFS pull kernels (reference cuda/bfs/bfs.h :206-260):
+// sweep unvisited owned rows looking for a parent at the current level.
+// No atomics, early exit; reuses the degree buckets.
+// ---------------------------------------------------------------------------
+__global__ void bfs_pull_small_kernel(const uint64_t* __restrict__ off,
+                                      const uint32_t* __restrict__ dst,
+                                      const uint32_t* __restrict__ rows,
+                                      uint64_t nrows, uint32_t v_begin,
+                                      uint32_t* __restrict__ depth,
+                                      uint32_t level, DevBitmap next_bm) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * blockDim.x +
+                    threadIdx.x;
+       i < nrows; i += stride) {
+    uint32_t r = rows[i];
+    uint32_t v = v_begin + r;
+    if (depth[v] != 0xFFFFFFFFu) continue;
+    uint64_t b = off[r], e = off[r + 1];
+    for (uint64_t k = b; k < e; ++k) {
+      if (depth[dst[k]] == level) {
+        depth[v] = level + 1;
+        next_bm.set_once(r);
+        break;
+      }
+    }
+  }
+}
+
+__global__ void bfs_pull_mid_kernel(const uint64_t* __restrict__ off,
+                                    const uint32_t* __restrict__ dst,
+                                    const uint32_t* __restrict__ rows,
+                                    uint64_t nrows, uint32_t v_begin,
+                                    uint32_t* __restrict__ depth,
+                                    uint32_t level, DevBitmap next_bm) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wpb = kBlock / kWave;
+  size_t wstride = static_cast<size_t>(gridDim.x) * wpb;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid; i < nrows;
+       i += wstride) {
+    uint32_t r = rows[i];
+    uint32_t v = v_begin + r;
+    if (depth[v] != 0xFFFFFFFFu) continue;
+    uint64_t b = off[r], e = off[r + 1];
+    bool found = false;
+    for (uint64_t kb = b; kb < e; kb += kWave) {
+      uint64_t k = kb + lane;
+      bool mine = k < e && depth[dst[k]] == level;
+      if (__any(mine)) {
+        found = true;
+        break;
+      }
+    }
+    if (found && lane == 0) {
+      depth[v] = level + 1;
+      next_bm.set_once(r);
+    }
+  }
+}
+
+__global__ void bfs_pull_large_kernel(const uint64_t* __restrict__ off,
+                                      const uint32_t* __restrict__ dst,
+                                      const uint32_t* __restrict__ rows,
+                                      uint64_t nrows, uint32_t v_begin,
+                                      uint32_t* __restrict__ depth,
+                                      uint32_t level, DevBitmap next_bm) {
+  __shared__ int s_found;
+  for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
+    uint32_t r = rows[i];
+    uint32_t v = v_begin + r;
+    if (depth[v] != 0xFFFFFFFFu) continue;
+    uint64_t b = off[r], e = off[r + 1];
+    if (threadIdx.x == 0) s_found = 0;
+    __syncthreads();
+    for (uint64_t kb = b; kb < e; kb += blockDim.x) {
+      uint64_t k = kb + threadIdx.x;
+      if (k < e && depth[dst[k]] == level) s_found = 1;
+      __syncthreads();
+      if (s_found) break;
+      __syncthreads();
+    }
+    if (threadIdx.x == 0 && s_found) {
+      depth[v] = level + 1;
+      next_bm.set_once(r);
     }
     __syncthreads();
   }
@@ -1040,6 +1199,53 @@ std::unique_ptr<DeviceGraph> GpuContext::upload(const Fragment& frag) {
 namespace grapehip {
 
 // ---------------------------------------------------------------------------
+// Degree buckets over the pull CSR (undirected: out == neighborhood;
+// directed: in-CSR). Built once per graph, reused by PR pull + BFS pull.
+// ---------------------------------------------------------------------------
+static void ensure_buckets(DeviceGraph& g, hipStream_t s) {
+  if (g.buckets_built) return;
+  uint32_t owned = g.owned();
+  const uint64_t* pull_off = !g.directed ? g.oe_off.data() : g.ie_off.data();
+  g.rows_small.resize(owned);
+  g.rows_mid.resize(owned);
+  g.rows_large.resize(owned);
+  DeviceBuffer<unsigned long long> cnts(3);
+  cnts.zero(s);
+  bucket_rows_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+      pull_off, owned, g.rows_small.data(), cnts.data() + 0,
+      g.rows_mid.data(), cnts.data() + 1, g.rows_large.data(),
+      cnts.data() + 2);
+  auto h = cnts.download(s);
+  g.n_small = h[0];
+  g.n_mid = h[1];
+  g.n_large = h[2];
+  g.buckets_built = true;
+}
+
+// ---------------------------------------------------------------------------
+// Materialize a bitmap frontier into a queue; returns the count.
+// bm covers bits [0, nbits) (owned rows); q entries are global vids
+// (bit_base = v_begin). Clears the bitmap.
+// ---------------------------------------------------------------------------
+static uint64_t compact_frontier(GpuContext::Impl& I, uint32_t* bm_words,
+                                 size_t nbits, uint32_t bit_base, uint32_t* q,
+                                 hipStream_t s) {
+  size_t nwords = (nbits + 31) / 32;
+  if (I.frontier_deg.size() < nwords)
+    I.frontier_deg.resize(nwords + (nwords >> 2) + 64);
+  if (I.frontier_off.size() < nwords + 1)
+    I.frontier_off.resize(nwords + (nwords >> 2) + 65);
+  popc_words_kernel<<<grid_for(nwords), kBlock, 0, s>>>(bm_words, nwords,
+                                                        I.frontier_deg.data());
+  uint64_t total = exclusive_scan(I.frontier_deg.data(),
+                                  I.frontier_off.data(), nwords, s, I.scan);
+  if (total)
+    fill_frontier_kernel<<<grid_for(nwords), kBlock, 0, s>>>(
+        bm_words, nwords, I.frontier_off.data(), bit_base, q);
+  return total;
+}
+
+// ---------------------------------------------------------------------------
 // Halo host-side flush: pack per-peer pairs, exchange counts over TCP,
 // ncclSend/Recv payloads over xGMI. Returns #received pairs (in recvbuf).
 // ---------------------------------------------------------------------------
@@ -1123,12 +1329,21 @@ GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source) {
   DevGraphView view = make_view(g, rank_, world_);
   uint32_t nv_pad = padded_nv(g, world_);
   uint32_t owned = g.owned();
+  uint32_t slice = view.slice;
   uint64_t cap = view.slice;
+  bool multi = world_ > 1;
+
+  const bool pull_capable = !g.directed || g.has_in;
+  if (pull_capable) ensure_buckets(g, s);
+  const uint64_t* pull_off = !g.directed ? g.oe_off.data() : g.ie_off.data();
+  const uint32_t* pull_dst = !g.directed ? g.oe_dst.data() : g.ie_dst.data();
 
   DeviceBuffer<uint32_t> depth(nv_pad);
-  DeviceBuffer<uint32_t> q0(owned + 64), q1(owned + 64);
-  DeviceBuffer<unsigned long long> qcnt(2);
-  bool multi = world_ > 1;
+  DeviceBuffer<uint32_t> q(owned + 64);
+  size_t bm_words = (static_cast<size_t>(owned) + 31) / 32 + 1;
+  DeviceBuffer<uint32_t> next_bm(bm_words);
+  DeviceBuffer<uint32_t> fdeg;   // frontier degrees (separate from compaction
+  DeviceBuffer<uint64_t> foff;   //  scratch so both stay live in one round)
   if (multi) {
     I.halo_idx.resize(static_cast<uint64_t>(world_) * cap);
     I.halo_cnt.resize(world_);
@@ -1140,57 +1355,90 @@ GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source) {
   double t0 = wall_s();
 
   depth.fill_bytes(0xFF, s);
-  qcnt.zero(s);
+  next_bm.zero(s);
   if (multi) {
     I.halo_cnt.zero(s);
     I.halo_bm.zero(s);
   }
   uint32_t src = static_cast<uint32_t>(source);
   if (src >= g.v_begin && src < g.v_end)
-    bfs_seed_kernel<<<1, 1, 0, s>>>(depth.data(), src, q0.data(),
-                                    qcnt.data());
-  uint32_t* qs[2] = {q0.data(), q1.data()};
-  int cur = 0;
+    bfs_seed_kernel<<<1, 1, 0, s>>>(depth.data(), src, g.v_begin,
+                                    next_bm.data());
+  uint32_t qn =
+      static_cast<uint32_t>(compact_frontier(I, next_bm.data(), owned,
+                                             g.v_begin, q.data(), s));
+  uint64_t global_curr = multi ? comm_->allreduce_sum(qn) : qn;
   uint32_t level = 0;
   int rounds = 0;
-  std::vector<unsigned long long> host_cnt(2);
-  for (;;) {
-    HIP_CHECK(hipMemcpyAsync(host_cnt.data(), qcnt.data(), 16,
-                             hipMemcpyDeviceToHost, s));
-    HIP_CHECK(hipStreamSynchronize(s));
-    uint32_t qn = static_cast<uint32_t>(host_cnt[cur]);
-    BfsOp op{depth.data(), level + 1,
-             DevQueue{qs[1 - cur], qcnt.data() + (1 - cur)},
-             g.v_begin, g.v_end, multi,
-             DevHalo{I.halo_idx.data(), I.halo_cnt.data(),
-                     DevBitmap{I.halo_bm.data()}, cap, view.slice, world_}};
-    if (qn) expand_frontier(I, view, qs[cur], qn, op, s);
-    uint64_t next_local = 0;
-    if (multi) {
-      uint64_t nrecv =
-          halo_flush<uint32_t>(I, comm_, rank_, world_, depth.data(), cap, s);
-      if (nrecv)
-        halo_process_kernel<uint32_t, BfsRecvOp>
-            <<<grid_for(nrecv), kBlock, 0, s>>>(
-                reinterpret_cast<HaloPair<uint32_t>*>(I.recvbuf.data()),
-                nrecv,
-                BfsRecvOp{depth.data(),
-                          DevQueue{qs[1 - cur], qcnt.data() + (1 - cur)}});
+  // Beamer-style switch: pull when the frontier's out-edges exceed a
+  // fraction of the stored edges (cuda/bfs/bfs.h heuristic, retuned).
+  const uint64_t pull_edge_threshold = g.total_edges / 16;
+  while (global_curr > 0) {
+    bool use_pull = false;
+    uint64_t fedges = 0;
+    if (qn) {
+      if (fdeg.size() < qn) fdeg.resize(qn + (qn >> 2) + 64);
+      if (foff.size() < qn + 1) foff.resize(qn + (qn >> 2) + 65);
+      gather_deg_kernel<<<grid_for(qn), kBlock, 0, s>>>(view, q.data(), qn,
+                                                        fdeg.data());
+      fedges = exclusive_scan(fdeg.data(), foff.data(), qn, s, I.scan);
     }
-    // reset current queue counter, read next counter
-    HIP_CHECK(hipMemcpyAsync(host_cnt.data(), qcnt.data(), 16,
-                             hipMemcpyDeviceToHost, s));
-    HIP_CHECK(hipStreamSynchronize(s));
-    next_local = host_cnt[1 - cur];
-    unsigned long long zero = 0;
-    HIP_CHECK(hipMemcpyAsync(qcnt.data() + cur, &zero, 8,
-                             hipMemcpyHostToDevice, s));
-    uint64_t next_global =
-        multi ? comm_->allreduce_sum(next_local) : next_local;
+    if (pull_capable) {
+      uint64_t gedges = multi ? comm_->allreduce_sum(fedges) : fedges;
+      use_pull = gedges > pull_edge_threshold;
+    }
+    if (use_pull) {
+      if (multi)
+        NCCL_CHECK(ncclAllGather(
+            depth.data() + static_cast<uint64_t>(rank_) * slice, depth.data(),
+            slice, ncclUint32, I.nccl, s));
+      DevBitmap nb{next_bm.data()};
+      if (g.n_small)
+        bfs_pull_small_kernel<<<grid_for(g.n_small), kBlock, 0, s>>>(
+            pull_off, pull_dst, g.rows_small.data(), g.n_small, g.v_begin,
+            depth.data(), level, nb);
+      if (g.n_mid)
+        bfs_pull_mid_kernel<<<grid_for(g.n_mid * kWave), kBlock, 0, s>>>(
+            pull_off, pull_dst, g.rows_mid.data(), g.n_mid, g.v_begin,
+            depth.data(), level, nb);
+      if (g.n_large)
+        bfs_pull_large_kernel<<<static_cast<int>(std::min<uint64_t>(
+                                    std::max<uint64_t>(g.n_large, 1),
+                                    kMaxGrid)),
+                                kBlock, 0, s>>>(
+            pull_off, pull_dst, g.rows_large.data(), g.n_large, g.v_begin,
+            depth.data(), level, nb);
+    } else {
+      BfsOp op{depth.data(), level + 1, DevBitmap{next_bm.data()}, g.v_begin,
+               g.v_end, multi,
+               DevHalo{I.halo_idx.data(), I.halo_cnt.data(),
+                       DevBitmap{I.halo_bm.data()}, cap, view.slice, world_}};
+      if (qn) {
+        int nchunks = static_cast<int>((qn + kBlock - 1) / kBlock);
+        expand_cm_frontier<BfsOp><<<std::min(nchunks, kMaxGrid), kBlock, 0,
+                                    s>>>(view, q.data(), qn, foff.data(), op);
+      }
+      if (multi) {
+        uint64_t nrecv = halo_flush<uint32_t>(I, comm_, rank_, world_,
+                                              depth.data(), cap, s);
+        if (nrecv)
+          halo_process_kernel<uint32_t, BfsRecvOp>
+              <<<grid_for(nrecv), kBlock, 0, s>>>(
+                  reinterpret_cast<HaloPair<uint32_t>*>(I.recvbuf.data()),
+                  nrecv,
+                  BfsRecvOp{depth.data(), DevBitmap{next_bm.data()},
+                            g.v_begin});
+      }
+    }
+    qn = static_cast<uint32_t>(compact_frontier(I, next_bm.data(), owned,
+                                                g.v_begin, q.data(), s));
+    global_curr = multi ? comm_->allreduce_sum(qn) : qn;
+    if (getenv("GRAPEHIP_DEBUG"))
+      fprintf(stderr, "[bfs] level=%u mode=%s fedges=%lu next=%u t=%.1fms\n",
+              level, use_pull ? "pull" : "push", (unsigned long)fedges, qn,
+              (wall_s() - t0) * 1e3);
     ++level;
     ++rounds;
-    cur = 1 - cur;
-    if (next_global == 0) break;
   }
   HIP_CHECK(hipDeviceSynchronize());
   if (comm_) comm_->barrier();
@@ -1226,20 +1474,20 @@ GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
   bool multi = world_ > 1;
 
   DeviceBuffer<float> dist(nv_pad);
-  DeviceBuffer<uint32_t> qn0(owned + 64), qn1(owned + 64);  // near dbl-buf
+  DeviceBuffer<uint32_t> q0(owned + 64), q1(owned + 64);   // near dbl-buf
   DeviceBuffer<uint32_t> qf0(owned + 64), qf1(owned + 64);  // far dbl-buf
-  DeviceBuffer<unsigned long long> qcnt(4);  // near0, near1, far0, far1
-  uint64_t bm_words = (static_cast<uint64_t>(nv_pad) + 31) / 32;
-  DeviceBuffer<uint32_t> near_bm0(bm_words), near_bm1(bm_words),
-      far_bm(bm_words);
+  DeviceBuffer<unsigned long long> fcnt(2);  // far queue counters
+  size_t own_words = (static_cast<size_t>(owned) + 31) / 32 + 1;
+  uint64_t glob_words = (static_cast<uint64_t>(nv_pad) + 31) / 32;
+  DeviceBuffer<uint32_t> near_bm(own_words), far_bm(glob_words);
+  DeviceBuffer<uint32_t> fdeg;
+  DeviceBuffer<uint64_t> foff;
   if (multi) {
     I.halo_idx.resize(static_cast<uint64_t>(world_) * cap);
     I.halo_cnt.resize(world_);
-    I.halo_bm.resize(bm_words);
+    I.halo_bm.resize(glob_words);
   }
   if (delta <= 0) {
-    // heuristic: 32 * avg_weight / avg_degree (Davidson et al., as in the
-    // reference cuda/sssp/sssp.h:80-92); avg_weight ~ 50 for [1,100)
     double avg_deg = static_cast<double>(g.total_edges) /
                      std::max<uint64_t>(1, g.nv_global);
     delta = static_cast<float>(32.0 * 50.0 / std::max(1.0, avg_deg));
@@ -1250,9 +1498,8 @@ GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
   double t0 = wall_s();
 
   fill(dist.data(), std::numeric_limits<float>::max(), nv_pad, s);
-  qcnt.zero(s);
-  near_bm0.zero(s);
-  near_bm1.zero(s);
+  fcnt.zero(s);
+  near_bm.zero(s);
   far_bm.zero(s);
   if (multi) {
     I.halo_cnt.zero(s);
@@ -1260,34 +1507,36 @@ GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
   }
   uint32_t src = static_cast<uint32_t>(source);
   if (src >= g.v_begin && src < g.v_end)
-    sssp_seed_kernel<<<1, 1, 0, s>>>(dist.data(), src, qn0.data(),
-                                     qcnt.data(), near_bm0.data());
-  uint32_t* nearq[2] = {qn0.data(), qn1.data()};
+    sssp_seed_kernel<<<1, 1, 0, s>>>(dist.data(), src, g.v_begin,
+                                     near_bm.data());
+  uint32_t* nearq[2] = {q0.data(), q1.data()};
   uint32_t* farq[2] = {qf0.data(), qf1.data()};
-  uint32_t* nearbm[2] = {near_bm0.data(), near_bm1.data()};
   int ncur = 0, fcur = 0;
   float prio = 0.0f;
   int rounds = 0;
-  std::vector<unsigned long long> hc(4);
   unsigned long long zero = 0;
+  std::vector<unsigned long long> hfc(2);
+  uint32_t qn = static_cast<uint32_t>(compact_frontier(
+      I, near_bm.data(), owned, g.v_begin, nearq[ncur], s));
+  uint64_t g_near = multi ? comm_->allreduce_sum(qn) : qn;
   for (;;) {
-    HIP_CHECK(hipMemcpyAsync(hc.data(), qcnt.data(), 32,
-                             hipMemcpyDeviceToHost, s));
-    HIP_CHECK(hipStreamSynchronize(s));
-    uint32_t qn = static_cast<uint32_t>(hc[ncur]);
     float prio_hi = prio + delta;
-    SsspOp op{dist.data(), prio_hi,
-              DevQueue{nearq[1 - ncur], qcnt.data() + (1 - ncur)},
-              DevQueue{farq[fcur], qcnt.data() + 2 + fcur},
-              DevBitmap{nearbm[1 - ncur]}, DevBitmap{far_bm.data()},
-              g.v_begin, g.v_end, multi,
-              DevHalo{I.halo_idx.data(), I.halo_cnt.data(),
-                      DevBitmap{I.halo_bm.data()}, cap, view.slice, world_}};
     if (qn) {
-      expand_frontier(I, view, nearq[ncur], qn, op, s);
-      // clear the consumed queue's bitmap so future rounds can re-enqueue
-      clear_bits_kernel<<<grid_for(qn), kBlock, 0, s>>>(
-          nearq[ncur], qn, DevBitmap{nearbm[ncur]});
+      if (fdeg.size() < qn) fdeg.resize(qn + (qn >> 2) + 64);
+      if (foff.size() < qn + 1) foff.resize(qn + (qn >> 2) + 65);
+      gather_deg_kernel<<<grid_for(qn), kBlock, 0, s>>>(view, nearq[ncur], qn,
+                                                        fdeg.data());
+      exclusive_scan(fdeg.data(), foff.data(), qn, s, I.scan);
+      SsspOp op{dist.data(), prio_hi, DevBitmap{near_bm.data()},
+                DevQueue{farq[fcur], fcnt.data() + fcur},
+                DevBitmap{far_bm.data()}, g.v_begin, g.v_end, multi,
+                DevHalo{I.halo_idx.data(), I.halo_cnt.data(),
+                        DevBitmap{I.halo_bm.data()}, cap, view.slice,
+                        world_}};
+      int nchunks = static_cast<int>((qn + kBlock - 1) / kBlock);
+      expand_cm_frontier<SsspOp><<<std::min(nchunks, kMaxGrid), kBlock, 0,
+                                   s>>>(view, nearq[ncur], qn, foff.data(),
+                                        op);
     }
     if (multi) {
       uint64_t nrecv =
@@ -1296,48 +1545,47 @@ GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
         halo_process_kernel<float, SsspRecvOp>
             <<<grid_for(nrecv), kBlock, 0, s>>>(
                 reinterpret_cast<HaloPair<float>*>(I.recvbuf.data()), nrecv,
-                SsspRecvOp{dist.data(), prio_hi,
-                           DevQueue{nearq[1 - ncur], qcnt.data() + (1 - ncur)},
-                           DevQueue{farq[fcur], qcnt.data() + 2 + fcur},
-                           DevBitmap{nearbm[1 - ncur]},
-                           DevBitmap{far_bm.data()}});
+                SsspRecvOp{dist.data(), prio_hi, DevBitmap{near_bm.data()},
+                           DevQueue{farq[fcur], fcnt.data() + fcur},
+                           DevBitmap{far_bm.data()}, g.v_begin});
     }
-    HIP_CHECK(hipMemcpyAsync(qcnt.data() + ncur, &zero, 8,
-                             hipMemcpyHostToDevice, s));
-    HIP_CHECK(hipMemcpyAsync(hc.data(), qcnt.data(), 32,
+    ++rounds;
+    uint32_t next_qn = static_cast<uint32_t>(compact_frontier(
+        I, near_bm.data(), owned, g.v_begin, nearq[1 - ncur], s));
+    g_near = multi ? comm_->allreduce_sum(next_qn) : next_qn;
+    if (getenv("GRAPEHIP_DEBUG"))
+      fprintf(stderr, "[sssp] round=%d prio=%.0f in=%u next=%u t=%.1fms\n",
+              rounds, prio, qn, next_qn, (wall_s() - t0) * 1e3);
+    ncur = 1 - ncur;
+    qn = next_qn;
+    if (g_near > 0) continue;
+    // near exhausted globally: advance priority, repartition far buckets
+    HIP_CHECK(hipMemcpyAsync(hfc.data(), fcnt.data(), 16,
                              hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
-    ++rounds;
-    uint64_t next_near = hc[1 - ncur];
-    uint64_t far_n = hc[2 + fcur];
-    uint64_t g_near = multi ? comm_->allreduce_sum(next_near) : next_near;
-    ncur = 1 - ncur;
-    if (g_near > 0) continue;
-    // near exhausted globally: advance priority, repartition far
+    uint64_t far_n = hfc[fcur];
     uint64_t g_far = multi ? comm_->allreduce_sum(far_n) : far_n;
     if (g_far == 0) break;
-    prio += delta;
     while (g_near == 0 && g_far > 0) {
+      prio += delta;
       float new_hi = prio + delta;
-      if (far_n) {
+      if (far_n)
         sssp_repart_kernel<<<grid_for(far_n), kBlock, 0, s>>>(
             farq[fcur], static_cast<uint32_t>(far_n), dist.data(), new_hi,
-            DevQueue{nearq[ncur], qcnt.data() + ncur},
-            DevQueue{farq[1 - fcur], qcnt.data() + 2 + (1 - fcur)},
-            DevBitmap{nearbm[ncur]}, DevBitmap{far_bm.data()});
-      }
-      HIP_CHECK(hipMemcpyAsync(qcnt.data() + 2 + fcur, &zero, 8,
+            DevBitmap{near_bm.data()}, g.v_begin,
+            DevQueue{farq[1 - fcur], fcnt.data() + (1 - fcur)},
+            DevBitmap{far_bm.data()});
+      HIP_CHECK(hipMemcpyAsync(fcnt.data() + fcur, &zero, 8,
                                hipMemcpyHostToDevice, s));
-      HIP_CHECK(hipMemcpyAsync(hc.data(), qcnt.data(), 32,
+      qn = static_cast<uint32_t>(compact_frontier(
+          I, near_bm.data(), owned, g.v_begin, nearq[ncur], s));
+      g_near = multi ? comm_->allreduce_sum(qn) : qn;
+      HIP_CHECK(hipMemcpyAsync(hfc.data(), fcnt.data(), 16,
                                hipMemcpyDeviceToHost, s));
       HIP_CHECK(hipStreamSynchronize(s));
       fcur = 1 - fcur;
-      far_n = hc[2 + fcur];
-      uint64_t near_n = hc[ncur];
-      g_near = multi ? comm_->allreduce_sum(near_n) : near_n;
+      far_n = hfc[fcur];
       g_far = multi ? comm_->allreduce_sum(far_n) : far_n;
-      if (g_near == 0 && g_far > 0) prio += delta;
-      if (g_near == 0 && g_far == 0) break;
     }
     if (g_near == 0 && g_far == 0) break;
   }
@@ -1361,10 +1609,6 @@ GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
   return res;
 }
 
-}  // namespace grapehip
-
-namespace grapehip {
-
 // ---------------------------------------------------------------------------
 // PageRank (push + fp64 hw atomics; reduce-scatter/allgather over xGMI)
 // ---------------------------------------------------------------------------
@@ -1386,22 +1630,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
       (!g.directed ? g.oe_off.data() : g.ie_off.data());
   const uint32_t* pull_dst =
       (!g.directed ? g.oe_dst.data() : g.ie_dst.data());
-  if (pull && !g.buckets_built) {
-    g.rows_small.resize(owned);
-    g.rows_mid.resize(owned);
-    g.rows_large.resize(owned);
-    DeviceBuffer<unsigned long long> cnts(3);
-    cnts.zero(s);
-    bucket_rows_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-        pull_off, owned, g.rows_small.data(), cnts.data() + 0,
-        g.rows_mid.data(), cnts.data() + 1, g.rows_large.data(),
-        cnts.data() + 2);
-    auto h = cnts.download(s);
-    g.n_small = h[0];
-    g.n_mid = h[1];
-    g.n_large = h[2];
-    g.buckets_built = true;
-  }
+  if (pull) ensure_buckets(g, s);
 
   DeviceBuffer<double> rank_arr(nv_pad), acc(nv_pad), contrib(nv_pad);
   DeviceBuffer<double> d_dangling(1);
