@@ -188,18 +188,22 @@ py::dict run_timed(PyEngine& eng, RunFn&& run) {
 }
 
 #ifdef GRAPEHIP_WITH_HIP
-py::dict gpu_dict(const GpuRunResult& r, const DeviceGraph& g, bool is_i64) {
+py::dict gpu_dict(const GpuRunResult& r, const DeviceGraph& g, bool is_i64,
+                  bool with_values) {
   py::dict out;
   out["rounds"] = r.rounds;
   out["seconds"] = r.seconds;
   out["traversed_edges"] = r.traversed_edges;
-  uint32_t owned = g.owned();
-  py::array_t<int64_t> oids(owned);
-  auto* p = oids.mutable_data();
-  for (uint32_t i = 0; i < owned; ++i)
-    p[i] = static_cast<int64_t>(g.v_begin) + i;
-  out["oids"] = oids;
-  out["values"] = is_i64 ? py::object(to_np(r.i64)) : py::object(to_np(r.f64));
+  if (with_values) {
+    uint32_t owned = g.owned();
+    py::array_t<int64_t> oids(owned);
+    auto* p = oids.mutable_data();
+    for (uint32_t i = 0; i < owned; ++i)
+      p[i] = static_cast<int64_t>(g.v_begin) + i;
+    out["oids"] = oids;
+    out["values"] =
+        is_i64 ? py::object(to_np(r.i64)) : py::object(to_np(r.f64));
+  }
   return out;
 }
 #endif
@@ -286,15 +290,15 @@ PYBIND11_MODULE(_core, m) {
            py::arg("directed") = false, py::arg("weighted") = false,
            py::arg("a") = 0.57, py::arg("b") = 0.19, py::arg("c") = 0.19)
       .def("bfs",
-           [](PyEngine& eng, PyGraph& g, int64_t source) {
+           [](PyEngine& eng, PyGraph& g, int64_t source, bool values) {
 #ifdef GRAPEHIP_WITH_HIP
              if (eng.use_gpu) {
                GpuRunResult r;
                {
                  py::gil_scoped_release rel;
-                 r = eng.gpu->bfs(*g.dev, source);
+                 r = eng.gpu->bfs(*g.dev, source, values);
                }
-               return gpu_dict(r, *g.dev, true);
+               return gpu_dict(r, *g.dev, true, values);
              }
 #endif
              BFSApp app;
@@ -311,17 +315,19 @@ PYBIND11_MODULE(_core, m) {
              meta["values"] = to_np(vals);
              return meta;
            },
-           py::arg("graph"), py::arg("source") = 0)
+           py::arg("graph"), py::arg("source") = 0,
+           py::arg("values") = true)
       .def("sssp",
-           [](PyEngine& eng, PyGraph& g, int64_t source, float delta) {
+           [](PyEngine& eng, PyGraph& g, int64_t source, float delta,
+              bool values) {
 #ifdef GRAPEHIP_WITH_HIP
              if (eng.use_gpu) {
                GpuRunResult r;
                {
                  py::gil_scoped_release rel;
-                 r = eng.gpu->sssp(*g.dev, source, delta);
+                 r = eng.gpu->sssp(*g.dev, source, delta, values);
                }
-               return gpu_dict(r, *g.dev, false);
+               return gpu_dict(r, *g.dev, false, values);
              }
 #endif
              SSSPApp app;
@@ -338,17 +344,19 @@ PYBIND11_MODULE(_core, m) {
              meta["values"] = to_np(vals);
              return meta;
            },
-           py::arg("graph"), py::arg("source") = 0, py::arg("delta") = -1.0f)
+           py::arg("graph"), py::arg("source") = 0, py::arg("delta") = -1.0f,
+           py::arg("values") = true)
       .def("pagerank",
-           [](PyEngine& eng, PyGraph& g, double damping, int iters) {
+           [](PyEngine& eng, PyGraph& g, double damping, int iters,
+              bool values) {
 #ifdef GRAPEHIP_WITH_HIP
              if (eng.use_gpu) {
                GpuRunResult r;
                {
                  py::gil_scoped_release rel;
-                 r = eng.gpu->pagerank(*g.dev, damping, iters);
+                 r = eng.gpu->pagerank(*g.dev, damping, iters, values);
                }
-               return gpu_dict(r, *g.dev, false);
+               return gpu_dict(r, *g.dev, false, values);
              }
 #endif
              PageRankApp app;
@@ -362,17 +370,18 @@ PYBIND11_MODULE(_core, m) {
              meta["values"] = to_np(ctx.rank);
              return meta;
            },
-           py::arg("graph"), py::arg("damping") = 0.85, py::arg("iters") = 10)
+           py::arg("graph"), py::arg("damping") = 0.85, py::arg("iters") = 10,
+           py::arg("values") = true)
       .def("wcc",
-           [](PyEngine& eng, PyGraph& g) {
+           [](PyEngine& eng, PyGraph& g, bool values) {
 #ifdef GRAPEHIP_WITH_HIP
              if (eng.use_gpu) {
                GpuRunResult r;
                {
                  py::gil_scoped_release rel;
-                 r = eng.gpu->wcc(*g.dev);
+                 r = eng.gpu->wcc(*g.dev, values);
                }
-               return gpu_dict(r, *g.dev, true);
+               return gpu_dict(r, *g.dev, true, values);
              }
 #endif
              WCCApp app;
@@ -389,17 +398,17 @@ PYBIND11_MODULE(_core, m) {
              meta["values"] = to_np(vals);
              return meta;
            },
-           py::arg("graph"))
+           py::arg("graph"), py::arg("values") = true)
       .def("cdlp",
-           [](PyEngine& eng, PyGraph& g, int iters) {
+           [](PyEngine& eng, PyGraph& g, int iters, bool values) {
 #ifdef GRAPEHIP_WITH_HIP
              if (eng.use_gpu) {
                GpuRunResult r;
                {
                  py::gil_scoped_release rel;
-                 r = eng.gpu->cdlp(*g.dev, iters);
+                 r = eng.gpu->cdlp(*g.dev, iters, values);
                }
-               return gpu_dict(r, *g.dev, true);
+               return gpu_dict(r, *g.dev, true, values);
              }
 #endif
              CDLPApp app;
@@ -415,17 +424,18 @@ PYBIND11_MODULE(_core, m) {
              meta["values"] = to_np(vals);
              return meta;
            },
-           py::arg("graph"), py::arg("iters") = 10)
+           py::arg("graph"), py::arg("iters") = 10,
+           py::arg("values") = true)
       .def("lcc",
-           [](PyEngine& eng, PyGraph& g) {
+           [](PyEngine& eng, PyGraph& g, bool values) {
 #ifdef GRAPEHIP_WITH_HIP
              if (eng.use_gpu) {
                GpuRunResult r;
                {
                  py::gil_scoped_release rel;
-                 r = eng.gpu->lcc(*g.dev);
+                 r = eng.gpu->lcc(*g.dev, values);
                }
-               return gpu_dict(r, *g.dev, false);
+               return gpu_dict(r, *g.dev, false, values);
              }
 #endif
              LCCApp app;
@@ -439,7 +449,7 @@ PYBIND11_MODULE(_core, m) {
              meta["values"] = to_np(ctx.lcc);
              return meta;
            },
-           py::arg("graph"));
+           py::arg("graph"), py::arg("values") = true);
 
   m.attr("WITH_HIP") =
 #ifdef GRAPEHIP_WITH_HIP

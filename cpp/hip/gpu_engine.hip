@@ -1069,8 +1069,9 @@ void build_csr_from_coo(const DeviceBuffer<uint32_t>& src,
                                                     deg.data());
   out_off.resize(owned + 1);
   uint64_t total = exclusive_scan(deg.data(), out_off.data(), owned, s, scan);
-  fprintf(stderr, "[csr] scan total=%lu n=%lu\n", (unsigned long)total,
-          (unsigned long)n);
+  if (getenv("GRAPEHIP_DEBUG"))
+    fprintf(stderr, "[csr] scan total=%lu n=%lu\n", (unsigned long)total,
+            (unsigned long)n);
   if (total != n) throw std::runtime_error("CSR build: degree sum mismatch");
   deg.free();
   DeviceBuffer<unsigned long long> cursor(owned);
@@ -1122,8 +1123,9 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
   DeviceBuffer<float> e_w(weighted ? est : 0);
   DeviceBuffer<unsigned long long> cnt(1);
   cnt.zero(s);
-  fprintf(stderr, "[gen] nv=%u ne=%lu est=%lu scale=%d\n", g->nv_global,
-          (unsigned long)ne, (unsigned long)est, scale);
+  if (getenv("GRAPEHIP_DEBUG"))
+    fprintf(stderr, "[gen] nv=%u ne=%lu est=%lu scale=%d\n", g->nv_global,
+            (unsigned long)ne, (unsigned long)est, scale);
   gen_edges_kernel<<<kMaxGrid, kBlock, 0, s>>>(
       ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, g->v_begin, g->v_end,
       !directed, weighted, e_src.data(), e_dst.data(),
@@ -1132,7 +1134,8 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
   HIP_CHECK(hipMemcpyAsync(&n_local, cnt.data(), 8, hipMemcpyDeviceToHost, s));
   HIP_CHECK(hipStreamSynchronize(s));
   if (n_local > est) throw std::runtime_error("gen_synthetic: overflow");
-  fprintf(stderr, "[gen] n_local=%llu\n", n_local);
+  if (getenv("GRAPEHIP_DEBUG"))
+    fprintf(stderr, "[gen] n_local=%llu\n", n_local);
 
   build_csr_from_coo(e_src, e_dst, e_w, n_local, g->v_begin, owned, weighted,
                      g->oe_off, g->oe_dst, g->oe_w, s, impl_->scan);
@@ -1323,7 +1326,8 @@ void expand_frontier(GpuContext::Impl& I, const DevGraphView& view,
 // ---------------------------------------------------------------------------
 // BFS
 // ---------------------------------------------------------------------------
-GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source) {
+GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source,
+                             bool fetch) {
   auto& I = *impl_;
   hipStream_t s = I.compute;
   DevGraphView view = make_view(g, rank_, world_);
@@ -1448,15 +1452,17 @@ GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source) {
   res.rounds = rounds;
   res.seconds = comm_ ? comm_->allreduce_max_double(t1 - t0) : (t1 - t0);
   res.traversed_edges = g.input_edges;
-  std::vector<uint32_t> d32(owned);
-  HIP_CHECK(hipMemcpyAsync(d32.data(), depth.data() + g.v_begin, owned * 4,
-                           hipMemcpyDeviceToHost, s));
-  HIP_CHECK(hipStreamSynchronize(s));
-  res.i64.resize(owned);
-  for (uint32_t i = 0; i < owned; ++i)
-    res.i64[i] = d32[i] == 0xFFFFFFFFu
-                     ? std::numeric_limits<int64_t>::max()
-                     : static_cast<int64_t>(d32[i]);
+  if (fetch) {
+    std::vector<uint32_t> d32(owned);
+    HIP_CHECK(hipMemcpyAsync(d32.data(), depth.data() + g.v_begin, owned * 4,
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    res.i64.resize(owned);
+    for (uint32_t i = 0; i < owned; ++i)
+      res.i64[i] = d32[i] == 0xFFFFFFFFu
+                       ? std::numeric_limits<int64_t>::max()
+                       : static_cast<int64_t>(d32[i]);
+  }
   return res;
 }
 
@@ -1464,7 +1470,7 @@ GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source) {
 // SSSP (near-far delta stepping; cuda/sssp/sssp.h parity)
 // ---------------------------------------------------------------------------
 GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
-                              float delta) {
+                              float delta, bool fetch) {
   auto& I = *impl_;
   hipStream_t s = I.compute;
   DevGraphView view = make_view(g, rank_, world_);
@@ -1597,15 +1603,17 @@ GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
   res.rounds = rounds;
   res.seconds = comm_ ? comm_->allreduce_max_double(t1 - t0) : (t1 - t0);
   res.traversed_edges = g.input_edges;
-  std::vector<float> d32(owned);
-  HIP_CHECK(hipMemcpyAsync(d32.data(), dist.data() + g.v_begin, owned * 4,
-                           hipMemcpyDeviceToHost, s));
-  HIP_CHECK(hipStreamSynchronize(s));
-  res.f64.resize(owned);
-  for (uint32_t i = 0; i < owned; ++i)
-    res.f64[i] = d32[i] >= std::numeric_limits<float>::max()
-                     ? std::numeric_limits<double>::max()
-                     : static_cast<double>(d32[i]);
+  if (fetch) {
+    std::vector<float> d32(owned);
+    HIP_CHECK(hipMemcpyAsync(d32.data(), dist.data() + g.v_begin, owned * 4,
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    res.f64.resize(owned);
+    for (uint32_t i = 0; i < owned; ++i)
+      res.f64[i] = d32[i] >= std::numeric_limits<float>::max()
+                       ? std::numeric_limits<double>::max()
+                       : static_cast<double>(d32[i]);
+  }
   return res;
 }
 
@@ -1613,7 +1621,7 @@ GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
 // PageRank (push + fp64 hw atomics; reduce-scatter/allgather over xGMI)
 // ---------------------------------------------------------------------------
 GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
-                                  int iters) {
+                                  int iters, bool fetch) {
   auto& I = *impl_;
   hipStream_t s = I.compute;
   uint32_t nv_pad = padded_nv(g, world_);
@@ -1710,17 +1718,19 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
   res.rounds = rounds;
   res.seconds = comm_ ? comm_->allreduce_max_double(t1 - t0) : (t1 - t0);
   res.traversed_edges = static_cast<uint64_t>(iters) * g.total_edges;
-  res.f64.resize(owned);
-  HIP_CHECK(hipMemcpyAsync(res.f64.data(), rank_arr.data() + g.v_begin,
-                           owned * 8, hipMemcpyDeviceToHost, s));
-  HIP_CHECK(hipStreamSynchronize(s));
+  if (fetch) {
+    res.f64.resize(owned);
+    HIP_CHECK(hipMemcpyAsync(res.f64.data(), rank_arr.data() + g.v_begin,
+                             owned * 8, hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+  }
   return res;
 }
 
 // ---------------------------------------------------------------------------
 // WCC (min-root union-find, replicated parent + allreduce-min merge)
 // ---------------------------------------------------------------------------
-GpuRunResult GpuContext::wcc(DeviceGraph& g) {
+GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
   auto& I = *impl_;
   hipStream_t s = I.compute;
   DevGraphView view = make_view(g, rank_, world_);
@@ -1775,18 +1785,20 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g) {
   res.rounds = rounds;
   res.seconds = comm_ ? comm_->allreduce_max_double(t1 - t0) : (t1 - t0);
   res.traversed_edges = g.input_edges;
-  std::vector<uint32_t> lab(owned);
-  HIP_CHECK(hipMemcpyAsync(lab.data(), parent.data() + g.v_begin, owned * 4,
-                           hipMemcpyDeviceToHost, s));
-  HIP_CHECK(hipStreamSynchronize(s));
-  res.i64.assign(lab.begin(), lab.end());
+  if (fetch) {
+    std::vector<uint32_t> lab(owned);
+    HIP_CHECK(hipMemcpyAsync(lab.data(), parent.data() + g.v_begin,
+                             owned * 4, hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    res.i64.assign(lab.begin(), lab.end());
+  }
   return res;
 }
 
-GpuRunResult GpuContext::cdlp(DeviceGraph&, int) {
+GpuRunResult GpuContext::cdlp(DeviceGraph&, int, bool) {
   throw std::runtime_error("GPU CDLP: not implemented yet (use CPU engine)");
 }
-GpuRunResult GpuContext::lcc(DeviceGraph&) {
+GpuRunResult GpuContext::lcc(DeviceGraph&, bool) {
   throw std::runtime_error("GPU LCC: not implemented yet (use CPU engine)");
 }
 

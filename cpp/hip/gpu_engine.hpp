@@ -64,12 +64,16 @@ class GpuContext {
                                              double rmat_a, double rmat_b,
                                              double rmat_c);
 
-  GpuRunResult bfs(DeviceGraph& g, int64_t source);
-  GpuRunResult sssp(DeviceGraph& g, int64_t source, float delta);
-  GpuRunResult pagerank(DeviceGraph& g, double damping, int iters);
-  GpuRunResult wcc(DeviceGraph& g);
-  GpuRunResult cdlp(DeviceGraph& g, int iters);
-  GpuRunResult lcc(DeviceGraph& g);
+  // fetch=false skips the result D2H/convert (bench times algorithm only,
+  // like the reference's "run algorithm" timer phase vs the Output phase).
+  GpuRunResult bfs(DeviceGraph& g, int64_t source, bool fetch = true);
+  GpuRunResult sssp(DeviceGraph& g, int64_t source, float delta,
+                    bool fetch = true);
+  GpuRunResult pagerank(DeviceGraph& g, double damping, int iters,
+                        bool fetch = true);
+  GpuRunResult wcc(DeviceGraph& g, bool fetch = true);
+  GpuRunResult cdlp(DeviceGraph& g, int iters, bool fetch = true);
+  GpuRunResult lcc(DeviceGraph& g, bool fetch = true);
 
   void device_sync();
   // test hook: exclusive scan of host u32 data on the device
