@@ -4,8 +4,8 @@ import numpy as np
 import pytest
 
 import grapehip
-from oracles import (bfs_oracle, pagerank_oracle, sssp_oracle, wcc_oracle,
-                     INT64_MAX)
+from oracles import (bfs_oracle, cdlp_oracle, lcc_oracle, pagerank_oracle,
+                     sssp_oracle, wcc_oracle, INT64_MAX)
 
 pytestmark = pytest.mark.gpu
 
@@ -103,3 +103,69 @@ def test_gpu_synthetic(eng):
     labs = r4["values"]
     # all BFS-reachable vertices share vertex 0's component label
     assert (labs[reached] == labs[0]).all()
+
+
+def test_gpu_cdlp_undirected(eng):
+    src, dst, _ = random_graph(num_v=4000, num_e=30000, seed=23)
+    g = eng.load_edges(src, dst, directed=False, num_vertices=4000)
+    _, vals = by_oid(eng.cdlp(g, 5))
+    assert np.array_equal(vals, cdlp_oracle(4000, src, dst, 5, directed=False))
+
+
+def test_gpu_cdlp_directed(eng):
+    src, dst, _ = random_graph(num_v=2000, num_e=15000, seed=29)
+    g = eng.load_edges(src, dst, directed=True, num_vertices=2000,
+                       build_in_csr=True)
+    _, vals = by_oid(eng.cdlp(g, 6))
+    assert np.array_equal(vals, cdlp_oracle(2000, src, dst, 6, directed=True))
+
+
+def test_gpu_cdlp_hub_tiers(eng):
+    # star hubs exercise the LDS-hash (mid) and global-hash (large) tiers
+    rng = np.random.default_rng(31)
+    nv = 30000
+    hub_e = np.stack([np.zeros(12000, np.int64),
+                      rng.integers(1, nv, 12000)], 1)
+    mid_e = np.stack([np.ones(2000, np.int64),
+                      rng.integers(2, nv, 2000)], 1)
+    rest = np.stack([rng.integers(0, nv, 40000),
+                     rng.integers(0, nv, 40000)], 1)
+    e = np.concatenate([hub_e, mid_e, rest])
+    keep = e[:, 0] != e[:, 1]
+    src, dst = e[keep, 0], e[keep, 1]
+    g = eng.load_edges(src, dst, directed=False, num_vertices=nv)
+    _, vals = by_oid(eng.cdlp(g, 4))
+    assert np.array_equal(vals, cdlp_oracle(nv, src, dst, 4, directed=False))
+
+
+def test_gpu_lcc_undirected(eng):
+    src, dst, _ = random_graph(num_v=2000, num_e=40000, seed=37)
+    g = eng.load_edges(src, dst, directed=False, num_vertices=2000)
+    _, vals = by_oid(eng.lcc(g))
+    expect = lcc_oracle(2000, src, dst, directed=False)
+    assert np.allclose(vals, expect, rtol=1e-12)
+
+
+def test_gpu_lcc_directed_raises(eng):
+    # directed LCC runs on the CPU engine; the GPU path must refuse loudly
+    src, dst, _ = random_graph(num_v=200, num_e=1000, seed=41)
+    g = eng.load_edges(src, dst, directed=True, num_vertices=200,
+                       build_in_csr=True)
+    with pytest.raises(RuntimeError, match="undirected"):
+        eng.lcc(g)
+
+
+def test_gpu_lcc_hub(eng):
+    # hub row exercises the global-hash tier + big-row bitonic sort
+    rng = np.random.default_rng(43)
+    nv = 20000
+    hub = np.stack([np.zeros(9000, np.int64), rng.integers(1, nv, 9000)], 1)
+    rest = np.stack([rng.integers(0, nv, 60000),
+                     rng.integers(0, nv, 60000)], 1)
+    e = np.concatenate([hub, rest])
+    keep = e[:, 0] != e[:, 1]
+    src, dst = e[keep, 0], e[keep, 1]
+    g = eng.load_edges(src, dst, directed=False, num_vertices=nv)
+    _, vals = by_oid(eng.lcc(g))
+    expect = lcc_oracle(nv, src, dst, directed=False)
+    assert np.allclose(vals, expect, rtol=1e-12)
