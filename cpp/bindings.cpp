@@ -12,8 +12,11 @@
 #include <memory>
 #include <optional>
 
+#include "apps/bc.hpp"
 #include "apps/bfs.hpp"
 #include "apps/cdlp.hpp"
+#include "apps/kclique.hpp"
+#include "apps/kcore.hpp"
 #include "apps/lcc.hpp"
 #include "apps/pagerank.hpp"
 #include "apps/sssp.hpp"
@@ -449,7 +452,90 @@ PYBIND11_MODULE(_core, m) {
              meta["values"] = to_np(ctx.lcc);
              return meta;
            },
-           py::arg("graph"), py::arg("values") = true);
+           py::arg("graph"), py::arg("values") = true)
+      .def("bc",
+           [](PyEngine& eng, PyGraph& g, int64_t source) {
+             if (!g.frag)
+               throw std::runtime_error(
+                   "bc runs on the CPU engine and needs a host fragment "
+                   "(load with load_edges)");
+             BCApp app;
+             BCContext ctx;
+             MessageManager mm;
+             mm.init(eng.c(), g.frag.get(), eng.n_threads);
+             ctx.init(*g.frag, source);
+             py::dict meta = run_timed(
+                 eng, [&] { return RunWorker(app, ctx, *g.frag, mm); });
+             std::vector<double> vals(g.frag->ivnum());
+             std::vector<double> sig(g.frag->ivnum());
+             std::vector<int64_t> dep(g.frag->ivnum());
+             for (vid_t v = 0; v < g.frag->ivnum(); ++v) {
+               vals[v] = ctx.delta[v].load(std::memory_order_relaxed);
+               sig[v] = ctx.sigma[v].load(std::memory_order_relaxed);
+               dep[v] = ctx.depth[v].load(std::memory_order_relaxed);
+             }
+             meta["oids"] = inner_oids(*g.frag);
+             meta["values"] = to_np(vals);
+             meta["path_num"] = to_np(sig);
+             meta["depth"] = to_np(dep);
+             return meta;
+           },
+           py::arg("graph"), py::arg("source") = 0)
+      .def("kcore",
+           [](PyEngine& eng, PyGraph& g, int k) {
+             if (!g.frag)
+               throw std::runtime_error("kcore needs a host fragment");
+             KCoreApp app;
+             KCoreContext ctx;
+             MessageManager mm;
+             mm.init(eng.c(), g.frag.get(), eng.n_threads);
+             ctx.init(*g.frag, k);
+             py::dict meta = run_timed(
+                 eng, [&] { return RunWorker(app, ctx, *g.frag, mm); });
+             std::vector<int64_t> vals(g.frag->ivnum());
+             for (vid_t v = 0; v < g.frag->ivnum(); ++v)
+               vals[v] = ctx.removed[v] ? 0 : 1;
+             meta["oids"] = inner_oids(*g.frag);
+             meta["values"] = to_np(vals);
+             return meta;
+           },
+           py::arg("graph"), py::arg("k") = 3)
+      .def("core_decomposition",
+           [](PyEngine& eng, PyGraph& g) {
+             if (!g.frag)
+               throw std::runtime_error(
+                   "core_decomposition needs a host fragment");
+             CoreDecompApp app;
+             CoreDecompContext ctx;
+             MessageManager mm;
+             mm.init(eng.c(), g.frag.get(), eng.n_threads);
+             ctx.init(*g.frag);
+             py::dict meta = run_timed(
+                 eng, [&] { return RunWorker(app, ctx, *g.frag, mm); });
+             std::vector<int64_t> vals(g.frag->ivnum());
+             for (vid_t v = 0; v < g.frag->ivnum(); ++v)
+               vals[v] = ctx.est[v].load(std::memory_order_relaxed);
+             meta["oids"] = inner_oids(*g.frag);
+             meta["values"] = to_np(vals);
+             return meta;
+           },
+           py::arg("graph"))
+      .def("kclique",
+           [](PyEngine& eng, PyGraph& g, int k) {
+             if (!g.frag)
+               throw std::runtime_error("kclique needs a host fragment");
+             if (k < 2) throw std::runtime_error("kclique: k must be >= 2");
+             KCliqueApp app;
+             KCliqueContext ctx;
+             MessageManager mm;
+             mm.init(eng.c(), g.frag.get(), eng.n_threads);
+             ctx.init(*g.frag, k);
+             py::dict meta = run_timed(
+                 eng, [&] { return RunWorker(app, ctx, *g.frag, mm); });
+             meta["clique_count"] = ctx.clique_num;
+             return meta;
+           },
+           py::arg("graph"), py::arg("k") = 3);
 
   m.attr("WITH_HIP") =
 #ifdef GRAPEHIP_WITH_HIP

@@ -125,6 +125,29 @@ class MessageManager {
     }
   }
 
+  // Like process(), but the payload may target a NON-owned vertex of the
+  // receiving fragment (owner pushing its inner state to mirrors, e.g.
+  // core decomposition estimates / BC depth refresh). Decodes through the
+  // fragment's full gid->lid map; drops gids this fragment doesn't hold.
+  template <typename MSG, typename F>
+  void process_any(F&& f) {
+    constexpr size_t unit = sizeof(vid_t) + sizeof(MSG);
+    for (int src = 0; src < fnum_; ++src) {
+      OutArchive& ar = recv_[src];
+      size_t n = ar.remaining() / unit;
+      const char* base = ar.cursor();
+      parallel_for_tid(0, n, [&](int tid, size_t i) {
+        vid_t gid;
+        MSG m;
+        std::memcpy(&gid, base + i * unit, sizeof(vid_t));
+        std::memcpy(&m, base + i * unit + sizeof(vid_t), sizeof(MSG));
+        vid_t lid = frag_->gid2lid(gid);
+        if (lid != kInvalidVid) f(tid, lid, m);
+      }, 2048);
+      ar.skip(n * unit);
+    }
+  }
+
  private:
   TcpComm* comm_ = nullptr;
   const Fragment* frag_ = nullptr;

@@ -58,6 +58,16 @@ def main():
         res = eng.cdlp(g, 10)
     elif app == "lcc":
         res = eng.lcc(g)
+    elif app == "bc":
+        res = eng.bc(g, cfg["source"])
+    elif app == "kcore":
+        res = eng.kcore(g, cfg.get("k", 3))
+    elif app == "core_decomposition":
+        res = eng.core_decomposition(g)
+    elif app == "kclique":
+        res = eng.kclique(g, cfg.get("k", 3))
+        res = dict(res, oids=np.array([0], dtype=np.int64),
+                   values=np.array([res["clique_count"]], dtype=np.int64))
     else:
         raise ValueError(app)
 

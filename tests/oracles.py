@@ -124,3 +124,115 @@ def lcc_oracle(num_v, src, dst, directed=True):
         cnt = sum(len(nv & outs[u]) for u in nv)
         out[v] = cnt / (deg * (deg - 1))
     return out
+
+
+def bc_oracle(num_v, src, dst, source, directed=True):
+    """Single-source Brandes dependency delta over the stored adjacency
+    (undirected: both orientations; duplicates count as parallel paths)."""
+    adj = [[] for _ in range(num_v)]
+    for s, d in zip(src, dst):
+        adj[s].append(d)
+        if not directed and s != d:
+            adj[d].append(s)
+    INF = float("inf")
+    depth = np.full(num_v, INF)
+    sigma = np.zeros(num_v)
+    depth[source] = 0
+    sigma[source] = 1.0
+    frontier = [source]
+    order = []
+    while frontier:
+        order.extend(frontier)
+        nxt = set()
+        for v in frontier:
+            for u in adj[v]:
+                if depth[u] == INF or depth[u] == depth[v] + 1:
+                    if depth[u] == INF:
+                        depth[u] = depth[v] + 1
+                        nxt.add(u)
+                    sigma[u] += sigma[v]
+        frontier = sorted(nxt)
+    delta = np.zeros(num_v)
+    for v in reversed(order):
+        for u in adj[v]:
+            if depth[u] == depth[v] - 1:
+                delta[u] += sigma[u] / sigma[v] * (1.0 + delta[v])
+    return delta, sigma, depth
+
+
+def _stored_adj(num_v, src, dst, directed):
+    """Multiset adjacency matching Fragment storage (self-loops kept on the
+    forward orientation, reverse of self-loops dropped for undirected)."""
+    adj = [[] for _ in range(num_v)]
+    for s, d in zip(src, dst):
+        adj[s].append(d)
+        if not directed and s != d:
+            adj[d].append(s)
+    return adj
+
+
+def coreness_oracle(num_v, src, dst, directed=False):
+    """Peeling coreness; degree = stored multiplicity, self-loops excluded."""
+    adj = _stored_adj(num_v, src, dst, directed)
+    rem = np.array([sum(1 for u in a if u != v)
+                    for v, a in enumerate(adj)], dtype=np.int64)
+    core = np.zeros(num_v, dtype=np.int64)
+    alive = np.ones(num_v, bool)
+    for k in range(0, int(rem.max(initial=0)) + 2):
+        changed = True
+        while changed:
+            changed = False
+            for v in range(num_v):
+                if alive[v] and rem[v] <= k:
+                    alive[v] = False
+                    core[v] = k
+                    for u in adj[v]:
+                        if u != v and alive[u]:
+                            rem[u] -= 1
+                    changed = True
+        if not alive.any():
+            break
+    return core
+
+
+def kcore_oracle(num_v, src, dst, k, directed=False):
+    adj = _stored_adj(num_v, src, dst, directed)
+    rem = np.array([sum(1 for u in a if u != v)
+                    for v, a in enumerate(adj)], dtype=np.int64)
+    alive = np.ones(num_v, bool)
+    changed = True
+    while changed:
+        changed = False
+        for v in range(num_v):
+            if alive[v] and rem[v] < k:
+                alive[v] = False
+                for u in adj[v]:
+                    if u != v and alive[u]:
+                        rem[u] -= 1
+                changed = True
+    return alive.astype(np.int64)
+
+
+def kclique_oracle(num_v, src, dst, k, directed=False):
+    """Count k-cliques on the simple undirected graph."""
+    nbrs = [set() for _ in range(num_v)]
+    for s, d in zip(src, dst):
+        if s == d:
+            continue
+        nbrs[s].add(d)
+        nbrs[d].add(s)
+    oriented = [sorted(u for u in nbrs[v] if u > v) for v in range(num_v)]
+
+    def rec(cand, depth):
+        if depth == k - 1:
+            return len(cand)
+        total = 0
+        for c in cand:
+            nxt = [x for x in cand if x in nbrs[c] and x > c]
+            if nxt:
+                total += rec(nxt, depth + 1)
+        return total
+
+    if k == 2:
+        return sum(len(o) for o in oriented)
+    return sum(rec(o, 1) for o in oriented if o)

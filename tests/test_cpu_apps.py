@@ -3,8 +3,10 @@ import numpy as np
 import pytest
 
 import grapehip
-from oracles import (bfs_oracle, cdlp_oracle, lcc_oracle, pagerank_oracle,
-                     sssp_oracle, wcc_oracle, INT64_MAX, DBL_MAX)
+from oracles import (bc_oracle, bfs_oracle, cdlp_oracle, coreness_oracle,
+                     kclique_oracle, kcore_oracle, lcc_oracle,
+                     pagerank_oracle, sssp_oracle, wcc_oracle, INT64_MAX,
+                     DBL_MAX)
 
 
 @pytest.fixture(scope="module")
@@ -137,3 +139,56 @@ def test_hashmap_oids(eng):
     oids, vals = sorted_by_oid(eng.bfs(g, 10))
     assert list(oids) == [10, 20, 30, 40, 77]
     assert list(vals) == [0, 1, 2, INT64_MAX, 3]
+
+
+# ---- extras: BC / k-core / core decomposition / k-clique -------------------
+
+def test_bc_undirected(eng):
+    src, dst, _ = random_graph(num_v=800, num_e=4000, seed=51)
+    g = eng.load_edges(src, dst, directed=False, num_vertices=800)
+    r = eng.bc(g, 5)
+    order = np.argsort(r["oids"])
+    vals = r["values"][order]
+    sig = r["path_num"][order]
+    dep = r["depth"][order]
+    e_delta, e_sigma, e_depth = bc_oracle(800, src, dst, 5, directed=False)
+    reach = e_depth < 1e300
+    assert np.array_equal(dep[reach], e_depth[reach].astype(np.int64))
+    assert np.allclose(sig[reach], e_sigma[reach])
+    assert np.allclose(vals[reach], e_delta[reach], rtol=1e-9, atol=1e-9)
+
+
+def test_bc_directed(eng):
+    src, dst, _ = random_graph(num_v=600, num_e=3000, seed=53)
+    g = eng.load_edges(src, dst, directed=True, num_vertices=600)
+    r = eng.bc(g, 2)
+    order = np.argsort(r["oids"])
+    e_delta, e_sigma, e_depth = bc_oracle(600, src, dst, 2, directed=True)
+    reach = e_depth < 1e300
+    assert np.allclose(r["values"][order][reach], e_delta[reach], rtol=1e-9)
+
+
+def test_kcore(eng):
+    src, dst, _ = random_graph(num_v=500, num_e=3000, seed=57)
+    g = eng.load_edges(src, dst, directed=False, num_vertices=500)
+    for k in (2, 4, 7):
+        r = eng.kcore(g, k)
+        order = np.argsort(r["oids"])
+        assert np.array_equal(r["values"][order],
+                              kcore_oracle(500, src, dst, k)), k
+
+
+def test_core_decomposition(eng):
+    src, dst, _ = random_graph(num_v=400, num_e=2500, seed=59)
+    g = eng.load_edges(src, dst, directed=False, num_vertices=400)
+    r = eng.core_decomposition(g)
+    order = np.argsort(r["oids"])
+    assert np.array_equal(r["values"][order], coreness_oracle(400, src, dst))
+
+
+def test_kclique(eng):
+    src, dst, _ = random_graph(num_v=120, num_e=1400, seed=61)
+    g = eng.load_edges(src, dst, directed=False, num_vertices=120)
+    for k in (3, 4, 5):
+        r = eng.kclique(g, k)
+        assert r["clique_count"] == kclique_oracle(120, src, dst, k), k

@@ -11,8 +11,9 @@ from pathlib import Path
 import numpy as np
 import pytest
 
-from oracles import (bfs_oracle, cdlp_oracle, lcc_oracle, pagerank_oracle,
-                     sssp_oracle, wcc_oracle)
+from oracles import (bc_oracle, bfs_oracle, cdlp_oracle, coreness_oracle,
+                     kclique_oracle, kcore_oracle, lcc_oracle,
+                     pagerank_oracle, sssp_oracle, wcc_oracle)
 
 REPO = Path(__file__).resolve().parent.parent
 WORKER = REPO / "tests" / "mp_worker.py"
@@ -118,3 +119,40 @@ def test_bfs_mp_hashmap_oids(world, free_port, tmp_path):
     expect = bfs_oracle(cfg["num_v"], src, dst, 3)
     assert np.array_equal(oids, np.arange(cfg["num_v"]) * 3 + 1)
     assert np.array_equal(vals, expect)
+
+
+@pytest.mark.parametrize("world", [2])
+def test_bc_mp(world, free_port, tmp_path):
+    cfg = dict(BASE, app="bc", directed=False, num_v=250, num_e=1200)
+    oids, vals = run_world(world, cfg, free_port, tmp_path)
+    src, dst, _ = graph_arrays(cfg)
+    delta, _, depth = bc_oracle(cfg["num_v"], src, dst, cfg["source"],
+                                directed=False)
+    reach = depth < 1e300
+    assert np.allclose(vals[reach], delta[reach], rtol=1e-9, atol=1e-9)
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_kcore_mp(world, free_port, tmp_path):
+    cfg = dict(BASE, app="kcore", directed=False, num_v=300, num_e=2000, k=4)
+    oids, vals = run_world(world, cfg, free_port, tmp_path)
+    src, dst, _ = graph_arrays(cfg)
+    assert np.array_equal(vals, kcore_oracle(cfg["num_v"], src, dst, 4))
+
+
+@pytest.mark.parametrize("world", [2])
+def test_core_decomposition_mp(world, free_port, tmp_path):
+    cfg = dict(BASE, app="core_decomposition", directed=False, num_v=250,
+               num_e=1500)
+    oids, vals = run_world(world, cfg, free_port, tmp_path)
+    src, dst, _ = graph_arrays(cfg)
+    assert np.array_equal(vals, coreness_oracle(cfg["num_v"], src, dst))
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_kclique_mp(world, free_port, tmp_path):
+    cfg = dict(BASE, app="kclique", directed=False, num_v=100, num_e=900, k=4)
+    oids, vals = run_world(world, cfg, free_port, tmp_path)
+    src, dst, _ = graph_arrays(cfg)
+    # every rank reports the same global count
+    assert (vals == kclique_oracle(cfg["num_v"], src, dst, 4)).all()
