@@ -22,6 +22,7 @@
 #include "apps/sssp.hpp"
 #include "apps/wcc.hpp"
 #include "core/fragment.hpp"
+#include "core/serialize.hpp"
 #include "core/message_manager.hpp"
 #include "core/net.hpp"
 
@@ -274,6 +275,31 @@ PYBIND11_MODULE(_core, m) {
            py::arg("num_vertices") = -1, py::arg("vertex_oids") = std::nullopt,
            py::arg("build_in_csr") = false,
            py::arg("partitioner") = "segmented")
+      .def("save_graph",
+           [](PyEngine& eng, PyGraph& g, const std::string& prefix) {
+             if (!g.frag)
+               throw std::runtime_error(
+                   "save_graph needs a host fragment (device-only synthetic "
+                   "graphs are regenerated, not checkpointed)");
+             py::gil_scoped_release rel;
+             serialize_graph(*g.frag, prefix);
+             if (eng.c()) eng.c()->barrier();
+           },
+           py::arg("graph"), py::arg("prefix"))
+      .def("load_serialized",
+           [](PyEngine& eng, const std::string& prefix) {
+             auto g = std::make_shared<PyGraph>();
+             py::gil_scoped_release rel;
+             auto [vm, frag] = deserialize_graph(prefix, eng.rank);
+             g->vm = vm;
+             g->frag = std::move(frag);
+#ifdef GRAPEHIP_WITH_HIP
+             if (eng.use_gpu) g->dev = eng.gpu->upload(*g->frag);
+#endif
+             if (eng.c()) eng.c()->barrier();
+             return g;
+           },
+           py::arg("prefix"))
       .def("load_synthetic",
            [](PyEngine& eng, uint64_t nv, uint64_t ne, uint64_t seed,
               bool directed, bool weighted, double a, double b, double c) {

@@ -280,6 +280,45 @@ class Fragment {
     return frag;
   }
 
+  // Rebuild from a serialized checkpoint (see core/serialize.hpp). R is any
+  // reader with pod<T>() / vec(std::vector<T>&).
+  template <typename R>
+  static std::unique_ptr<Fragment> FromParts(std::shared_ptr<VertexMap> vm,
+                                             R& r, uint32_t fnum) {
+    auto frag = std::make_unique<Fragment>();
+    Fragment& F = *frag;
+    F.vm_ = std::move(vm);
+    F.fnum_ = static_cast<int>(fnum);
+    F.parser_ = F.vm_->parser();
+    F.fid_ = static_cast<fid_t>(r.template pod<uint32_t>());
+    F.directed_ = r.template pod<uint8_t>() != 0;
+    F.total_edges_ = r.template pod<uint64_t>();
+    F.input_edges_ = r.template pod<uint64_t>();
+    F.total_vertices_ = F.vm_->total_vertices();
+    F.ivnum_ = F.vm_->frag_vnum(F.fid_);
+    r.vec(F.oe_off_);
+    r.vec(F.oe_dst_);
+    r.vec(F.oe_w_);
+    r.vec(F.ie_off_);
+    r.vec(F.ie_dst_);
+    r.vec(F.ie_w_);
+    r.vec(F.ovgid_);
+    F.ovnum_ = static_cast<vid_t>(F.ovgid_.size());
+    F.ovg2l_.clear();
+    F.ovg2l_.reserve(F.ovnum_ * 2);
+    for (vid_t i = 0; i < F.ovnum_; ++i)
+      F.ovg2l_.emplace(F.ovgid_[i], F.ivnum_ + i);
+    F.outer_range_.resize(fnum);
+    F.mirrors_.assign(fnum, {});
+    for (uint32_t f = 0; f < fnum; ++f) {
+      vid_t b = r.template pod<vid_t>();
+      vid_t e = r.template pod<vid_t>();
+      F.outer_range_[f] = {b, e};
+      r.vec(F.mirrors_[f]);
+    }
+    return frag;
+  }
+
  private:
   fid_t fid_ = 0;
   int fnum_ = 1;

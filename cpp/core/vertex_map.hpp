@@ -142,6 +142,26 @@ class VertexMap {
     return static_cast<vid_t>(l2o_[f].size());
   }
 
+  // Rebuild hashmap mode from already-replicated per-fragment oid lists
+  // (deserialization path — no communication).
+  void init_hashmap_local(int fnum, PartitionerKind pk,
+                          std::vector<std::vector<oid_t>> l2o) {
+    fnum_ = fnum;
+    parser_.init(fnum);
+    idxer_ = IdxerKind::kHashmap;
+    pkind_ = pk;
+    l2o_ = std::move(l2o);
+    o2g_.clear();
+    total_vertices_ = 0;
+    for (int f = 0; f < fnum; ++f) total_vertices_ += l2o_[f].size();
+    o2g_.reserve(total_vertices_);
+    for (int f = 0; f < fnum; ++f)
+      for (size_t l = 0; l < l2o_[f].size(); ++l)
+        o2g_.emplace(l2o_[f][l], parser_.gid(f, static_cast<vid_t>(l)));
+  }
+
+  const std::vector<oid_t>& frag_oids(fid_t f) const { return l2o_[f]; }
+
   uint64_t total_vertices() const { return total_vertices_; }
   const IdParser& parser() const { return parser_; }
   int fnum() const { return fnum_; }

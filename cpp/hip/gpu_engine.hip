@@ -2500,22 +2500,41 @@ __global__ void lcc_compact_kernel(const uint64_t* __restrict__ ooff,
   }
 }
 
-// bucket rows for the sort tiers by oriented count
+// bucket rows for the sort tiers by oriented count (block-aggregated
+// counter reservation, same pattern as bucket_rows_kernel)
 __global__ void lcc_sortbucket_kernel(const uint64_t* __restrict__ goff,
                                       uint32_t owned, uint32_t v_begin,
                                       uint32_t* lds_rows,
                                       unsigned long long* c_lds,
                                       uint32_t* big_rows,
                                       unsigned long long* c_big) {
-  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
-  for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < owned;
-       r += stride) {
-    uint64_t n = goff[v_begin + r + 1] - goff[v_begin + r];
-    if (n < 2) continue;
-    if (n <= kLccSortLds)
-      lds_rows[atomicAdd(c_lds, 1ull)] = r;
-    else
-      big_rows[atomicAdd(c_big, 1ull)] = r;
+  __shared__ uint32_t s_cnt[2];
+  __shared__ unsigned long long s_base[2];
+  uint32_t* lists[2] = {lds_rows, big_rows};
+  unsigned long long* gcnt[2] = {c_lds, c_big};
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t base = blockIdx.x * blockDim.x; base < owned;
+       base += stride) {
+    if (threadIdx.x < 2) s_cnt[threadIdx.x] = 0;
+    __syncthreads();
+    uint32_t r = base + threadIdx.x;
+    int b = -1;
+    uint32_t loc = 0;
+    if (r < owned) {
+      uint64_t n = goff[v_begin + r + 1] - goff[v_begin + r];
+      if (n >= 2) {
+        b = n <= kLccSortLds ? 0 : 1;
+        loc = atomicAdd(&s_cnt[b], 1u);
+      }
+    }
+    __syncthreads();
+    if (threadIdx.x < 2 && s_cnt[threadIdx.x])
+      s_base[threadIdx.x] =
+          atomicAdd(gcnt[threadIdx.x],
+                    static_cast<unsigned long long>(s_cnt[threadIdx.x]));
+    __syncthreads();
+    if (b >= 0) lists[b][s_base[b] + loc] = r;
+    __syncthreads();
   }
 }
 
@@ -2626,7 +2645,28 @@ __global__ void lcc_bigcap_kernel(const uint64_t* __restrict__ goff,
   }
 }
 
-// wave-per-row triangle counting over sorted oriented CSR
+// wave-per-row triangle counting over the sorted oriented CSR.
+// Per oriented edge (u,v) a lane intersects O(u) with O(v) by walking the
+// SMALLER list and binary-searching the larger (the naive per-lane serial
+// merge rescans all of O(u) per edge — quadratic in hub rows; this pass
+// dropped 2.11s -> see profiles/). Hits credit u (wave-aggregated), v and
+// the witness w.
+__device__ __forceinline__ bool lcc_bsearch(const uint32_t* __restrict__ a,
+                                            uint32_t n, uint32_t key) {
+  uint32_t lo = 0, hi = n;
+  while (lo < hi) {
+    uint32_t mid = (lo + hi) >> 1;
+    uint32_t x = a[mid];
+    if (x < key)
+      lo = mid + 1;
+    else if (x > key)
+      hi = mid;
+    else
+      return true;
+  }
+  return false;
+}
+
 __global__ void lcc_triangle_kernel(const uint64_t* __restrict__ goff,
                                     const uint32_t* __restrict__ gdst,
                                     uint32_t owned, uint32_t v_begin,
@@ -2641,24 +2681,31 @@ __global__ void lcc_triangle_kernel(const uint64_t* __restrict__ goff,
     uint64_t ub = goff[u], ue = goff[u + 1];
     uint32_t un = static_cast<uint32_t>(ue - ub);
     if (un < 1) continue;
-    unsigned long long my_u = 0;  // triangles found with u as pivot
+    const uint32_t* A = gdst + ub;
+    unsigned long long my_u = 0;
     for (uint32_t k = lane; k < un; k += kWave) {
-      uint32_t v = gdst[ub + k];
-      uint64_t vb = goff[v], vend = goff[v + 1];
-      // merge-intersect gdst[ub..ue) with gdst[vb..vend)
-      uint64_t a = ub, b = vb;
+      uint32_t v = A[k];
+      uint64_t vb = goff[v];
+      uint32_t vn = static_cast<uint32_t>(goff[v + 1] - vb);
+      if (vn == 0) continue;
+      const uint32_t* B = gdst + vb;
+      // walk the smaller, search the larger
+      const uint32_t* small = A;
+      uint32_t sn = un;
+      const uint32_t* big = B;
+      uint32_t bn = vn;
+      if (vn < un) {
+        small = B;
+        sn = vn;
+        big = A;
+        bn = un;
+      }
       unsigned long long hits = 0;
-      while (a < ue && b < vend) {
-        uint32_t x = gdst[a], y = gdst[b];
-        if (x < y) {
-          ++a;
-        } else if (x > y) {
-          ++b;
-        } else {
+      for (uint32_t t = 0; t < sn; ++t) {
+        uint32_t w = small[t];
+        if (lcc_bsearch(big, bn, w)) {
           ++hits;
-          atomicAdd(&T[x], 1ull);  // witness w
-          ++a;
-          ++b;
+          atomicAdd(&T[w], 1ull);  // witness
         }
       }
       if (hits) {

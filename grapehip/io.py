@@ -33,16 +33,30 @@ def _byte_slice(path: str, rank: int, world: int):
     return data
 
 
+def _ncols(path: str) -> int:
+    with open(path, "rb") as f:
+        line = f.readline()
+    return len(line.split())
+
+
 def read_ldbc_edges(efile: str, weighted: bool = False, rank: int = 0,
                     world: int = 1):
-    """Parse this rank's slice of an LDBC .e file (src dst [weight])."""
+    """Parse this rank's slice of an LDBC .e file (src dst [weight]).
+
+    The file's column count is probed from its first line: a weight column
+    present in the file is parsed (and dropped if weighted=False), so
+    unweighted apps work on weighted .e files, like the reference loader.
+    """
     data = _byte_slice(efile, rank, world)
     if not data.strip():
         empty = np.zeros(0, dtype=np.int64)
         return (empty, empty.copy(),
                 np.zeros(0, dtype=np.float32) if weighted else None)
-    fields = np.array(data.split(), dtype=np.float64 if weighted else np.int64)
-    ncols = 3 if weighted else 2
+    ncols = _ncols(efile)
+    if weighted and ncols < 3:
+        raise ValueError("%s has no weight column" % efile)
+    fields = np.array(data.split(),
+                      dtype=np.float64 if ncols >= 3 else np.int64)
     fields = fields.reshape(-1, ncols)
     src = fields[:, 0].astype(np.int64)
     dst = fields[:, 1].astype(np.int64)

@@ -1,0 +1,80 @@
+"""CLI driver (grapehip.run_app) end-to-end on the reference's bundled
+p2p-31 dataset, verified against the LDBC golden outputs like the
+reference's misc/app_tests.sh ExactVerify/EpsVerify."""
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+import numpy as np
+import pytest
+
+REPO = Path(__file__).resolve().parent.parent
+DATASET = Path("/root/reference/dataset")
+
+pytestmark = pytest.mark.skipif(
+    not DATASET.exists(), reason="reference dataset not available")
+
+
+def run_cli(tmp_path, *extra):
+    out = tmp_path / "out"
+    cmd = [sys.executable, "-m", "grapehip.run_app",
+           "--efile", str(DATASET / "p2p-31.e"),
+           "--vfile", str(DATASET / "p2p-31.v"),
+           "--out_prefix", str(out), *extra]
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    r = subprocess.run(cmd, cwd=REPO, env=env, capture_output=True, text=True,
+                       timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    return out / "result_frag_0", r.stdout
+
+
+def load_pairs(path):
+    oids, vals = [], []
+    for line in open(path):
+        a, b = line.split()
+        oids.append(int(a))
+        vals.append(b)
+    return np.array(oids), vals
+
+
+def test_cli_bfs_golden(tmp_path):
+    res, out = run_cli(tmp_path, "--application", "bfs", "--bfs_source", "6")
+    assert "run algorithm" in out
+    g_oids, g_vals = load_pairs(DATASET / "p2p-31-BFS")
+    oids, vals = load_pairs(res)
+    order, gorder = np.argsort(oids), np.argsort(g_oids)
+    assert np.array_equal(oids[order], g_oids[gorder])
+    got = np.array([int(v) for v in vals], dtype=np.int64)[order]
+    exp = np.array([int(v) for v in g_vals], dtype=np.int64)[gorder]
+    assert np.array_equal(got, exp)
+
+
+def test_cli_sssp_golden(tmp_path):
+    res, _ = run_cli(tmp_path, "--application", "sssp", "--sssp_source", "6")
+    g_oids, g_vals = load_pairs(DATASET / "p2p-31-SSSP")
+    oids, vals = load_pairs(res)
+    order, gorder = np.argsort(oids), np.argsort(g_oids)
+    got = np.array([float(v) for v in vals])[order]
+    exp = np.array([float(v) for v in g_vals])[gorder]
+    finite = exp < 1e300
+    assert np.allclose(got[finite], exp[finite], rtol=1e-5)
+    assert (got[~finite] > 1e300).all()
+
+
+def test_cli_serialize_roundtrip(tmp_path):
+    ser = tmp_path / "ckpt"
+    res1, _ = run_cli(tmp_path, "--application", "wcc",
+                      "--serialize", "--serialization_prefix", str(ser))
+    out2 = tmp_path / "out2"
+    cmd = [sys.executable, "-m", "grapehip.run_app",
+           "--application", "wcc", "--deserialize",
+           "--serialization_prefix", str(ser),
+           "--out_prefix", str(out2)]
+    env = dict(os.environ, PYTHONPATH=str(REPO))
+    r = subprocess.run(cmd, cwd=REPO, env=env, capture_output=True,
+                       text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    a = open(res1).read()
+    b = open(out2 / "result_frag_0").read()
+    assert a == b
