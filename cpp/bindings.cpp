@@ -11,6 +11,7 @@
 #include <chrono>
 #include <memory>
 #include <optional>
+#include <unordered_set>
 
 #include "apps/bc.hpp"
 #include "apps/bfs.hpp"
@@ -19,6 +20,7 @@
 #include "apps/kcore.hpp"
 #include "apps/lcc.hpp"
 #include "apps/pagerank.hpp"
+#include "apps/sampler.hpp"
 #include "apps/sssp.hpp"
 #include "apps/wcc.hpp"
 #include "core/fragment.hpp"
@@ -275,6 +277,95 @@ PYBIND11_MODULE(_core, m) {
            py::arg("num_vertices") = -1, py::arg("vertex_oids") = std::nullopt,
            py::arg("build_in_csr") = false,
            py::arg("partitioner") = "segmented")
+      .def("mutate_graph",
+           [](PyEngine& eng, PyGraph& g, arr_i64 add_src, arr_i64 add_dst,
+              std::optional<arr_f32> add_w, arr_i64 rm_src, arr_i64 rm_dst,
+              arr_i64 rm_vertices) {
+             // Reference parity: LoadGraphAndMutate / Mutation{add/remove
+             // edges,vertices} (loader.h:55-68, basic_fragment_mutator.h).
+             // grapehip mutates by functional rebuild: owned triples are
+             // filtered against the (replicated) removal sets, additions
+             // are merged, and the distributed builder reruns.
+             if (!g.frag)
+               throw std::runtime_error("mutate_graph needs a host fragment");
+             size_t na = add_src.size();
+             if (static_cast<size_t>(add_dst.size()) != na)
+               throw std::runtime_error("add src/dst size mismatch");
+             size_t nr = rm_src.size();
+             if (static_cast<size_t>(rm_dst.size()) != nr)
+               throw std::runtime_error("remove src/dst size mismatch");
+             auto out = std::make_shared<PyGraph>();
+             out->vm = g.vm;
+             bool weighted =
+                 g.frag->has_weights() || add_w.has_value();
+             py::gil_scoped_release rel;
+
+             // replicate removal lists to every rank
+             auto replicate = [&](const int64_t* p, size_t n) {
+               std::vector<oid_t> all(p, p + n);
+               if (eng.c()) {
+                 std::string mine(reinterpret_cast<const char*>(p),
+                                  n * sizeof(int64_t));
+                 std::vector<std::string> send(eng.world, mine);
+                 send[eng.rank].clear();
+                 auto recv = eng.c()->exchange_all(send);
+                 for (int f = 0; f < eng.world; ++f) {
+                   if (f == eng.rank) continue;
+                   size_t m = recv[f].size() / sizeof(int64_t);
+                   const int64_t* q =
+                       reinterpret_cast<const int64_t*>(recv[f].data());
+                   all.insert(all.end(), q, q + m);
+                 }
+               }
+               return all;
+             };
+             std::vector<oid_t> rms = replicate(rm_src.data(), nr);
+             std::vector<oid_t> rmd = replicate(rm_dst.data(), nr);
+             std::vector<oid_t> rmv =
+                 replicate(rm_vertices.data(), rm_vertices.size());
+
+             struct PairHash {
+               size_t operator()(const std::pair<oid_t, oid_t>& p) const {
+                 return std::hash<uint64_t>()(
+                     static_cast<uint64_t>(p.first) * 0x9e3779b97f4a7c15ULL ^
+                     static_cast<uint64_t>(p.second));
+               }
+             };
+             std::unordered_set<std::pair<oid_t, oid_t>, PairHash> rm_edges;
+             for (size_t i = 0; i < rms.size(); ++i) {
+               rm_edges.emplace(rms[i], rmd[i]);
+               if (!g.frag->directed()) rm_edges.emplace(rmd[i], rms[i]);
+             }
+             std::unordered_set<oid_t> rm_verts(rmv.begin(), rmv.end());
+
+             std::vector<EdgeTriple> triples = g.frag->to_triples();
+             uint64_t kept = 0;
+             for (auto& e : triples) {
+               if (rm_edges.count({e.src, e.dst})) continue;
+               if (rm_verts.count(e.src) || rm_verts.count(e.dst)) continue;
+               triples[kept++] = e;
+             }
+             triples.resize(kept);
+             {
+               auto sp = add_src.unchecked<1>();
+               auto dp = add_dst.unchecked<1>();
+               const float* wp = add_w ? add_w->data() : nullptr;
+               for (size_t i = 0; i < na; ++i)
+                 triples.push_back({sp(i), dp(i), wp ? wp[i] : 1.0f});
+             }
+             uint64_t n_input = triples.size();
+             out->frag = Fragment::Build(
+                 out->vm, eng.c(), eng.rank, eng.world, std::move(triples),
+                 g.frag->directed(), weighted, g.frag->has_in_csr(),
+                 n_input);
+#ifdef GRAPEHIP_WITH_HIP
+             if (eng.use_gpu) out->dev = eng.gpu->upload(*out->frag);
+#endif
+             return out;
+           },
+           py::arg("graph"), py::arg("add_src"), py::arg("add_dst"),
+           py::arg("add_weights") = std::nullopt, py::arg("remove_src"),
+           py::arg("remove_dst"), py::arg("remove_vertices"))
       .def("save_graph",
            [](PyEngine& eng, PyGraph& g, const std::string& prefix) {
              if (!g.frag)
@@ -546,6 +637,62 @@ PYBIND11_MODULE(_core, m) {
              return meta;
            },
            py::arg("graph"))
+      .def("sample",
+           [](PyEngine& eng, PyGraph& g, arr_i64 starts, int hops,
+              const std::string& strategy, int top_k, uint64_t seed) {
+             if (!g.frag)
+               throw std::runtime_error("sample needs a host fragment");
+             SampleStrategy st;
+             if (strategy == "random")
+               st = SampleStrategy::kRandom;
+             else if (strategy == "edge_weight")
+               st = SampleStrategy::kEdgeWeight;
+             else if (strategy == "top_k")
+               st = SampleStrategy::kTopK;
+             else
+               throw std::runtime_error(
+                   "strategy must be random|edge_weight|top_k");
+             std::vector<oid_t> st_oids(starts.data(),
+                                        starts.data() + starts.size());
+             SamplerApp app;
+             SamplerContext ctx;
+             MessageManager mm;
+             py::dict meta;
+             {
+               py::gil_scoped_release rel;
+               mm.init(eng.c(), g.frag.get(), eng.n_threads);
+               ctx.init(*g.frag, st_oids, hops, st, top_k, seed);
+               app.origin_map_.resize(st_oids.size());
+               for (size_t i = 0; i < st_oids.size(); ++i) {
+                 vid_t gid;
+                 app.origin_map_[i] =
+                     g.frag->vm().get_gid(st_oids[i], &gid)
+                         ? g.frag->parser().fid(gid)
+                         : 0;
+               }
+               RunWorker(app, ctx, *g.frag, mm);
+             }
+             size_t n = ctx.paths.size();
+             py::array_t<int64_t> walks(n);
+             py::array_t<int64_t> paths({n, static_cast<size_t>(hops + 1)});
+             auto* wp = walks.mutable_data();
+             auto* pp = paths.mutable_data();
+             for (size_t i = 0; i < n; ++i) {
+               wp[i] = static_cast<int64_t>(ctx.walk_ids[i]);
+               for (int h = 0; h <= hops; ++h) {
+                 vid_t gv = ctx.paths[i][h];
+                 pp[i * (hops + 1) + h] =
+                     gv == kInvalidVid ? -1 : g.frag->vm().get_oid(gv);
+               }
+             }
+             py::dict out;
+             out["walk_ids"] = walks;
+             out["paths"] = paths;
+             return out;
+           },
+           py::arg("graph"), py::arg("starts"), py::arg("hops") = 2,
+           py::arg("strategy") = "random", py::arg("top_k") = 4,
+           py::arg("seed") = 7)
       .def("kclique",
            [](PyEngine& eng, PyGraph& g, int k) {
              if (!g.frag)

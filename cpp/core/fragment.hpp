@@ -280,6 +280,31 @@ class Fragment {
     return frag;
   }
 
+  // Decode this fragment's stored edges back to oid-space triples, one per
+  // ORIGINAL input edge (undirected storage holds both orientations; only
+  // the src<=dst copy is emitted so a rebuild reproduces the multiset).
+  // Used by the mutation path (reference mutable_edgecut_fragment.h:289-399
+  // applies deltas in place; grapehip mutates by functional rebuild — the
+  // parallel builder is fast enough that slack-tracking CSRs don't pay).
+  std::vector<EdgeTriple> to_triples() const {
+    std::vector<EdgeTriple> out;
+    out.reserve(oe_dst_.size());
+    for (vid_t v = 0; v < ivnum_; ++v) {
+      oid_t so = lid2oid(v);
+      auto adj = out_edges(v);
+      for (size_t i = 0; i < adj.n; ++i) {
+        oid_t dg = lid2oid(adj.dst[i]);
+        if (!directed_) {
+          // keep one orientation per input edge; ties (self loops) kept
+          vid_t sg = lid2gid(v), dgid = lid2gid(adj.dst[i]);
+          if (sg > dgid) continue;
+        }
+        out.push_back({so, dg, adj.w ? adj.w[i] : 1.0f});
+      }
+    }
+    return out;
+  }
+
   // Rebuild from a serialized checkpoint (see core/serialize.hpp). R is any
   // reader with pod<T>() / vec(std::vector<T>&).
   template <typename R>

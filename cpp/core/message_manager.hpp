@@ -125,6 +125,26 @@ class MessageManager {
     }
   }
 
+  // Raw variant: no gid->lid decode at all — the routing gid is passed
+  // through untouched (payload-routed protocols like the GNN sampler).
+  template <typename MSG, typename F>
+  void process_raw(F&& f) {
+    constexpr size_t unit = sizeof(vid_t) + sizeof(MSG);
+    for (int src = 0; src < fnum_; ++src) {
+      OutArchive& ar = recv_[src];
+      size_t n = ar.remaining() / unit;
+      const char* base = ar.cursor();
+      parallel_for_tid(0, n, [&](int tid, size_t i) {
+        vid_t gid;
+        MSG m;
+        std::memcpy(&gid, base + i * unit, sizeof(vid_t));
+        std::memcpy(&m, base + i * unit + sizeof(vid_t), sizeof(MSG));
+        f(tid, gid, m);
+      }, 2048);
+      ar.skip(n * unit);
+    }
+  }
+
   // Like process(), but the payload may target a NON-owned vertex of the
   // receiving fragment (owner pushing its inner state to mirrors, e.g.
   // core decomposition estimates / BC depth refresh). Decodes through the

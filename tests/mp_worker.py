@@ -64,6 +64,10 @@ def main():
         res = eng.kcore(g, cfg.get("k", 3))
     elif app == "core_decomposition":
         res = eng.core_decomposition(g)
+    elif app == "sample":
+        r = eng.sample(g, np.arange(cfg.get("n_walks", 20), dtype=np.int64),
+                       hops=cfg.get("hops", 3), seed=5)
+        res = {"oids": r["walk_ids"], "values": r["paths"]}
     elif app == "kclique":
         res = eng.kclique(g, cfg.get("k", 3))
         res = dict(res, oids=np.array([0], dtype=np.int64),

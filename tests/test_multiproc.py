@@ -156,3 +156,23 @@ def test_kclique_mp(world, free_port, tmp_path):
     src, dst, _ = graph_arrays(cfg)
     # every rank reports the same global count
     assert (vals == kclique_oracle(cfg["num_v"], src, dst, 4)).all()
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_sampler_mp(world, free_port, tmp_path):
+    # walks hop across fragments; every consecutive pair must be an edge
+    cfg = dict(BASE, app="sample", directed=True, num_v=200, num_e=1600,
+               n_walks=20, hops=3)
+    wids, paths = run_world(world, cfg, free_port, tmp_path)
+    assert np.array_equal(np.sort(wids), np.arange(20))
+    src, dst, _ = graph_arrays(cfg)
+    adj = {}
+    for s, d in zip(src, dst):
+        adj.setdefault(int(s), set()).add(int(d))
+    for wid, path in zip(wids, paths):
+        assert path[0] == wid
+        for h in range(3):
+            a, b = int(path[h]), int(path[h + 1])
+            if a == -1 or b == -1:
+                continue
+            assert b in adj.get(a, set()), (wid, h)

@@ -1,0 +1,97 @@
+"""Graph mutation (add/remove edges and vertices): results on the mutated
+graph must equal a fresh build of the mutated edge list. Reference parity:
+LoadGraphAndMutate + MutableEdgecutFragment::Mutate + mutation app tests
+(misc/app_tests.sh:114-165)."""
+import numpy as np
+import pytest
+
+import grapehip
+from oracles import bfs_oracle, pagerank_oracle, sssp_oracle, wcc_oracle
+
+
+@pytest.fixture(scope="module")
+def eng():
+    return grapehip.Engine(rank=0, world=1, master_port=29651)
+
+
+def base_graph(seed=81, num_v=600, num_e=3000):
+    rng = np.random.default_rng(seed)
+    src = rng.integers(0, num_v, size=num_e, dtype=np.int64)
+    dst = rng.integers(0, num_v, size=num_e, dtype=np.int64)
+    keep = src != dst
+    return src[keep], dst[keep]
+
+
+def test_add_remove_edges(eng):
+    src, dst = base_graph()
+    nv = 600
+    g = eng.load_edges(src, dst, directed=False, num_vertices=nv)
+    # remove the first 200 edges, add 300 new ones
+    rm_s, rm_d = src[:200].copy(), dst[:200].copy()
+    rng = np.random.default_rng(83)
+    ad_s = rng.integers(0, nv, 300).astype(np.int64)
+    ad_d = rng.integers(0, nv, 300).astype(np.int64)
+    k = ad_s != ad_d
+    ad_s, ad_d = ad_s[k], ad_d[k]
+    g2 = eng.mutate_graph(g, add_src=ad_s, add_dst=ad_d,
+                          remove_src=rm_s, remove_dst=rm_d,
+                          remove_vertices=np.array([], dtype=np.int64))
+    # expected edge multiset: drop ALL copies of removed pairs, then add
+    pairs = set()
+    for a, b in zip(rm_s, rm_d):
+        pairs.add((a, b))
+        pairs.add((b, a))
+    keep = np.array([(a, b) not in pairs for a, b in zip(src, dst)])
+    e_src = np.concatenate([src[keep], ad_s])
+    e_dst = np.concatenate([dst[keep], ad_d])
+    r = eng.bfs(g2, 5)
+    order = np.argsort(r["oids"])
+    assert np.array_equal(r["values"][order],
+                          bfs_oracle(nv, e_src, e_dst, 5, directed=False))
+    r = eng.wcc(g2)
+    assert np.array_equal(r["values"][np.argsort(r["oids"])],
+                          wcc_oracle(nv, e_src, e_dst))
+
+
+def test_remove_vertices(eng):
+    src, dst = base_graph(seed=91)
+    nv = 600
+    g = eng.load_edges(src, dst, directed=True, num_vertices=nv,
+                       build_in_csr=True)
+    dead = np.array([3, 77, 200, 401], dtype=np.int64)
+    g2 = eng.mutate_graph(g, add_src=np.array([], dtype=np.int64),
+                          add_dst=np.array([], dtype=np.int64),
+                          remove_src=np.array([], dtype=np.int64),
+                          remove_dst=np.array([], dtype=np.int64),
+                          remove_vertices=dead)
+    deadset = set(dead.tolist())
+    keep = np.array([s not in deadset and d not in deadset
+                     for s, d in zip(src, dst)])
+    r = eng.pagerank(g2, 0.85, 10)
+    order = np.argsort(r["oids"])
+    expect = pagerank_oracle(nv, src[keep], dst[keep], 0.85, 10,
+                             directed=True)
+    assert np.allclose(r["values"][order], expect, rtol=1e-9)
+
+
+def test_mutate_weighted(eng):
+    src, dst = base_graph(seed=95)
+    nv = 600
+    rng = np.random.default_rng(97)
+    w = rng.random(len(src), dtype=np.float32) * 9 + 1
+    g = eng.load_edges(src, dst, weights=w, directed=True, num_vertices=nv)
+    ad_s = np.array([1, 2, 3], dtype=np.int64)
+    ad_d = np.array([4, 5, 6], dtype=np.int64)
+    ad_w = np.array([0.5, 0.25, 0.125], dtype=np.float32)
+    g2 = eng.mutate_graph(g, add_src=ad_s, add_dst=ad_d, add_weights=ad_w,
+                          remove_src=np.array([], dtype=np.int64),
+                          remove_dst=np.array([], dtype=np.int64),
+                          remove_vertices=np.array([], dtype=np.int64))
+    e_src = np.concatenate([src, ad_s])
+    e_dst = np.concatenate([dst, ad_d])
+    e_w = np.concatenate([w, ad_w])
+    r = eng.sssp(g2, 1)
+    order = np.argsort(r["oids"])
+    expect = sssp_oracle(nv, e_src, e_dst, e_w, 1, directed=True)
+    finite = expect < 1e300
+    assert np.allclose(r["values"][order][finite], expect[finite], rtol=1e-6)
