@@ -2645,30 +2645,72 @@ __global__ void lcc_bigcap_kernel(const uint64_t* __restrict__ goff,
   }
 }
 
-// wave-per-row triangle counting over the sorted oriented CSR.
-// Per oriented edge (u,v) a lane intersects O(u) with O(v) by walking the
-// SMALLER list and binary-searching the larger (the naive per-lane serial
-// merge rescans all of O(u) per edge — quadratic in hub rows; this pass
-// dropped 2.11s -> see profiles/). Hits credit u (wave-aggregated), v and
-// the witness w.
-__device__ __forceinline__ bool lcc_bsearch(const uint32_t* __restrict__ a,
-                                            uint32_t n, uint32_t key) {
-  uint32_t lo = 0, hi = n;
-  while (lo < hi) {
-    uint32_t mid = (lo + hi) >> 1;
-    uint32_t x = a[mid];
-    if (x < key)
-      lo = mid + 1;
-    else if (x > key)
-      hi = mid;
-    else
-      return true;
+// Triangle counting via per-vertex hash sets of oriented neighbors.
+// v2 history: per-lane serial merge rescanned O(u) per edge (quadratic in
+// hub rows); binary search made it WORSE (log n random reads beat by the
+// merge's streaming locality — 2.3s -> 3.7s at 40M). Hash probing is O(1)
+// random reads per element: enumerate the smaller list, probe the larger
+// list's open-addressing set. Tables are built once over the exchanged
+// oriented CSR (capacity 2x, pow2), ~8 bytes per oriented edge of HBM.
+__global__ void lcc_hashcap_kernel(const uint64_t* __restrict__ goff,
+                                   uint32_t nv, uint32_t* __restrict__ caps) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint32_t v = blockIdx.x * blockDim.x + threadIdx.x; v < nv;
+       v += stride) {
+    uint64_t n = goff[v + 1] - goff[v];
+    uint64_t cap = 1;  // empty rows are never probed
+    if (n) {
+      cap = 4;
+      while (cap < 2 * n) cap <<= 1;
+    }
+    caps[v] = static_cast<uint32_t>(cap);
   }
-  return false;
+}
+
+__global__ void lcc_hashfill_kernel(const uint64_t* __restrict__ goff,
+                                    const uint32_t* __restrict__ gdst,
+                                    const uint64_t* __restrict__ hoff,
+                                    uint32_t nv,
+                                    uint32_t* __restrict__ htab) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wpb = kBlock / kWave;
+  size_t wstride = static_cast<size_t>(gridDim.x) * wpb;
+  for (uint64_t v = static_cast<uint64_t>(blockIdx.x) * wpb + wid; v < nv;
+       v += wstride) {
+    uint64_t b = goff[v];
+    uint32_t n = static_cast<uint32_t>(goff[v + 1] - b);
+    if (n == 0) continue;
+    uint64_t hb = hoff[v];
+    const uint64_t mask = (hoff[v + 1] - hb) - 1;
+    uint32_t* t = htab + hb;
+    for (uint32_t k = lane; k < n; k += kWave) {
+      uint32_t x = gdst[b + k];
+      uint64_t idx = cdlp_hash(x) & mask;
+      for (;;) {
+        uint32_t old = atomicCAS(&t[idx], kCdlpEmpty, x);
+        if (old == kCdlpEmpty || old == x) break;
+        idx = (idx + 1) & mask;
+      }
+    }
+  }
+}
+
+__device__ __forceinline__ bool lcc_probe(const uint32_t* __restrict__ t,
+                                          uint64_t mask, uint32_t key) {
+  uint64_t idx = cdlp_hash(key) & mask;
+  for (;;) {
+    uint32_t x = t[idx];
+    if (x == key) return true;
+    if (x == kCdlpEmpty) return false;
+    idx = (idx + 1) & mask;
+  }
 }
 
 __global__ void lcc_triangle_kernel(const uint64_t* __restrict__ goff,
                                     const uint32_t* __restrict__ gdst,
+                                    const uint64_t* __restrict__ hoff,
+                                    const uint32_t* __restrict__ htab,
                                     uint32_t owned, uint32_t v_begin,
                                     unsigned long long* __restrict__ T) {
   const int lane = threadIdx.x & 63;
@@ -2681,29 +2723,31 @@ __global__ void lcc_triangle_kernel(const uint64_t* __restrict__ goff,
     uint64_t ub = goff[u], ue = goff[u + 1];
     uint32_t un = static_cast<uint32_t>(ue - ub);
     if (un < 1) continue;
-    const uint32_t* A = gdst + ub;
     unsigned long long my_u = 0;
     for (uint32_t k = lane; k < un; k += kWave) {
-      uint32_t v = A[k];
+      uint32_t v = gdst[ub + k];
       uint64_t vb = goff[v];
       uint32_t vn = static_cast<uint32_t>(goff[v + 1] - vb);
       if (vn == 0) continue;
-      const uint32_t* B = gdst + vb;
-      // walk the smaller, search the larger
-      const uint32_t* small = A;
-      uint32_t sn = un;
-      const uint32_t* big = B;
-      uint32_t bn = vn;
+      // enumerate the smaller list, probe the larger one's hash set
+      uint64_t eb, hb_big;
+      uint32_t en;
       if (vn < un) {
-        small = B;
-        sn = vn;
-        big = A;
-        bn = un;
+        eb = vb;
+        en = vn;
+        hb_big = hoff[u];
+      } else {
+        eb = ub;
+        en = un;
+        hb_big = hoff[v];
       }
+      const uint64_t mask =
+          ((vn < un ? hoff[u + 1] : hoff[v + 1]) - hb_big) - 1;
+      const uint32_t* t = htab + hb_big;
       unsigned long long hits = 0;
-      for (uint32_t t = 0; t < sn; ++t) {
-        uint32_t w = small[t];
-        if (lcc_bsearch(big, bn, w)) {
+      for (uint32_t e = 0; e < en; ++e) {
+        uint32_t w = gdst[eb + e];
+        if (lcc_probe(t, mask, w)) {
           ++hits;
           atomicAdd(&T[w], 1ull);  // witness
         }
@@ -2917,12 +2961,27 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
     }
   }
 
+  // per-vertex oriented hash sets (intersection via O(1) probes)
+  DeviceBuffer<uint32_t> hcap(nv_pad);
+  DeviceBuffer<uint64_t> hoff(static_cast<size_t>(nv_pad) + 1);
+  lcc_hashcap_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(goff.data(), nv_pad,
+                                                         hcap.data());
+  uint64_t htotal =
+      exclusive_scan(hcap.data(), hoff.data(), nv_pad, s, I.scan);
+  hcap.free();
+  DeviceBuffer<uint32_t> htab(htotal);
+  fill(htab.data(), kCdlpEmpty, htotal, s);
+  lcc_hashfill_kernel<<<grid_for(static_cast<size_t>(nv_pad) * kWave),
+                        kBlock, 0, s>>>(goff.data(), gdst.data(),
+                                        hoff.data(), nv_pad, htab.data());
+
   // triangle counting
   Tcnt.resize(nv_pad);
   Tcnt.zero(s);
   if (owned)
     lcc_triangle_kernel<<<grid_for(static_cast<size_t>(owned) * kWave),
-                          kBlock, 0, s>>>(goff.data(), gdst.data(), owned,
+                          kBlock, 0, s>>>(goff.data(), gdst.data(),
+                                          hoff.data(), htab.data(), owned,
                                           g.v_begin, Tcnt.data());
   if (multi)
     NCCL_CHECK(ncclAllReduce(Tcnt.data(), Tcnt.data(), nv_pad, ncclUint64,
