@@ -17,6 +17,7 @@
 #include <algorithm>
 #include <cmath>
 #include <cstring>
+#include <unordered_map>
 
 #include "dev_graph.hpp"
 
@@ -739,6 +740,34 @@ __global__ void wcc_compress_kernel(uint32_t* parent, uint32_t n) {
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (uint32_t v = blockIdx.x * blockDim.x + threadIdx.x; v < n; v += stride)
     parent[v] = wcc_find(parent, v);
+}
+
+// Afforest-style neighbor sampling (Sutton et al.): hook only the first K
+// edges of each row — a cheap streaming pass that already collapses most
+// of a power-law graph into its giant component.
+__global__ void wcc_sample_kernel(DevGraphView g, int k_sample,
+                                  uint32_t* parent, int* changed) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  const uint32_t rows = g.owned();
+  WccOp op{parent, changed};
+  for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < rows;
+       r += stride) {
+    uint64_t b = g.oe_off[r], e = g.oe_off[r + 1];
+    uint64_t hi = b + k_sample < e ? b + k_sample : e;
+    for (uint64_t i = b; i < hi; ++i)
+      op(g.v_begin + r, g.oe_dst[i], 1.0f);
+  }
+}
+
+// mark owned rows NOT in the giant component (bitmap -> frontier queue);
+// their remaining edges are the only ones the full pass must visit
+__global__ void wcc_mark_rest_kernel(const uint32_t* __restrict__ parent,
+                                     uint32_t owned, uint32_t v_begin,
+                                     uint32_t giant, DevBitmap bm) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < owned;
+       r += stride)
+    if (parent[v_begin + r] != giant) bm.set_once(r);
 }
 
 }  // namespace grapehip
@@ -1751,15 +1780,64 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
 
   iota_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(parent.data(), 0, nv_pad);
   int rounds = 0;
-  int nchunks = static_cast<int>((owned + kBlock - 1) / kBlock);
+  constexpr int kSample = 2;
+  // scratch for the remaining-rows frontier
+  DeviceBuffer<uint32_t> rest_bm((owned + 31) / 32 + 1);
+  DeviceBuffer<uint32_t> rest_q(owned ? owned : 1);
+  rest_bm.zero(s);
   for (;;) {
-    // local fixpoint: hook over local edges until no change
     int local_changed_any = 0;
+    // 1) sample pass: first K edges per row (streaming), then compress
+    d_changed.zero(s);
+    if (owned)
+      wcc_sample_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+          view, kSample, parent.data(), d_changed.data());
+    wcc_compress_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(parent.data(),
+                                                            nv_pad);
+    ++rounds;
+    if (multi) {
+      int ch0 = 0;
+      HIP_CHECK(hipMemcpyAsync(&ch0, d_changed.data(), 4,
+                               hipMemcpyDeviceToHost, s));
+      HIP_CHECK(hipStreamSynchronize(s));
+      if (ch0) local_changed_any = 1;
+    }
+    // 2) identify the giant component from a small host probe
+    uint32_t giant = 0;
+    {
+      const int kProbe = 509;
+      std::vector<uint32_t> roots(kProbe);
+      uint32_t step = nv_pad / kProbe ? nv_pad / kProbe : 1;
+      for (int i = 0; i < kProbe; ++i)
+        HIP_CHECK(hipMemcpyAsync(
+            &roots[i],
+            parent.data() + (static_cast<uint64_t>(i) * step) % nv_pad, 4,
+            hipMemcpyDeviceToHost, s));
+      HIP_CHECK(hipStreamSynchronize(s));
+      std::unordered_map<uint32_t, int> freq;
+      int best = 0;
+      for (uint32_t r : roots) {
+        int c = ++freq[r];
+        if (c > best) {
+          best = c;
+          giant = r;
+        }
+      }
+    }
+    // 3) full pass restricted to rows outside the giant component,
+    //    repeated to the local fixpoint
     for (;;) {
+      rest_bm.zero(s);
+      if (owned)
+        wcc_mark_rest_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+            parent.data(), owned, g.v_begin, giant,
+            DevBitmap{rest_bm.data()});
+      uint64_t nrest = compact_frontier(I, rest_bm.data(), owned, g.v_begin,
+                                        rest_q.data(), s);
+      if (nrest == 0) break;
       d_changed.zero(s);
-      expand_cm_range<false, WccOp>
-          <<<std::min(nchunks, kMaxGrid), kBlock, 0, s>>>(
-              view, WccOp{parent.data(), d_changed.data()});
+      expand_frontier(I, view, rest_q.data(), static_cast<uint32_t>(nrest),
+                      WccOp{parent.data(), d_changed.data()}, s);
       int ch = 0;
       HIP_CHECK(hipMemcpyAsync(&ch, d_changed.data(), 4,
                                hipMemcpyDeviceToHost, s));
@@ -1772,10 +1850,12 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
     }
     if (!multi) break;
     bool any = comm_->allreduce_or(local_changed_any != 0);
-    if (!any) break;
     NCCL_CHECK(ncclAllReduce(parent.data(), parent.data(), nv_pad,
                              ncclUint32, ncclMin, I.nccl, s));
+    wcc_compress_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(parent.data(),
+                                                            nv_pad);
     HIP_CHECK(hipStreamSynchronize(s));
+    if (!any) break;
   }
   HIP_CHECK(hipDeviceSynchronize());
   if (comm_) comm_->barrier();
@@ -2712,7 +2792,10 @@ __global__ void lcc_triangle_kernel(const uint64_t* __restrict__ goff,
                                     const uint64_t* __restrict__ hoff,
                                     const uint32_t* __restrict__ htab,
                                     uint32_t owned, uint32_t v_begin,
-                                    unsigned long long* __restrict__ T) {
+                                    unsigned long long* __restrict__ T,
+                                    uint32_t heavy_thresh,
+                                    unsigned long long* __restrict__ heavy_q,
+                                    unsigned long long* __restrict__ heavy_n) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int wpb = kBlock / kWave;
@@ -2729,27 +2812,35 @@ __global__ void lcc_triangle_kernel(const uint64_t* __restrict__ goff,
       uint64_t vb = goff[v];
       uint32_t vn = static_cast<uint32_t>(goff[v + 1] - vb);
       if (vn == 0) continue;
-      // enumerate the smaller list, probe the larger one's hash set
-      uint64_t eb, hb_big;
+      uint32_t mn = vn < un ? vn : un;
+      if (mn > heavy_thresh) {
+        // defer: one lane grinding thousands of probes stalls its whole
+        // wave; heavy edges get a wave each in the second pass
+        heavy_q[atomicAdd(heavy_n, 1ull)] =
+            (static_cast<unsigned long long>(u) << 32) | v;
+        continue;
+      }
+      uint64_t eb, hb_big, hcap_end;
       uint32_t en;
       if (vn < un) {
         eb = vb;
         en = vn;
         hb_big = hoff[u];
+        hcap_end = hoff[u + 1];
       } else {
         eb = ub;
         en = un;
         hb_big = hoff[v];
+        hcap_end = hoff[v + 1];
       }
-      const uint64_t mask =
-          ((vn < un ? hoff[u + 1] : hoff[v + 1]) - hb_big) - 1;
+      const uint64_t mask = (hcap_end - hb_big) - 1;
       const uint32_t* t = htab + hb_big;
       unsigned long long hits = 0;
       for (uint32_t e = 0; e < en; ++e) {
         uint32_t w = gdst[eb + e];
         if (lcc_probe(t, mask, w)) {
           ++hits;
-          atomicAdd(&T[w], 1ull);  // witness
+          atomicAdd(&T[w], 1ull);
         }
       }
       if (hits) {
@@ -2761,6 +2852,58 @@ __global__ void lcc_triangle_kernel(const uint64_t* __restrict__ goff,
     for (int d = 32; d > 0; d >>= 1)
       my_u += __shfl_down(my_u, d, 64);
     if (lane == 0 && my_u) atomicAdd(&T[u], my_u);
+  }
+}
+
+// second pass: one wave per heavy edge; lanes stride the smaller list
+__global__ void lcc_triangle_heavy_kernel(
+    const uint64_t* __restrict__ goff, const uint32_t* __restrict__ gdst,
+    const uint64_t* __restrict__ hoff, const uint32_t* __restrict__ htab,
+    const unsigned long long* __restrict__ heavy_q, uint64_t heavy_n,
+    unsigned long long* __restrict__ T) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wpb = kBlock / kWave;
+  size_t wstride = static_cast<size_t>(gridDim.x) * wpb;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid;
+       i < heavy_n; i += wstride) {
+    unsigned long long pk = heavy_q[i];
+    uint32_t u = static_cast<uint32_t>(pk >> 32);
+    uint32_t v = static_cast<uint32_t>(pk);
+    uint64_t ub = goff[u];
+    uint32_t un = static_cast<uint32_t>(goff[u + 1] - ub);
+    uint64_t vb = goff[v];
+    uint32_t vn = static_cast<uint32_t>(goff[v + 1] - vb);
+    uint64_t eb, hb_big, hcap_end;
+    uint32_t en;
+    if (vn < un) {
+      eb = vb;
+      en = vn;
+      hb_big = hoff[u];
+      hcap_end = hoff[u + 1];
+    } else {
+      eb = ub;
+      en = un;
+      hb_big = hoff[v];
+      hcap_end = hoff[v + 1];
+    }
+    const uint64_t mask = (hcap_end - hb_big) - 1;
+    const uint32_t* t = htab + hb_big;
+    unsigned long long hits = 0;
+    for (uint32_t e = lane; e < en; e += kWave) {
+      uint32_t w = gdst[eb + e];
+      if (lcc_probe(t, mask, w)) {
+        ++hits;
+        atomicAdd(&T[w], 1ull);
+      }
+    }
+#pragma unroll
+    for (int d = 32; d > 0; d >>= 1)
+      hits += __shfl_down(hits, d, 64);
+    if (lane == 0 && hits) {
+      atomicAdd(&T[u], hits);
+      atomicAdd(&T[v], hits);
+    }
   }
 }
 
@@ -2797,6 +2940,7 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
   const uint32_t* dst2 = nullptr;
 
   // graph-sized working set (freed on return)
+  uint64_t oriented_total = 0;
   DeviceBuffer<uint32_t> Dv;                 // distinct degree, global idx
   DeviceBuffer<uint32_t> gcnt;               // oriented count, global idx
   DeviceBuffer<uint64_t> goff;               // global oriented CSR offsets
@@ -2896,6 +3040,7 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
     goff.resize(static_cast<size_t>(nv_pad) + 1);
     uint64_t g_total =
         exclusive_scan(gcnt.data(), goff.data(), nv_pad, s, I.scan);
+    oriented_total = g_total;
     gdst.resize(g_total ? g_total : 1);
     if (owned)
       lcc_compact_kernel<<<grid_for(static_cast<size_t>(owned) * kWave),
@@ -2975,14 +3120,48 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
                         kBlock, 0, s>>>(goff.data(), gdst.data(),
                                         hoff.data(), nv_pad, htab.data());
 
-  // triangle counting
+  // triangle counting: light edges inline, heavy edges deferred to a
+  // wave-per-edge pass (power-law tail otherwise stalls single lanes)
   Tcnt.resize(nv_pad);
   Tcnt.zero(s);
+  constexpr uint32_t kHeavyThresh = 96;
+  DeviceBuffer<unsigned long long> heavy_q;
+  DeviceBuffer<unsigned long long> heavy_n(1);
+  heavy_n.zero(s);
+  {
+    // worst case: every local oriented edge is heavy
+    uint64_t local_oedges = 0;
+    HIP_CHECK(hipMemcpyAsync(&local_oedges,
+                             goff.data() + std::min<uint64_t>(
+                                 static_cast<uint64_t>(g.v_end), nv_pad),
+                             8, hipMemcpyDeviceToHost, s));
+    uint64_t base_off = 0;
+    HIP_CHECK(hipMemcpyAsync(&base_off, goff.data() + g.v_begin, 8,
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    heavy_q.resize(std::max<uint64_t>(local_oedges - base_off, 1));
+  }
   if (owned)
     lcc_triangle_kernel<<<grid_for(static_cast<size_t>(owned) * kWave),
                           kBlock, 0, s>>>(goff.data(), gdst.data(),
                                           hoff.data(), htab.data(), owned,
-                                          g.v_begin, Tcnt.data());
+                                          g.v_begin, Tcnt.data(),
+                                          kHeavyThresh, heavy_q.data(),
+                                          heavy_n.data());
+  {
+    unsigned long long hn = 0;
+    HIP_CHECK(hipMemcpyAsync(&hn, heavy_n.data(), 8, hipMemcpyDeviceToHost,
+                             s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    if (getenv("GRAPEHIP_DEBUG"))
+      fprintf(stderr, "[lcc] oriented_total=%llu heavy_edges=%llu\n",
+              static_cast<unsigned long long>(oriented_total),
+              static_cast<unsigned long long>(hn));
+    if (hn)
+      lcc_triangle_heavy_kernel<<<grid_for(hn * kWave), kBlock, 0, s>>>(
+          goff.data(), gdst.data(), hoff.data(), htab.data(),
+          heavy_q.data(), hn, Tcnt.data());
+  }
   if (multi)
     NCCL_CHECK(ncclAllReduce(Tcnt.data(), Tcnt.data(), nv_pad, ncclUint64,
                              ncclSum, I.nccl, s));
