@@ -393,14 +393,15 @@ PYBIND11_MODULE(_core, m) {
            py::arg("prefix"))
       .def("load_synthetic",
            [](PyEngine& eng, uint64_t nv, uint64_t ne, uint64_t seed,
-              bool directed, bool weighted, double a, double b, double c) {
+              bool directed, bool weighted, bool build_in_csr, double a,
+              double b, double c) {
 #ifdef GRAPEHIP_WITH_HIP
              if (!eng.use_gpu)
                throw std::runtime_error("load_synthetic requires gpu=True");
              auto g = std::make_shared<PyGraph>();
              py::gil_scoped_release rel;
              g->dev = eng.gpu->gen_synthetic(nv, ne, seed, directed, weighted,
-                                             false, a, b, c);
+                                             build_in_csr, a, b, c);
              return g;
 #else
              throw std::runtime_error("built without HIP");
@@ -408,6 +409,7 @@ PYBIND11_MODULE(_core, m) {
            },
            py::arg("num_vertices"), py::arg("num_edges"), py::arg("seed") = 42,
            py::arg("directed") = false, py::arg("weighted") = false,
+           py::arg("build_in_csr") = false,
            py::arg("a") = 0.57, py::arg("b") = 0.19, py::arg("c") = 0.19)
       .def("bfs",
            [](PyEngine& eng, PyGraph& g, int64_t source, bool values) {

@@ -250,8 +250,9 @@ __global__ void gen_edges_kernel(uint64_t ne, uint64_t seed, int scale,
                                  uint32_t nv, uint32_t t_a, uint32_t t_ab,
                                  uint32_t t_abc, uint32_t my_begin,
                                  uint32_t my_end, bool undirected,
-                                 bool weighted, uint32_t* out_src,
-                                 uint32_t* out_dst, float* out_w,
+                                 bool reverse, bool weighted,
+                                 uint32_t* out_src, uint32_t* out_dst,
+                                 float* out_w,
                                  unsigned long long* out_cnt) {
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * blockDim.x +
@@ -280,14 +281,22 @@ __global__ void gen_edges_kernel(uint64_t ne, uint64_t seed, int scale,
     if (weighted)
       w = static_cast<float>((h >> 16) & 0xFFFFFF) * (99.0f / 16777216.0f) +
           1.0f;
-    // materialize both orientations for undirected storage; keep owned src
-    if (s >= my_begin && s < my_end) {
+    // materialize both orientations for undirected storage; keep owned src.
+    // reverse=true keeps edges by OWNED DST as (d, s) pairs — the same
+    // deterministic edge stream builds the in-CSR of a directed graph.
+    if (!reverse && s >= my_begin && s < my_end) {
       uint64_t pos = atomicAdd(out_cnt, 1ull);
       out_src[pos] = s;
       out_dst[pos] = d;
       if (weighted) out_w[pos] = w;
     }
     if (undirected && d >= my_begin && d < my_end && d != s) {
+      uint64_t pos = atomicAdd(out_cnt, 1ull);
+      out_src[pos] = d;
+      out_dst[pos] = s;
+      if (weighted) out_w[pos] = w;
+    }
+    if (reverse && d >= my_begin && d < my_end) {
       uint64_t pos = atomicAdd(out_cnt, 1ull);
       out_src[pos] = d;
       out_dst[pos] = s;
@@ -745,7 +754,7 @@ __global__ void wcc_compress_kernel(uint32_t* parent, uint32_t n) {
 // Afforest-style neighbor sampling (Sutton et al.): hook only the first K
 // edges of each row — a cheap streaming pass that already collapses most
 // of a power-law graph into its giant component.
-__global__ void wcc_sample_kernel(DevGraphView g, int k_sample,
+__global__ void wcc_sample_kernel(DevGraphView g, int which,
                                   uint32_t* parent, int* changed) {
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   const uint32_t rows = g.owned();
@@ -753,21 +762,22 @@ __global__ void wcc_sample_kernel(DevGraphView g, int k_sample,
   for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < rows;
        r += stride) {
     uint64_t b = g.oe_off[r], e = g.oe_off[r + 1];
-    uint64_t hi = b + k_sample < e ? b + k_sample : e;
-    for (uint64_t i = b; i < hi; ++i)
-      op(g.v_begin + r, g.oe_dst[i], 1.0f);
+    uint64_t i = b + which;
+    if (i < e) op(g.v_begin + r, g.oe_dst[i], 1.0f);
   }
 }
 
 // mark owned rows NOT in the giant component (bitmap -> frontier queue);
 // their remaining edges are the only ones the full pass must visit
 __global__ void wcc_mark_rest_kernel(const uint32_t* __restrict__ parent,
+                                     const uint64_t* __restrict__ off,
                                      uint32_t owned, uint32_t v_begin,
                                      uint32_t giant, DevBitmap bm) {
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < owned;
        r += stride)
-    if (parent[v_begin + r] != giant) bm.set_once(r);
+    if (parent[v_begin + r] != giant && off[r + 1] != off[r])
+      bm.set_once(r);
 }
 
 }  // namespace grapehip
@@ -1157,7 +1167,7 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
             (unsigned long)ne, (unsigned long)est, scale);
   gen_edges_kernel<<<kMaxGrid, kBlock, 0, s>>>(
       ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, g->v_begin, g->v_end,
-      !directed, weighted, e_src.data(), e_dst.data(),
+      !directed, false, weighted, e_src.data(), e_dst.data(),
       weighted ? e_w.data() : nullptr, cnt.data());
   unsigned long long n_local = 0;
   HIP_CHECK(hipMemcpyAsync(&n_local, cnt.data(), 8, hipMemcpyDeviceToHost, s));
@@ -1169,13 +1179,21 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
   build_csr_from_coo(e_src, e_dst, e_w, n_local, g->v_begin, owned, weighted,
                      g->oe_off, g->oe_dst, g->oe_w, s, impl_->scan);
   if (directed && build_in_csr) {
-    // regenerate reversed edges owned by dst
+    // regenerate the SAME edge stream, keeping edges whose dst is owned,
+    // reversed — builds the in-CSR without materializing the global list
     cnt.zero(s);
     gen_edges_kernel<<<kMaxGrid, kBlock, 0, s>>>(
-        ne, seed ^ 0x1234567ULL, scale, g->nv_global, t_a, t_ab, t_abc, 0, 0,
-        false, weighted, e_src.data(), e_dst.data(),
+        ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, g->v_begin,
+        g->v_end, false, true, weighted, e_src.data(), e_dst.data(),
         weighted ? e_w.data() : nullptr, cnt.data());
-    throw std::runtime_error("gen_synthetic: directed in-CSR TODO");
+    unsigned long long n_in = 0;
+    HIP_CHECK(hipMemcpyAsync(&n_in, cnt.data(), 8, hipMemcpyDeviceToHost,
+                             s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    if (n_in > est) throw std::runtime_error("gen_synthetic: in overflow");
+    build_csr_from_coo(e_src, e_dst, e_w, n_in, g->v_begin, owned, weighted,
+                       g->ie_off, g->ie_dst, g->ie_w, s, impl_->scan);
+    g->has_in = true;
   }
   e_src.free();
   e_dst.free();
@@ -1787,13 +1805,17 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
   rest_bm.zero(s);
   for (;;) {
     int local_changed_any = 0;
-    // 1) sample pass: first K edges per row (streaming), then compress
+    // 1) Afforest sampling: hook edge i per row in pass i, compressing
+    //    between passes (linking via compressed roots collapses the giant
+    //    component far better than hooking both edges in one pass)
     d_changed.zero(s);
-    if (owned)
-      wcc_sample_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-          view, kSample, parent.data(), d_changed.data());
-    wcc_compress_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(parent.data(),
-                                                            nv_pad);
+    for (int which = 0; which < kSample; ++which) {
+      if (owned)
+        wcc_sample_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+            view, which, parent.data(), d_changed.data());
+      wcc_compress_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(parent.data(),
+                                                              nv_pad);
+    }
     ++rounds;
     if (multi) {
       int ch0 = 0;
@@ -1830,7 +1852,7 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
       rest_bm.zero(s);
       if (owned)
         wcc_mark_rest_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-            parent.data(), owned, g.v_begin, giant,
+            parent.data(), view.oe_off, owned, g.v_begin, giant,
             DevBitmap{rest_bm.data()});
       uint64_t nrest = compact_frontier(I, rest_bm.data(), owned, g.v_begin,
                                         rest_q.data(), s);

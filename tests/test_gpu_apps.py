@@ -169,3 +169,14 @@ def test_gpu_lcc_hub(eng):
     _, vals = by_oid(eng.lcc(g))
     expect = lcc_oracle(nv, src, dst, directed=False)
     assert np.allclose(vals, expect, rtol=1e-12)
+
+
+def test_gpu_synthetic_directed_incsr(eng):
+    # directed synthetic with in-CSR: CDLP runs (needs in+out multiset) and
+    # in/out edge totals agree
+    g = eng.load_synthetic(num_vertices=50000, num_edges=400000, seed=3,
+                           directed=True, weighted=True, build_in_csr=True)
+    r = eng.cdlp(g, 3)
+    assert len(r["values"]) == 50000
+    r2 = eng.pagerank(g, 0.85, 5)
+    assert abs(r2["values"].sum() - 1.0) < 1e-6
