@@ -25,6 +25,7 @@
 #include "apps/wcc.hpp"
 #include "core/fragment.hpp"
 #include "core/serialize.hpp"
+#include "core/vertexcut.hpp"
 #include "core/message_manager.hpp"
 #include "core/net.hpp"
 
@@ -218,6 +219,13 @@ py::dict gpu_dict(const GpuRunResult& r, const DeviceGraph& g, bool is_i64,
 
 PYBIND11_MODULE(_core, m) {
   m.doc() = "grapehip core engine";
+
+  py::class_<VertexcutFragment, std::shared_ptr<VertexcutFragment>>(
+      m, "VertexcutGraph")
+      .def_property_readonly("num_vertices",
+                             &VertexcutFragment::num_vertices)
+      .def_property_readonly("num_edges", &VertexcutFragment::total_edges)
+      .def_property_readonly("local_edges", &VertexcutFragment::local_edges);
 
   py::class_<PyGraph, std::shared_ptr<PyGraph>>(m, "Graph")
       .def_property_readonly("num_vertices",
@@ -639,6 +647,56 @@ PYBIND11_MODULE(_core, m) {
              return meta;
            },
            py::arg("graph"))
+      .def("load_vertexcut",
+           [](PyEngine& eng, arr_i64 src, arr_i64 dst, int64_t num_vertices,
+              int bucket_num) {
+             size_t n = src.size();
+             std::vector<uint32_t> s32(n), d32(n);
+             auto sp = src.unchecked<1>();
+             auto dp = dst.unchecked<1>();
+             for (size_t i = 0; i < n; ++i) {
+               s32[i] = static_cast<uint32_t>(sp(i));
+               d32[i] = static_cast<uint32_t>(dp(i));
+             }
+             py::gil_scoped_release rel;
+             return std::shared_ptr<VertexcutFragment>(
+                 VertexcutFragment::Build(eng.c(), eng.rank, eng.world,
+                                          static_cast<uint64_t>(num_vertices),
+                                          s32, d32, bucket_num));
+           },
+           py::arg("src"), py::arg("dst"), py::arg("num_vertices"),
+           py::arg("bucket_num") = 8)
+      .def("pagerank_vc",
+           [](PyEngine& eng, VertexcutFragment& g, double damping,
+              int iters) {
+             std::vector<double> r;
+             double t0, t1;
+             {
+               py::gil_scoped_release rel;
+               if (eng.c()) eng.c()->barrier();
+               t0 = now_s();
+               r = pagerank_vc(g, eng.c(), damping, iters);
+               if (eng.c()) eng.c()->barrier();
+               t1 = now_s();
+             }
+             // result is fully replicated; report this rank's segment
+             uint64_t b = g.segments()[eng.rank];
+             uint64_t e = g.segments()[eng.rank + 1];
+             py::dict out;
+             out["seconds"] = t1 - t0;
+             out["rounds"] = iters;
+             py::array_t<int64_t> oids(e - b);
+             py::array_t<double> vals(e - b);
+             for (uint64_t v = b; v < e; ++v) {
+               oids.mutable_data()[v - b] = static_cast<int64_t>(v);
+               vals.mutable_data()[v - b] = r[v];
+             }
+             out["oids"] = oids;
+             out["values"] = vals;
+             return out;
+           },
+           py::arg("graph"), py::arg("damping") = 0.85,
+           py::arg("iters") = 10)
       .def("sample",
            [](PyEngine& eng, PyGraph& g, arr_i64 starts, int hops,
               const std::string& strategy, int top_k, uint64_t seed) {

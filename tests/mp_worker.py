@@ -64,6 +64,10 @@ def main():
         res = eng.kcore(g, cfg.get("k", 3))
     elif app == "core_decomposition":
         res = eng.core_decomposition(g)
+    elif app == "pagerank_vc":
+        gvc = eng.load_vertexcut(src[sl], dst[sl],
+                                 num_vertices=cfg["num_v"])
+        res = eng.pagerank_vc(gvc, 0.85, 10)
     elif app == "sample":
         r = eng.sample(g, np.arange(cfg.get("n_walks", 20), dtype=np.int64),
                        hops=cfg.get("hops", 3), seed=5)

@@ -176,3 +176,12 @@ def test_sampler_mp(world, free_port, tmp_path):
             if a == -1 or b == -1:
                 continue
             assert b in adj.get(a, set()), (wid, h)
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_pagerank_vc_mp(world, free_port, tmp_path):
+    cfg = dict(BASE, app="pagerank_vc", num_v=400, num_e=2600)
+    oids, vals = run_world(world, cfg, free_port, tmp_path)
+    src, dst, _ = graph_arrays(cfg)
+    assert np.allclose(vals, pagerank_oracle(cfg["num_v"], src, dst),
+                       rtol=1e-9)
