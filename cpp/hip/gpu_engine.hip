@@ -660,24 +660,29 @@ __global__ void sssp_repart_kernel(const uint32_t* __restrict__ far_in,
 }
 
 struct PrPushOp {
-  const double* __restrict__ contrib;  // per owned row
+  const float* __restrict__ contrib;  // per owned row
   double* acc;
   uint32_t v_begin;
   __device__ __forceinline__ void operator()(uint32_t u, uint32_t d,
                                              float) const {
-    unsafeAtomicAdd(&acc[d], contrib[u - v_begin]);
+    unsafeAtomicAdd(&acc[d], static_cast<double>(contrib[u - v_begin]));
   }
 };
 
+// fp32 contributions (the reference's GPU PageRank precision,
+// cuda/pagerank/pagerank.h:27-35 rank_t=float): halves the random-gather
+// bytes of the pull pass; the rank/accumulator side stays fp64.
 __global__ void pr_contrib_kernel(const double* __restrict__ rank,
                                   const uint64_t* __restrict__ off,
                                   uint32_t owned, uint32_t v_begin,
-                                  double* __restrict__ contrib) {
+                                  float* __restrict__ contrib) {
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < owned;
        r += stride) {
     uint64_t deg = off[r + 1] - off[r];
-    contrib[r] = deg ? rank[v_begin + r] / static_cast<double>(deg) : 0.0;
+    contrib[r] = deg ? static_cast<float>(rank[v_begin + r] /
+                                          static_cast<double>(deg))
+                     : 0.0f;
   }
 }
 
@@ -836,7 +841,7 @@ __global__ void bucket_rows_kernel(const uint64_t* __restrict__ off,
 // thread per row
 __global__ void pr_pull_small_kernel(const uint64_t* __restrict__ off,
                                      const uint32_t* __restrict__ dst,
-                                     const double* __restrict__ contrib,
+                                     const float* __restrict__ contrib,
                                      const uint32_t* __restrict__ rows,
                                      uint64_t nrows, uint32_t v_begin,
                                      double* __restrict__ acc) {
@@ -855,7 +860,7 @@ __global__ void pr_pull_small_kernel(const uint64_t* __restrict__ off,
 // wave per row (4 waves per 256-block)
 __global__ void pr_pull_mid_kernel(const uint64_t* __restrict__ off,
                                    const uint32_t* __restrict__ dst,
-                                   const double* __restrict__ contrib,
+                                   const float* __restrict__ contrib,
                                    const uint32_t* __restrict__ rows,
                                    uint64_t nrows, uint32_t v_begin,
                                    double* __restrict__ acc) {
@@ -878,7 +883,7 @@ __global__ void pr_pull_mid_kernel(const uint64_t* __restrict__ off,
 // block per row
 __global__ void pr_pull_large_kernel(const uint64_t* __restrict__ off,
                                      const uint32_t* __restrict__ dst,
-                                     const double* __restrict__ contrib,
+                                     const float* __restrict__ contrib,
                                      const uint32_t* __restrict__ rows,
                                      uint64_t nrows, uint32_t v_begin,
                                      double* __restrict__ acc) {
@@ -1687,7 +1692,8 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
       (!g.directed ? g.oe_dst.data() : g.ie_dst.data());
   if (pull) ensure_buckets(g, s);
 
-  DeviceBuffer<double> rank_arr(nv_pad), acc(nv_pad), contrib(nv_pad);
+  DeviceBuffer<double> rank_arr(nv_pad), acc(nv_pad);
+  DeviceBuffer<float> contrib(nv_pad);
   DeviceBuffer<double> d_dangling(1);
   DevGraphView view = make_view(g, rank_, world_);
 
@@ -1724,7 +1730,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
       if (multi)
         NCCL_CHECK(ncclAllGather(contrib.data() + static_cast<uint64_t>(
                                                       rank_) * slice,
-                                 contrib.data(), slice, ncclDouble, I.nccl,
+                                 contrib.data(), slice, ncclFloat, I.nccl,
                                  s));
       if (g.n_small)
         pr_pull_small_kernel<<<grid_for(g.n_small), kBlock, 0, s>>>(
@@ -2112,7 +2118,9 @@ __global__ void cdlp_heavy_cap_kernel(const uint64_t* __restrict__ off1,
   }
 }
 
-// block-per-row global hash (heavy rows)
+// block-per-row global hash (heavy rows). Slots are epoch-tagged: the
+// count word packs (epoch << 24 | count), so tables never need clearing
+// between iterations — a stale slot (old epoch) reads as empty.
 __global__ void cdlp_large_kernel(const uint64_t* __restrict__ off1,
                                   const uint32_t* __restrict__ dst1,
                                   const uint64_t* __restrict__ off2,
@@ -2121,10 +2129,16 @@ __global__ void cdlp_large_kernel(const uint64_t* __restrict__ off1,
                                   const uint32_t* __restrict__ rows,
                                   uint64_t nrows,
                                   const uint64_t* __restrict__ tbl_off,
-                                  uint32_t* __restrict__ tbl_lab,
-                                  uint32_t* __restrict__ tbl_cnt,
-                                  uint32_t v_begin,
+                                  unsigned long long* __restrict__ tbl,
+                                  uint32_t epoch, uint32_t v_begin,
                                   uint32_t* __restrict__ next) {
+  // slot u64 = (epoch:16 | label:32 hashed into low? ) — layout:
+  // high 16 bits epoch, next 32 bits label, low 16 bits... counts can
+  // exceed 16 bits, so: slot = (epoch<<48) | (label<<16) is unsafe.
+  // Instead: two u32 halves in one u64: hi = label, lo = (epoch<<24|cnt)
+  // with cnt capped at 2^24 (heavy rows cap the count at deg < 2^24 per
+  // label; larger multiplicities clamp — mode selection unaffected since
+  // clamp only at astronomically heavy rows).
   for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
     uint32_t r = rows[i];
     uint64_t b1 = off1[r], e1 = off1[r + 1];
@@ -2138,15 +2152,29 @@ __global__ void cdlp_large_kernel(const uint64_t* __restrict__ off1,
     uint64_t tb = tbl_off[i];
     const uint64_t cap = tbl_off[i + 1] - tb;
     const uint64_t mask = cap - 1;
-    uint32_t* tl = tbl_lab + tb;
-    uint32_t* tc = tbl_cnt + tb;
+    unsigned long long* t = tbl + tb;
+    const uint32_t etag = epoch << 24;
     for (uint64_t k = threadIdx.x; k < deg; k += blockDim.x) {
       uint32_t l = k < d1 ? lab[dst1[b1 + k]] : lab[dst2[b2 + (k - d1)]];
       uint64_t idx = cdlp_hash(l) & mask;
       for (;;) {
-        uint32_t old = atomicCAS(&tl[idx], kCdlpEmpty, l);
-        if (old == kCdlpEmpty || old == l) {
-          atomicAdd(&tc[idx], 1u);
+        unsigned long long cur = t[idx];
+        uint32_t cur_lab = static_cast<uint32_t>(cur >> 32);
+        uint32_t cur_lo = static_cast<uint32_t>(cur);
+        bool stale = (cur_lo >> 24) != epoch;
+        if (stale) {
+          // try to claim the slot for this epoch with count 1
+          unsigned long long want =
+              (static_cast<unsigned long long>(l) << 32) | (etag | 1u);
+          unsigned long long old = atomicCAS(&t[idx], cur, want);
+          if (old == cur) break;      // claimed
+          cur = old;                  // somebody else raced; re-inspect
+          cur_lab = static_cast<uint32_t>(cur >> 32);
+          cur_lo = static_cast<uint32_t>(cur);
+          if ((cur_lo >> 24) != epoch) continue;  // still stale: retry slot
+        }
+        if (cur_lab == l) {
+          if ((cur_lo & 0xFFFFFF) != 0xFFFFFF) atomicAdd(&t[idx], 1ull);
           break;
         }
         idx = (idx + 1) & mask;
@@ -2155,10 +2183,12 @@ __global__ void cdlp_large_kernel(const uint64_t* __restrict__ off1,
     __syncthreads();
     uint64_t key = 0;
     for (uint64_t k = threadIdx.x; k < cap; k += blockDim.x) {
-      uint32_t c = tc[k];
-      if (c) {
-        uint64_t cand = (static_cast<uint64_t>(c) << 32) |
-                        static_cast<uint32_t>(~tl[k]);
+      unsigned long long cur = t[k];
+      uint32_t lo = static_cast<uint32_t>(cur);
+      if ((lo >> 24) == epoch && (lo & 0xFFFFFF)) {
+        uint64_t cand =
+            (static_cast<uint64_t>(lo & 0xFFFFFF) << 32) |
+            static_cast<uint32_t>(~static_cast<uint32_t>(cur >> 32));
         if (cand > key) key = cand;
       }
     }
@@ -2203,10 +2233,11 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
   auto hc = cnts.download(s);
   uint64_t n_small = hc[0], n_mid = hc[1], n_large = hc[2];
 
-  // heavy-row global hash pool
+  // heavy-row global hash pool (epoch-tagged u64 slots: no per-iteration
+  // clears; epoch 0 == the zeroed virgin state)
   DeviceBuffer<uint32_t> heavy_caps;
   DeviceBuffer<uint64_t> heavy_off;
-  DeviceBuffer<uint32_t> tbl_lab, tbl_cnt;
+  DeviceBuffer<unsigned long long> heavy_tbl;
   uint64_t heavy_total = 0;
   if (n_large) {
     heavy_caps.resize(n_large);
@@ -2215,8 +2246,8 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
         off1, off2, t_large.data(), n_large, heavy_caps.data());
     heavy_total = exclusive_scan(heavy_caps.data(), heavy_off.data(),
                                  n_large, s, I.scan);
-    tbl_lab.resize(heavy_total);
-    tbl_cnt.resize(heavy_total);
+    heavy_tbl.resize(heavy_total);
+    heavy_tbl.zero(s);
   }
 
   DeviceBuffer<uint32_t> lab(nv_pad);
@@ -2237,14 +2268,13 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
       cdlp_mid_kernel<<<std::min<int>(n_mid, kMaxGrid), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, lab.data(), t_mid.data(), n_mid, g.v_begin,
           next.data());
-    if (n_large) {
-      fill(tbl_lab.data(), kCdlpEmpty, heavy_total, s);
-      tbl_cnt.zero(s);
+    if (n_large && it > 0 && it % 254 == 0)
+      heavy_tbl.zero(s);  // epoch tag is 8-bit; re-zero on wrap
+    if (n_large)
       cdlp_large_kernel<<<std::min<int>(n_large, kMaxGrid), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, lab.data(), t_large.data(), n_large,
-          heavy_off.data(), tbl_lab.data(), tbl_cnt.data(), g.v_begin,
-          next.data());
-    }
+          heavy_off.data(), heavy_tbl.data(),
+          static_cast<uint32_t>((it % 254) + 1), g.v_begin, next.data());
     if (owned)
       cdlp_commit_kernel<<<grid_for(owned), kBlock, 0, s>>>(
           next.data(), owned, g.v_begin, lab.data());

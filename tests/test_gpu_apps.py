@@ -70,8 +70,9 @@ def test_gpu_pagerank(eng):
     g = eng.load_edges(src, dst, directed=True, num_vertices=5000)
     _, vals = by_oid(eng.pagerank(g, 0.85, 10))
     expect = pagerank_oracle(5000, src, dst, 0.85, 10, directed=True)
-    assert np.allclose(vals, expect, rtol=1e-9)
-    assert abs(vals.sum() - 1.0) < 1e-9
+    # fp32 contributions (the reference's GPU precision) vs fp64 oracle
+    assert np.allclose(vals, expect, rtol=3e-5, atol=1e-12)
+    assert abs(vals.sum() - 1.0) < 1e-5
 
 
 def test_gpu_wcc(eng):
@@ -94,7 +95,7 @@ def test_gpu_synthetic(eng):
     reached = d < INT64_MAX
     assert reached.sum() > 1000  # hub-connected RMAT core
     r2 = eng.pagerank(g, 0.85, 5)
-    assert abs(r2["values"].sum() - 1.0) < 1e-6
+    assert abs(r2["values"].sum() - 1.0) < 1e-5
     r3 = eng.sssp(g, 0)
     # any BFS-reachable vertex must be SSSP-reachable and dist >= depth
     sd = r3["values"]
@@ -179,4 +180,4 @@ def test_gpu_synthetic_directed_incsr(eng):
     r = eng.cdlp(g, 3)
     assert len(r["values"]) == 50000
     r2 = eng.pagerank(g, 0.85, 5)
-    assert abs(r2["values"].sum() - 1.0) < 1e-6
+    assert abs(r2["values"].sum() - 1.0) < 1e-5
