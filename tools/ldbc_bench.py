@@ -28,6 +28,10 @@ DATASETS = {
     "datagen-9_0-fb": (404_817_003, 1_010_447_118),
     "graph500-26": (67_108_864, 1_073_741_824),
     "p2p-31-shaped": (6_300, 148_000),
+    # reference GPU-table graphs (Performance.md:80-97, 8x V100)
+    "soc-LiveJournal1-shaped": (4_847_571, 68_993_773),
+    "soc-orkut-shaped": (2_997_166, 212_698_418),
+    "soc-twitter-2010-shaped": (21_297_772, 530_051_090),
 }
 
 ALGOS = ("bfs", "sssp", "pagerank", "wcc", "cdlp", "lcc")
