@@ -78,3 +78,38 @@ def test_cli_serialize_roundtrip(tmp_path):
     a = open(res1).read()
     b = open(out2 / "result_frag_0").read()
     assert a == b
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_cli_multirank_bfs_golden(tmp_path, world):
+    # torchrun-style env; each rank partial-reads its file slice
+    out = tmp_path / "out"
+    procs = []
+    port = 29755
+    for rank in range(world):
+        env = dict(os.environ, PYTHONPATH=str(REPO), RANK=str(rank),
+                   WORLD_SIZE=str(world), MASTER_ADDR="127.0.0.1",
+                   MASTER_PORT=str(port - 17))
+        cmd = [sys.executable, "-m", "grapehip.run_app",
+               "--application", "bfs", "--bfs_source", "6",
+               "--efile", str(DATASET / "p2p-31.e"),
+               "--vfile", str(DATASET / "p2p-31.v"),
+               "--out_prefix", str(out)]
+        procs.append(subprocess.Popen(cmd, cwd=REPO, env=env,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for p in procs:
+        o, _ = p.communicate(timeout=300)
+        assert p.returncode == 0, o.decode()
+    oids, vals = [], []
+    for rank in range(world):
+        o, v = load_pairs(out / ("result_frag_%d" % rank))
+        oids.extend(o)
+        vals.extend(int(x) for x in v)
+    oids = np.array(oids)
+    vals = np.array(vals)
+    g_oids, g_vals = load_pairs(DATASET / "p2p-31-BFS")
+    order, gorder = np.argsort(oids), np.argsort(g_oids)
+    assert np.array_equal(oids[order], g_oids[gorder])
+    exp = np.array([int(v) for v in g_vals], dtype=np.int64)[gorder]
+    assert np.array_equal(vals[order], exp)
