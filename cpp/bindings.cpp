@@ -125,26 +125,34 @@ std::shared_ptr<PyGraph> load_edges(
   if (vertex_oids.has_value()) {
     const int64_t* po = vertex_oids->data();
     size_t nv = vertex_oids->size();
-    std::vector<std::vector<oid_t>> bins(eng.world);
-    for (size_t i = 0; i < nv; ++i)
-      bins[hash_oid(po[i]) % eng.world].push_back(po[i]);
-    std::vector<std::string> send(eng.world);
-    for (int f = 0; f < eng.world; ++f)
-      send[f].assign(reinterpret_cast<const char*>(bins[f].data()),
-                     bins[f].size() * sizeof(oid_t));
-    std::vector<std::string> recv =
-        eng.c() ? eng.c()->exchange_all(send) : std::move(send);
     std::vector<oid_t> owned;
-    for (auto& blob : recv) {
-      size_t m = blob.size() / sizeof(oid_t);
-      const oid_t* p = reinterpret_cast<const oid_t*>(blob.data());
-      owned.insert(owned.end(), p, p + m);
+    if (partitioner == "map") {
+      // MapPartitioner semantics (reference partitioner.h:103): ownership
+      // = the rank that supplied the oid (e.g. a rebalanced range)
+      owned.assign(po, po + nv);
+      std::sort(owned.begin(), owned.end());
+      owned.erase(std::unique(owned.begin(), owned.end()), owned.end());
+    } else {
+      // HashPartitioner: shuffle oids to hash owners
+      std::vector<std::vector<oid_t>> bins(eng.world);
+      for (size_t i = 0; i < nv; ++i)
+        bins[hash_oid(po[i]) % eng.world].push_back(po[i]);
+      std::vector<std::string> send(eng.world);
+      for (int f = 0; f < eng.world; ++f)
+        send[f].assign(reinterpret_cast<const char*>(bins[f].data()),
+                       bins[f].size() * sizeof(oid_t));
+      std::vector<std::string> recv =
+          eng.c() ? eng.c()->exchange_all(send) : std::move(send);
+      for (auto& blob : recv) {
+        size_t m = blob.size() / sizeof(oid_t);
+        const oid_t* p = reinterpret_cast<const oid_t*>(blob.data());
+        owned.insert(owned.end(), p, p + m);
+      }
+      std::sort(owned.begin(), owned.end());
+      owned.erase(std::unique(owned.begin(), owned.end()), owned.end());
     }
-    std::sort(owned.begin(), owned.end());
-    owned.erase(std::unique(owned.begin(), owned.end()), owned.end());
     g->vm->init_hashmap(eng.world, PartitionerKind::kHash, eng.c(),
                         std::move(owned));
-    (void)partitioner;
   } else {
     if (num_vertices <= 0)
       throw std::runtime_error("num_vertices required for identity mapping");
@@ -280,6 +288,26 @@ PYBIND11_MODULE(_core, m) {
              if (e.gpu) e.gpu->device_sync();
 #endif
            })
+      .def("_exchange_all",
+           [](PyEngine& eng, std::vector<py::bytes> blobs) {
+             if (static_cast<int>(blobs.size()) != eng.world)
+               throw std::runtime_error(
+                   "_exchange_all needs one blob per rank");
+             std::vector<std::string> send(eng.world);
+             for (int f = 0; f < eng.world; ++f)
+               send[f] = std::string(blobs[f]);
+             std::vector<std::string> recv;
+             {
+               py::gil_scoped_release rel;
+               recv = eng.c() ? eng.c()->exchange_all(send)
+                              : std::move(send);
+             }
+             std::vector<py::bytes> out;
+             out.reserve(recv.size());
+             for (auto& r : recv) out.emplace_back(r);
+             return out;
+           },
+           py::arg("blobs"))
       .def("load_edges", &load_edges, py::arg("src"), py::arg("dst"),
            py::arg("weights") = std::nullopt, py::arg("directed") = false,
            py::arg("num_vertices") = -1, py::arg("vertex_oids") = std::nullopt,

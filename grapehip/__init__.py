@@ -13,7 +13,8 @@ import os
 from grapehip._core import Engine, Graph, WITH_HIP  # noqa: F401
 
 from grapehip.generate import rmat_edges, uniform_edges  # noqa: F401
-from grapehip.io import read_ldbc_edges  # noqa: F401
+from grapehip.io import (read_ldbc_edges, read_ldbc_vertices,  # noqa: F401
+                         load_string_edges, rebalance_partition)  # noqa: F401
 
 __version__ = "0.1.0"
 

@@ -32,7 +32,12 @@ def main():
     # each rank takes a strided slice of the same global edge list
     sl = slice(rank, None, world)
     kw = {}
-    if cfg.get("vertex_oids"):
+    if cfg.get("rebalance"):
+        oids = grapehip.rebalance_partition(eng, src[sl], dst[sl],
+                                            cfg["num_v"])
+        kw["vertex_oids"] = oids
+        kw["partitioner"] = "map"
+    elif cfg.get("vertex_oids"):
         # split vertex list round-robin too
         all_oids = np.arange(cfg["num_v"], dtype=np.int64) * 3 + 1
         kw["vertex_oids"] = all_oids[sl]

@@ -185,3 +185,13 @@ def test_pagerank_vc_mp(world, free_port, tmp_path):
     src, dst, _ = graph_arrays(cfg)
     assert np.allclose(vals, pagerank_oracle(cfg["num_v"], src, dst),
                        rtol=1e-9)
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_bfs_mp_rebalanced(world, free_port, tmp_path):
+    # degree-balanced contiguous map partition (reference rebalancer.h)
+    cfg = dict(BASE, app="bfs", rebalance=True)
+    oids, vals = run_world(world, cfg, free_port, tmp_path)
+    src, dst, _ = graph_arrays(cfg)
+    assert np.array_equal(np.sort(oids), np.arange(cfg["num_v"]))
+    assert np.array_equal(vals, bfs_oracle(cfg["num_v"], src, dst, 3))
