@@ -14,6 +14,7 @@
 #include <unordered_set>
 
 #include "apps/bc.hpp"
+#include "apps/auto_app.hpp"
 #include "apps/bfs.hpp"
 #include "apps/cdlp.hpp"
 #include "apps/kclique.hpp"
@@ -781,6 +782,25 @@ PYBIND11_MODULE(_core, m) {
            py::arg("graph"), py::arg("starts"), py::arg("hops") = 2,
            py::arg("strategy") = "random", py::arg("top_k") = 4,
            py::arg("seed") = 7)
+      .def("sssp_auto",
+           [](PyEngine& eng, PyGraph& g, int64_t source) {
+             if (!g.frag)
+               throw std::runtime_error("sssp_auto needs a host fragment");
+             SSSPAutoApp app;
+             SSSPAutoContext ctx;
+             MessageManager mm;
+             mm.init(eng.c(), g.frag.get(), eng.n_threads);
+             ctx.init(*g.frag, source);
+             py::dict meta = run_timed(
+                 eng, [&] { return RunWorker(app, ctx, *g.frag, mm); });
+             std::vector<double> vals(g.frag->ivnum());
+             for (vid_t v = 0; v < g.frag->ivnum(); ++v)
+               vals[v] = ctx.dist[v];
+             meta["oids"] = inner_oids(*g.frag);
+             meta["values"] = to_np(vals);
+             return meta;
+           },
+           py::arg("graph"), py::arg("source") = 0)
       .def("kclique",
            [](PyEngine& eng, PyGraph& g, int k) {
              if (!g.frag)

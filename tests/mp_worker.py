@@ -51,7 +51,9 @@ def main():
                        build_in_csr=cfg["in_csr"], **kw)
 
     app = cfg["app"]
-    if app == "bfs":
+    if app == "sssp_auto":
+        res = eng.sssp_auto(g, cfg["source"])
+    elif app == "bfs":
         res = eng.bfs(g, cfg["source"])
     elif app == "sssp":
         res = eng.sssp(g, cfg["source"])

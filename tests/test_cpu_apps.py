@@ -192,3 +192,16 @@ def test_kclique(eng):
     for k in (3, 4, 5):
         r = eng.kclique(g, k)
         assert r["clique_count"] == kclique_oracle(120, src, dst, k), k
+
+
+def test_sssp_auto(eng):
+    # auto-app (sync-buffer) variant must agree with the explicit app
+    src, dst, w = random_graph(num_v=400, num_e=2500, seed=67, weighted=True)
+    g = eng.load_edges(src, dst, weights=w, directed=True, num_vertices=400)
+    r_auto = eng.sssp_auto(g, 3)
+    r_par = eng.sssp(g, 3)
+    o1, o2 = np.argsort(r_auto["oids"]), np.argsort(r_par["oids"])
+    a, b = r_auto["values"][o1], r_par["values"][o2]
+    finite = b < 1e300
+    assert np.allclose(a[finite], b[finite], rtol=1e-9)
+    assert (a[~finite] > 1e300).all()

@@ -195,3 +195,13 @@ def test_bfs_mp_rebalanced(world, free_port, tmp_path):
     src, dst, _ = graph_arrays(cfg)
     assert np.array_equal(np.sort(oids), np.arange(cfg["num_v"]))
     assert np.array_equal(vals, bfs_oracle(cfg["num_v"], src, dst, 3))
+
+
+@pytest.mark.parametrize("world", [2])
+def test_sssp_auto_mp(world, free_port, tmp_path):
+    cfg = dict(BASE, app="sssp_auto", weighted=True)
+    oids, vals = run_world(world, cfg, free_port, tmp_path)
+    src, dst, w = graph_arrays(cfg)
+    expect = sssp_oracle(cfg["num_v"], src, dst, w, 3)
+    finite = expect < 1e300
+    assert np.allclose(vals[finite], expect[finite], rtol=1e-9)
