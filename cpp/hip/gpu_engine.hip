@@ -2484,7 +2484,7 @@ __global__ void lcc_orient_small_kernel(
     const uint64_t* __restrict__ off2, const uint32_t* __restrict__ dst2,
     const uint32_t* __restrict__ rows, uint64_t nrows, uint32_t v_begin,
     const uint32_t* __restrict__ D, const uint64_t* __restrict__ ooff,
-    uint32_t* __restrict__ oadj, uint32_t* __restrict__ ocnt) {
+    uint32_t* __restrict__ oadj, uint32_t* __restrict__ ocnt, bool filter) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int wpb = kBlock / kWave;
@@ -2519,7 +2519,7 @@ __global__ void lcc_orient_small_kernel(
     }
     uint32_t dv = D[v];
     bool keep = active && ((first_mask >> lane) & 1) &&
-                lcc_keep(D[u], u, dv, v);
+                (!filter || lcc_keep(D[u], u, dv, v));
     unsigned long long keep_mask = __ballot(keep);
     uint32_t pos = __popcll(keep_mask & ((1ull << lane) - 1));
     uint64_t base = ooff[r];
@@ -2533,7 +2533,7 @@ __global__ void lcc_orient_mid_kernel(
     const uint64_t* __restrict__ off2, const uint32_t* __restrict__ dst2,
     const uint32_t* __restrict__ rows, uint64_t nrows, uint32_t v_begin,
     const uint32_t* __restrict__ D, const uint64_t* __restrict__ ooff,
-    uint32_t* __restrict__ oadj, uint32_t* __restrict__ ocnt) {
+    uint32_t* __restrict__ oadj, uint32_t* __restrict__ ocnt, bool filter) {
   __shared__ uint32_t s_lab[kLccLdsSlots];
   __shared__ uint32_t s_cursor;
   for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
@@ -2564,7 +2564,7 @@ __global__ void lcc_orient_mid_kernel(
       for (;;) {
         uint32_t old = atomicCAS(&s_lab[idx], kCdlpEmpty, u);
         if (old == kCdlpEmpty) {
-          if (lcc_keep(D[u], u, dv, v))
+          if (!filter || lcc_keep(D[u], u, dv, v))
             oadj[base + atomicAdd(&s_cursor, 1u)] = u;
           break;
         }
@@ -2585,7 +2585,7 @@ __global__ void lcc_orient_large_kernel(
     const uint64_t* __restrict__ tbl_off, uint32_t* __restrict__ tbl_lab,
     uint32_t v_begin, const uint32_t* __restrict__ D,
     const uint64_t* __restrict__ ooff, uint32_t* __restrict__ oadj,
-    uint32_t* __restrict__ ocnt) {
+    uint32_t* __restrict__ ocnt, bool filter) {
   __shared__ uint32_t s_cursor;
   for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
     uint32_t r = rows[i];
@@ -2612,7 +2612,7 @@ __global__ void lcc_orient_large_kernel(
       for (;;) {
         uint32_t old = atomicCAS(&tl[idx], kCdlpEmpty, u);
         if (old == kCdlpEmpty) {
-          if (lcc_keep(D[u], u, dv, v))
+          if (!filter || lcc_keep(D[u], u, dv, v))
             oadj[base + atomicAdd(&s_cursor, 1u)] = u;
           break;
         }
@@ -2973,6 +2973,128 @@ __global__ void lcc_triangle_heavy_kernel(
   }
 }
 
+// --- directed LCC (reference cuda/lcc/lcc_directed*.h semantics) ---------
+// numerator(v) = sum over u in N(v) of |N(v) ∩ Nout(u)| computed directly
+// (no orientation trick applies to the directed definition): enumerate the
+// smaller of U(v) / Nout(u) and probe the other family's hash set.
+__global__ void lcc_dir_count_kernel(
+    const uint64_t* __restrict__ goffU, const uint32_t* __restrict__ gdstU,
+    const uint64_t* __restrict__ hoffU, const uint32_t* __restrict__ htabU,
+    const uint64_t* __restrict__ goffO, const uint32_t* __restrict__ gdstO,
+    const uint64_t* __restrict__ hoffO, const uint32_t* __restrict__ htabO,
+    uint32_t owned, uint32_t v_begin, unsigned long long* __restrict__ T,
+    uint32_t heavy_thresh, unsigned long long* __restrict__ heavy_q,
+    unsigned long long* __restrict__ heavy_n) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wpb = kBlock / kWave;
+  size_t wstride = static_cast<size_t>(gridDim.x) * wpb;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid; i < owned;
+       i += wstride) {
+    uint32_t v = v_begin + i;
+    uint64_t ub = goffU[v];
+    uint32_t un = static_cast<uint32_t>(goffU[v + 1] - ub);
+    if (un < 2) continue;
+    unsigned long long my_v = 0;
+    for (uint32_t k = lane; k < un; k += kWave) {
+      uint32_t u = gdstU[ub + k];
+      uint64_t ob = goffO[u];
+      uint32_t on = static_cast<uint32_t>(goffO[u + 1] - ob);
+      if (on == 0) continue;
+      uint32_t mn = on < un ? on : un;
+      if (mn > heavy_thresh) {
+        heavy_q[atomicAdd(heavy_n, 1ull)] =
+            (static_cast<unsigned long long>(v) << 32) | u;
+        continue;
+      }
+      unsigned long long hits = 0;
+      if (on <= un) {
+        const uint64_t mask = (hoffU[v + 1] - hoffU[v]) - 1;
+        const uint32_t* t = htabU + hoffU[v];
+        for (uint32_t e = 0; e < on; ++e)
+          if (lcc_probe(t, mask, gdstO[ob + e])) ++hits;
+      } else {
+        const uint64_t mask = (hoffO[u + 1] - hoffO[u]) - 1;
+        const uint32_t* t = htabO + hoffO[u];
+        for (uint32_t e = 0; e < un; ++e)
+          if (lcc_probe(t, mask, gdstU[ub + e])) ++hits;
+      }
+      my_v += hits;
+    }
+#pragma unroll
+    for (int d = 32; d > 0; d >>= 1)
+      my_v += __shfl_down(my_v, d, 64);
+    if (lane == 0 && my_v) atomicAdd(&T[v], my_v);
+  }
+}
+
+__global__ void lcc_dir_heavy_kernel(
+    const uint64_t* __restrict__ goffU, const uint32_t* __restrict__ gdstU,
+    const uint64_t* __restrict__ hoffU, const uint32_t* __restrict__ htabU,
+    const uint64_t* __restrict__ goffO, const uint32_t* __restrict__ gdstO,
+    const uint64_t* __restrict__ hoffO, const uint32_t* __restrict__ htabO,
+    const unsigned long long* __restrict__ heavy_q, uint64_t heavy_n,
+    unsigned long long* __restrict__ T) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wpb = kBlock / kWave;
+  size_t wstride = static_cast<size_t>(gridDim.x) * wpb;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid;
+       i < heavy_n; i += wstride) {
+    unsigned long long pk = heavy_q[i];
+    uint32_t v = static_cast<uint32_t>(pk >> 32);
+    uint32_t u = static_cast<uint32_t>(pk);
+    uint64_t ub = goffU[v];
+    uint32_t un = static_cast<uint32_t>(goffU[v + 1] - ub);
+    uint64_t ob = goffO[u];
+    uint32_t on = static_cast<uint32_t>(goffO[u + 1] - ob);
+    unsigned long long hits = 0;
+    if (on <= un) {
+      const uint64_t mask = (hoffU[v + 1] - hoffU[v]) - 1;
+      const uint32_t* t = htabU + hoffU[v];
+      for (uint32_t e = lane; e < on; e += kWave)
+        if (lcc_probe(t, mask, gdstO[ob + e])) ++hits;
+    } else {
+      const uint64_t mask = (hoffO[u + 1] - hoffO[u]) - 1;
+      const uint32_t* t = htabO + hoffO[u];
+      for (uint32_t e = lane; e < un; e += kWave)
+        if (lcc_probe(t, mask, gdstU[ub + e])) ++hits;
+    }
+#pragma unroll
+    for (int d = 32; d > 0; d >>= 1)
+      hits += __shfl_down(hits, d, 64);
+    if (lane == 0 && hits) atomicAdd(&T[v], hits);
+  }
+}
+
+// caps for the dedup'd OUT family: distinct-out <= min(out_deg, D)
+__global__ void lcc_outcap_kernel(const uint64_t* __restrict__ off,
+                                  const uint32_t* __restrict__ D,
+                                  uint32_t owned, uint32_t v_begin,
+                                  uint32_t* __restrict__ caps) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < owned;
+       r += stride) {
+    uint64_t d = off[r + 1] - off[r];
+    uint32_t dd = D[v_begin + r];
+    caps[r] = static_cast<uint32_t>(d < dd ? d : dd);
+  }
+}
+
+__global__ void lcc_dir_finalize_kernel(
+    const unsigned long long* __restrict__ T,
+    const uint32_t* __restrict__ D, uint32_t owned, uint32_t v_begin,
+    double* __restrict__ lcc) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < owned;
+       r += stride) {
+    uint32_t d = D[v_begin + r];
+    lcc[r] = d < 2 ? 0.0
+                   : static_cast<double>(T[v_begin + r]) /
+                         (static_cast<double>(d) * (d - 1));
+  }
+}
+
 __global__ void lcc_finalize_kernel(const unsigned long long* __restrict__ T,
                                     const uint32_t* __restrict__ D,
                                     uint32_t owned, uint32_t v_begin,
@@ -2987,6 +3109,269 @@ __global__ void lcc_finalize_kernel(const unsigned long long* __restrict__ T,
   }
 }
 
+namespace {
+
+// Shared piece: turn capacity-layout per-owned-row lists into a replicated
+// global CSR (allgather counts -> scan -> compact -> per-rank broadcast).
+struct GlobalDedupCsr {
+  DeviceBuffer<uint32_t> gcnt;
+  DeviceBuffer<uint64_t> goff;
+  DeviceBuffer<uint32_t> gdst;
+  uint64_t total = 0;
+};
+
+void build_global_csr(GpuContext::Impl& I, ncclComm_t nccl, int rank,
+                      int world, uint32_t nv_pad, uint32_t slice,
+                      uint32_t owned, uint32_t v_begin,
+                      const DeviceBuffer<uint64_t>& ooff,
+                      const DeviceBuffer<uint32_t>& ocnt,
+                      const DeviceBuffer<uint32_t>& oadj,
+                      GlobalDedupCsr& out, hipStream_t s) {
+  bool multi = world > 1;
+  out.gcnt.resize(nv_pad);
+  out.gcnt.zero(s);
+  HIP_CHECK(hipMemcpyAsync(out.gcnt.data() + v_begin, ocnt.data(),
+                           owned * 4, hipMemcpyDeviceToDevice, s));
+  if (multi)
+    NCCL_CHECK(ncclAllGather(
+        out.gcnt.data() + static_cast<uint64_t>(rank) * slice,
+        out.gcnt.data(), slice, ncclUint32, nccl, s));
+  out.goff.resize(static_cast<size_t>(nv_pad) + 1);
+  out.total =
+      exclusive_scan(out.gcnt.data(), out.goff.data(), nv_pad, s, I.scan);
+  out.gdst.resize(out.total ? out.total : 1);
+  if (owned)
+    lcc_compact_kernel<<<grid_for(static_cast<size_t>(owned) * kWave),
+                         kBlock, 0, s>>>(ooff.data(), ocnt.data(),
+                                         oadj.data(), out.goff.data(),
+                                         owned, v_begin, out.gdst.data());
+  if (multi) {
+    std::vector<uint64_t> region(world + 1);
+    for (int f = 0; f <= world; ++f) {
+      uint64_t idx =
+          std::min<uint64_t>(static_cast<uint64_t>(f) * slice, nv_pad);
+      HIP_CHECK(hipMemcpyAsync(&region[f], out.goff.data() + idx, 8,
+                               hipMemcpyDeviceToHost, s));
+    }
+    HIP_CHECK(hipStreamSynchronize(s));
+    NCCL_CHECK(ncclGroupStart());
+    for (int f = 0; f < world; ++f) {
+      uint64_t cnt_f = region[f + 1] - region[f];
+      if (cnt_f)
+        NCCL_CHECK(ncclBroadcast(out.gdst.data() + region[f],
+                                 out.gdst.data() + region[f], cnt_f,
+                                 ncclUint32, f, nccl, s));
+    }
+    NCCL_CHECK(ncclGroupEnd());
+  }
+}
+
+void build_hash_sets(GpuContext::Impl& I, uint32_t nv_pad,
+                     const GlobalDedupCsr& csr, DeviceBuffer<uint64_t>& hoff,
+                     DeviceBuffer<uint32_t>& htab, hipStream_t s) {
+  DeviceBuffer<uint32_t> hcap(nv_pad);
+  hoff.resize(static_cast<size_t>(nv_pad) + 1);
+  lcc_hashcap_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(csr.goff.data(),
+                                                         nv_pad,
+                                                         hcap.data());
+  uint64_t total = exclusive_scan(hcap.data(), hoff.data(), nv_pad, s,
+                                  I.scan);
+  htab.resize(total);
+  fill(htab.data(), kCdlpEmpty, total, s);
+  lcc_hashfill_kernel<<<grid_for(static_cast<size_t>(nv_pad) * kWave),
+                        kBlock, 0, s>>>(csr.goff.data(), csr.gdst.data(),
+                                        hoff.data(), nv_pad, htab.data());
+}
+
+}  // namespace
+
+// Directed LCC: build the union family U(v)=N(v) and the deduped OUT
+// family Nout(v), then count |U(v) ∩ Nout(u)| per (v, u∈U(v)) pair.
+GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
+  auto& I = *impl_;
+  hipStream_t s = I.compute;
+  uint32_t nv_pad = padded_nv(g, world_);
+  uint32_t owned = g.owned();
+  uint32_t slice = nv_pad / (world_ ? world_ : 1);
+  bool multi = world_ > 1;
+  if (!g.has_in)
+    throw std::runtime_error(
+        "GPU directed LCC needs the in-CSR (build_in_csr=True)");
+  const uint64_t* off1 = g.oe_off.data();
+  const uint32_t* dst1 = g.oe_dst.data();
+  const uint64_t* off2 = g.ie_off.data();
+  const uint32_t* dst2 = g.ie_dst.data();
+
+  if (comm_) comm_->barrier();
+  HIP_CHECK(hipDeviceSynchronize());
+  double t0 = wall_s();
+
+  DeviceBuffer<uint32_t> Dv(nv_pad);
+  Dv.zero(s);
+  GlobalDedupCsr U, O;
+  {
+    DeviceBuffer<uint32_t> t_small(owned ? owned : 1),
+        t_mid(owned ? owned : 1), t_large(owned ? owned : 1);
+    DeviceBuffer<unsigned long long> cnts(3);
+    cnts.zero(s);
+    if (owned)
+      cdlp_bucket_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+          off1, off2, owned, t_small.data(), cnts.data() + 0, t_mid.data(),
+          cnts.data() + 1, t_large.data(), cnts.data() + 2);
+    auto hc = cnts.download(s);
+    uint64_t n_small = hc[0], n_mid = hc[1], n_large = hc[2];
+
+    DeviceBuffer<uint32_t> heavy_caps;
+    DeviceBuffer<uint64_t> heavy_off;
+    DeviceBuffer<uint32_t> tbl_lab;
+    uint64_t heavy_total = 0;
+    if (n_large) {
+      heavy_caps.resize(n_large);
+      heavy_off.resize(n_large + 1);
+      cdlp_heavy_cap_kernel<<<grid_for(n_large), kBlock, 0, s>>>(
+          off1, off2, t_large.data(), n_large, heavy_caps.data());
+      heavy_total = exclusive_scan(heavy_caps.data(), heavy_off.data(),
+                                   n_large, s, I.scan);
+      tbl_lab.resize(heavy_total);
+    }
+
+    // distinct union counts
+    if (n_small)
+      lcc_distinct_small_kernel<<<grid_for(n_small * kWave), kBlock, 0,
+                                  s>>>(off1, dst1, off2, dst2,
+                                       t_small.data(), n_small, g.v_begin,
+                                       Dv.data());
+    if (n_mid)
+      lcc_distinct_mid_kernel<<<std::min<int>(n_mid, kMaxGrid), kBlock, 0,
+                                s>>>(off1, dst1, off2, dst2, t_mid.data(),
+                                     n_mid, g.v_begin, Dv.data());
+    if (n_large) {
+      fill(tbl_lab.data(), kCdlpEmpty, heavy_total, s);
+      lcc_distinct_large_kernel<<<std::min<int>(n_large, kMaxGrid), kBlock,
+                                  0, s>>>(off1, dst1, off2, dst2,
+                                          t_large.data(), n_large,
+                                          heavy_off.data(), tbl_lab.data(),
+                                          g.v_begin, Dv.data());
+    }
+    if (multi)
+      NCCL_CHECK(ncclAllGather(
+          Dv.data() + static_cast<uint64_t>(rank_) * slice, Dv.data(),
+          slice, ncclUint32, I.nccl, s));
+
+    // family builder: run the dedup pass over (o1,d1,o2,d2) with given
+    // capacities, then lift to a replicated global CSR
+    auto build_family = [&](const uint64_t* o1, const uint32_t* d1,
+                            const uint64_t* o2, const uint32_t* d2,
+                            const DeviceBuffer<uint32_t>& caps,
+                            GlobalDedupCsr& out) {
+      DeviceBuffer<uint64_t> ooff(owned + 1);
+      uint64_t cap_total =
+          exclusive_scan(caps.data(), ooff.data(), owned, s, I.scan);
+      DeviceBuffer<uint32_t> oadj(cap_total ? cap_total : 1);
+      DeviceBuffer<uint32_t> ocnt(owned ? owned : 1);
+      ocnt.zero(s);
+      if (n_small)
+        lcc_orient_small_kernel<<<grid_for(n_small * kWave), kBlock, 0,
+                                  s>>>(o1, d1, o2, d2, t_small.data(),
+                                       n_small, g.v_begin, Dv.data(),
+                                       ooff.data(), oadj.data(),
+                                       ocnt.data(), false);
+      if (n_mid)
+        lcc_orient_mid_kernel<<<std::min<int>(n_mid, kMaxGrid), kBlock, 0,
+                                s>>>(o1, d1, o2, d2, t_mid.data(), n_mid,
+                                     g.v_begin, Dv.data(), ooff.data(),
+                                     oadj.data(), ocnt.data(), false);
+      if (n_large) {
+        fill(tbl_lab.data(), kCdlpEmpty, heavy_total, s);
+        lcc_orient_large_kernel<<<std::min<int>(n_large, kMaxGrid), kBlock,
+                                  0, s>>>(o1, d1, o2, d2, t_large.data(),
+                                          n_large, heavy_off.data(),
+                                          tbl_lab.data(), g.v_begin,
+                                          Dv.data(), ooff.data(),
+                                          oadj.data(), ocnt.data(), false);
+      }
+      build_global_csr(I, I.nccl, rank_, world_, nv_pad, slice, owned,
+                       g.v_begin, ooff, ocnt, oadj, out, s);
+    };
+
+    // U family: in ∪ out, cap = D
+    {
+      DeviceBuffer<uint32_t> caps(owned ? owned : 1);
+      HIP_CHECK(hipMemcpyAsync(caps.data(), Dv.data() + g.v_begin,
+                               owned * 4, hipMemcpyDeviceToDevice, s));
+      build_family(off1, dst1, off2, dst2, caps, U);
+    }
+    // O family: distinct out, cap = min(out_deg, D)
+    {
+      DeviceBuffer<uint32_t> caps(owned ? owned : 1);
+      if (owned)
+        lcc_outcap_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+            off1, Dv.data(), owned, g.v_begin, caps.data());
+      build_family(off1, dst1, nullptr, nullptr, caps, O);
+    }
+  }
+
+  DeviceBuffer<uint64_t> hoffU, hoffO;
+  DeviceBuffer<uint32_t> htabU, htabO;
+  build_hash_sets(I, nv_pad, U, hoffU, htabU, s);
+  build_hash_sets(I, nv_pad, O, hoffO, htabO, s);
+
+  DeviceBuffer<unsigned long long> Tcnt(nv_pad);
+  Tcnt.zero(s);
+  constexpr uint32_t kHeavyThresh = 96;
+  DeviceBuffer<unsigned long long> heavy_q;
+  DeviceBuffer<unsigned long long> heavy_n(1);
+  heavy_n.zero(s);
+  {
+    uint64_t lo = 0, hi = 0;
+    HIP_CHECK(hipMemcpyAsync(&lo, U.goff.data() + g.v_begin, 8,
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipMemcpyAsync(&hi,
+                             U.goff.data() + std::min<uint64_t>(g.v_end,
+                                                                nv_pad),
+                             8, hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    heavy_q.resize(std::max<uint64_t>(hi - lo, 1));
+  }
+  if (owned)
+    lcc_dir_count_kernel<<<grid_for(static_cast<size_t>(owned) * kWave),
+                           kBlock, 0, s>>>(
+        U.goff.data(), U.gdst.data(), hoffU.data(), htabU.data(),
+        O.goff.data(), O.gdst.data(), hoffO.data(), htabO.data(), owned,
+        g.v_begin, Tcnt.data(), kHeavyThresh, heavy_q.data(),
+        heavy_n.data());
+  {
+    unsigned long long hn = 0;
+    HIP_CHECK(hipMemcpyAsync(&hn, heavy_n.data(), 8, hipMemcpyDeviceToHost,
+                             s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    if (hn)
+      lcc_dir_heavy_kernel<<<grid_for(hn * kWave), kBlock, 0, s>>>(
+          U.goff.data(), U.gdst.data(), hoffU.data(), htabU.data(),
+          O.goff.data(), O.gdst.data(), hoffO.data(), htabO.data(),
+          heavy_q.data(), hn, Tcnt.data());
+  }
+  DeviceBuffer<double> lcc_out(owned ? owned : 1);
+  if (owned)
+    lcc_dir_finalize_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+        Tcnt.data(), Dv.data(), owned, g.v_begin, lcc_out.data());
+  HIP_CHECK(hipDeviceSynchronize());
+  if (comm_) comm_->barrier();
+  double t1 = wall_s();
+
+  GpuRunResult res;
+  res.rounds = 1;
+  res.seconds = comm_ ? comm_->allreduce_max_double(t1 - t0) : (t1 - t0);
+  res.traversed_edges = g.input_edges;
+  if (fetch) {
+    res.f64.resize(owned);
+    HIP_CHECK(hipMemcpyAsync(res.f64.data(), lcc_out.data(), owned * 8,
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+  }
+  return res;
+}
+
 GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
   auto& I = *impl_;
   hipStream_t s = I.compute;
@@ -2994,11 +3379,7 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
   uint32_t owned = g.owned();
   uint32_t slice = nv_pad / (world_ ? world_ : 1);
   bool multi = world_ > 1;
-  if (g.directed)
-    throw std::runtime_error(
-        "GPU LCC currently supports undirected graphs only (the directed "
-        "numerator intersects N(v) with out-neighborhoods; use the CPU "
-        "engine for directed LCC)");
+  if (g.directed) return lcc_directed(g, fetch);
 
   const uint64_t* off1 = g.oe_off.data();
   const uint32_t* dst1 = g.oe_dst.data();
@@ -3080,18 +3461,19 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
     if (n_small)
       lcc_orient_small_kernel<<<grid_for(n_small * kWave), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, t_small.data(), n_small, g.v_begin,
-          Dv.data(), ooff.data(), oadj.data(), ocnt.data());
+          Dv.data(), ooff.data(), oadj.data(), ocnt.data(), true);
     if (n_mid)
       lcc_orient_mid_kernel<<<std::min<int>(n_mid, kMaxGrid), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, t_mid.data(), n_mid, g.v_begin, Dv.data(),
-          ooff.data(), oadj.data(), ocnt.data());
+          ooff.data(), oadj.data(), ocnt.data(), true);
     if (n_large) {
       fill(tbl_lab.data(), kCdlpEmpty, heavy_total, s);
       lcc_orient_large_kernel<<<std::min<int>(n_large, kMaxGrid), kBlock, 0,
                                 s>>>(off1, dst1, off2, dst2, t_large.data(),
                                      n_large, heavy_off.data(),
                                      tbl_lab.data(), g.v_begin, Dv.data(),
-                                     ooff.data(), oadj.data(), ocnt.data());
+                                     ooff.data(), oadj.data(), ocnt.data(),
+                                     true);
     }
 
     // global oriented CSR: allgather counts, scan, compact, sort, exchange

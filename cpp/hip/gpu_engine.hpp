@@ -74,6 +74,7 @@ class GpuContext {
   GpuRunResult wcc(DeviceGraph& g, bool fetch = true);
   GpuRunResult cdlp(DeviceGraph& g, int iters, bool fetch = true);
   GpuRunResult lcc(DeviceGraph& g, bool fetch = true);
+  GpuRunResult lcc_directed(DeviceGraph& g, bool fetch = true);
 
   void device_sync();
   // test hook: exclusive scan of host u32 data on the device

@@ -147,13 +147,30 @@ def test_gpu_lcc_undirected(eng):
     assert np.allclose(vals, expect, rtol=1e-12)
 
 
-def test_gpu_lcc_directed_raises(eng):
-    # directed LCC runs on the CPU engine; the GPU path must refuse loudly
-    src, dst, _ = random_graph(num_v=200, num_e=1000, seed=41)
-    g = eng.load_edges(src, dst, directed=True, num_vertices=200,
+def test_gpu_lcc_directed(eng):
+    src, dst, _ = random_graph(num_v=1500, num_e=20000, seed=41)
+    g = eng.load_edges(src, dst, directed=True, num_vertices=1500,
                        build_in_csr=True)
-    with pytest.raises(RuntimeError, match="undirected"):
-        eng.lcc(g)
+    _, vals = by_oid(eng.lcc(g))
+    expect = lcc_oracle(1500, src, dst, directed=True)
+    assert np.allclose(vals, expect, rtol=1e-12)
+
+
+def test_gpu_lcc_directed_hub(eng):
+    rng = np.random.default_rng(47)
+    nv = 8000
+    hub = np.stack([np.zeros(6000, np.int64), rng.integers(1, nv, 6000)], 1)
+    rev = np.stack([rng.integers(1, nv, 3000), np.zeros(3000, np.int64)], 1)
+    rest = np.stack([rng.integers(0, nv, 30000),
+                     rng.integers(0, nv, 30000)], 1)
+    e = np.concatenate([hub, rev, rest])
+    keep = e[:, 0] != e[:, 1]
+    src, dst = e[keep, 0], e[keep, 1]
+    g = eng.load_edges(src, dst, directed=True, num_vertices=nv,
+                       build_in_csr=True)
+    _, vals = by_oid(eng.lcc(g))
+    expect = lcc_oracle(nv, src, dst, directed=True)
+    assert np.allclose(vals, expect, rtol=1e-12)
 
 
 def test_gpu_lcc_hub(eng):
