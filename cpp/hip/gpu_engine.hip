@@ -416,6 +416,46 @@ __global__ void expand_cm_range(DevGraphView g, EdgeOp op) {
   }
 }
 
+// STRICT load balancing (reference cuda/parallel/parallel_engine.h
+// LBSTRICT :883-979): perfectly edge-balanced — every block owns exactly
+// ceil(total/grid) edge outputs regardless of row skew. Row ownership is
+// found by binary search in the frontier edge prefix; the search is
+// re-anchored once per thread and advanced incrementally (a thread's
+// edges are strided, so its owner only moves forward).
+__device__ __forceinline__ uint32_t strict_owner(
+    const uint64_t* __restrict__ foff, uint32_t qn, uint64_t e) {
+  uint32_t lo = 0, hi = qn;  // find last row with foff[row] <= e
+  while (lo + 1 < hi) {
+    uint32_t mid = (lo + hi) >> 1;
+    if (foff[mid] <= e)
+      lo = mid;
+    else
+      hi = mid;
+  }
+  return lo;
+}
+
+template <typename EdgeOp>
+__global__ void expand_strict_frontier(DevGraphView g,
+                                       const uint32_t* __restrict__ q,
+                                       uint32_t qn,
+                                       const uint64_t* __restrict__ foff,
+                                       uint64_t total, EdgeOp op) {
+  uint64_t per = (total + gridDim.x - 1) / gridDim.x;
+  uint64_t eb = per * blockIdx.x;
+  if (eb >= total) return;
+  uint64_t ee = eb + per < total ? eb + per : total;
+  uint64_t e = eb + threadIdx.x;
+  if (e >= ee) return;
+  uint32_t owner = strict_owner(foff, qn, e);
+  for (; e < ee; e += blockDim.x) {
+    while (owner + 1 < qn && foff[owner + 1] <= e) ++owner;
+    uint32_t u = q[owner];
+    uint64_t eid = g.oe_off[g.row(u)] + (e - foff[owner]);
+    op(u, g.oe_dst[eid], g.oe_w ? g.oe_w[eid] : 1.0f);
+  }
+}
+
 // Thread-per-item baseline (LB=none), frontier form.
 template <typename EdgeOp>
 __global__ void expand_none_frontier(DevGraphView g,
@@ -1369,10 +1409,22 @@ void expand_frontier(GpuContext::Impl& I, const DevGraphView& view,
     I.frontier_off.resize(qn + (qn >> 2) + 65);
   gather_deg_kernel<<<grid_for(qn), kBlock, 0, s>>>(view, q, qn,
                                                     I.frontier_deg.data());
-  exclusive_scan(I.frontier_deg.data(), I.frontier_off.data(), qn, s, I.scan);
-  int nchunks = static_cast<int>((qn + kBlock - 1) / kBlock);
-  expand_cm_frontier<Op><<<std::min(nchunks, kMaxGrid), kBlock, 0, s>>>(
-      view, q, qn, I.frontier_off.data(), op);
+  uint64_t total = exclusive_scan(I.frontier_deg.data(),
+                                  I.frontier_off.data(), qn, s, I.scan);
+  // LB strategy (reference --lb flag, default cta/cm): cm = LDS-staged
+  // owner search; strict = perfectly edge-balanced; none = thread-per-row
+  static const char* lb = getenv("GRAPEHIP_LB");
+  if (lb && lb[0] == 's') {
+    expand_strict_frontier<Op><<<grid_for(total), kBlock, 0, s>>>(
+        view, q, qn, I.frontier_off.data(), total, op);
+  } else if (lb && lb[0] == 'n') {
+    expand_none_frontier<Op><<<grid_for(qn), kBlock, 0, s>>>(view, q, qn,
+                                                             op);
+  } else {
+    int nchunks = static_cast<int>((qn + kBlock - 1) / kBlock);
+    expand_cm_frontier<Op><<<std::min(nchunks, kMaxGrid), kBlock, 0, s>>>(
+        view, q, qn, I.frontier_off.data(), op);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -2861,7 +2913,8 @@ __global__ void lcc_triangle_kernel(const uint64_t* __restrict__ goff,
                                     unsigned long long* __restrict__ T,
                                     uint32_t heavy_thresh,
                                     unsigned long long* __restrict__ heavy_q,
-                                    unsigned long long* __restrict__ heavy_n) {
+                                    unsigned long long* __restrict__ heavy_n,
+                                    bool skip_witness) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int wpb = kBlock / kWave;
@@ -2906,7 +2959,7 @@ __global__ void lcc_triangle_kernel(const uint64_t* __restrict__ goff,
         uint32_t w = gdst[eb + e];
         if (lcc_probe(t, mask, w)) {
           ++hits;
-          atomicAdd(&T[w], 1ull);
+          if (!skip_witness) atomicAdd(&T[w], 1ull);
         }
       }
       if (hits) {
@@ -3573,6 +3626,7 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
   Tcnt.resize(nv_pad);
   Tcnt.zero(s);
   constexpr uint32_t kHeavyThresh = 96;
+  const bool nowit = getenv("GRAPEHIP_LCC_NOWIT") != nullptr;
   DeviceBuffer<unsigned long long> heavy_q;
   DeviceBuffer<unsigned long long> heavy_n(1);
   heavy_n.zero(s);
@@ -3595,7 +3649,7 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
                                           hoff.data(), htab.data(), owned,
                                           g.v_begin, Tcnt.data(),
                                           kHeavyThresh, heavy_q.data(),
-                                          heavy_n.data());
+                                          heavy_n.data(), nowit);
   {
     unsigned long long hn = 0;
     HIP_CHECK(hipMemcpyAsync(&hn, heavy_n.data(), 8, hipMemcpyDeviceToHost,
