@@ -2905,26 +2905,34 @@ __device__ __forceinline__ bool lcc_probe(const uint32_t* __restrict__ t,
   }
 }
 
-__global__ void lcc_triangle_kernel(const uint64_t* __restrict__ goff,
-                                    const uint32_t* __restrict__ gdst,
-                                    const uint64_t* __restrict__ hoff,
-                                    const uint32_t* __restrict__ htab,
-                                    uint32_t owned, uint32_t v_begin,
-                                    unsigned long long* __restrict__ T,
-                                    uint32_t heavy_thresh,
-                                    unsigned long long* __restrict__ heavy_q,
-                                    unsigned long long* __restrict__ heavy_n,
-                                    bool skip_witness) {
+// Light-edge pass, two tiers by oriented row size:
+//   wee rows (un <= 32): wave-per-row, lanes probe global hash sets
+//   staged rows: block-per-row; the row's hash set is staged into LDS once
+//   and every thread's probes hit LDS (the global-probe version spent its
+//   time on scattered 4B HBM reads; witness atomics measured ~5%)
+constexpr uint32_t kLccWeeRows = 32;
+constexpr uint32_t kLccStageSlots = 8192;  // 32 KB LDS
+
+__global__ void lcc_tri_wee_kernel(const uint64_t* __restrict__ goff,
+                                   const uint32_t* __restrict__ gdst,
+                                   const uint64_t* __restrict__ hoff,
+                                   const uint32_t* __restrict__ htab,
+                                   const uint32_t* __restrict__ rows,
+                                   uint64_t nrows, uint32_t v_begin,
+                                   unsigned long long* __restrict__ T,
+                                   uint32_t heavy_thresh,
+                                   unsigned long long* __restrict__ heavy_q,
+                                   unsigned long long* __restrict__ heavy_n,
+                                   bool skip_witness) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int wpb = kBlock / kWave;
   size_t wstride = static_cast<size_t>(gridDim.x) * wpb;
-  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid; i < owned;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid; i < nrows;
        i += wstride) {
-    uint32_t u = v_begin + i;
-    uint64_t ub = goff[u], ue = goff[u + 1];
-    uint32_t un = static_cast<uint32_t>(ue - ub);
-    if (un < 1) continue;
+    uint32_t u = v_begin + rows[i];
+    uint64_t ub = goff[u];
+    uint32_t un = static_cast<uint32_t>(goff[u + 1] - ub);
     unsigned long long my_u = 0;
     for (uint32_t k = lane; k < un; k += kWave) {
       uint32_t v = gdst[ub + k];
@@ -2933,27 +2941,25 @@ __global__ void lcc_triangle_kernel(const uint64_t* __restrict__ goff,
       if (vn == 0) continue;
       uint32_t mn = vn < un ? vn : un;
       if (mn > heavy_thresh) {
-        // defer: one lane grinding thousands of probes stalls its whole
-        // wave; heavy edges get a wave each in the second pass
         heavy_q[atomicAdd(heavy_n, 1ull)] =
             (static_cast<unsigned long long>(u) << 32) | v;
         continue;
       }
-      uint64_t eb, hb_big, hcap_end;
+      uint64_t eb, hb;
       uint32_t en;
+      uint64_t mask;
       if (vn < un) {
         eb = vb;
         en = vn;
-        hb_big = hoff[u];
-        hcap_end = hoff[u + 1];
+        hb = hoff[u];
+        mask = (hoff[u + 1] - hb) - 1;
       } else {
         eb = ub;
         en = un;
-        hb_big = hoff[v];
-        hcap_end = hoff[v + 1];
+        hb = hoff[v];
+        mask = (hoff[v + 1] - hb) - 1;
       }
-      const uint64_t mask = (hcap_end - hb_big) - 1;
-      const uint32_t* t = htab + hb_big;
+      const uint32_t* t = htab + hb;
       unsigned long long hits = 0;
       for (uint32_t e = 0; e < en; ++e) {
         uint32_t w = gdst[eb + e];
@@ -2971,6 +2977,124 @@ __global__ void lcc_triangle_kernel(const uint64_t* __restrict__ goff,
     for (int d = 32; d > 0; d >>= 1)
       my_u += __shfl_down(my_u, d, 64);
     if (lane == 0 && my_u) atomicAdd(&T[u], my_u);
+  }
+}
+
+__device__ __forceinline__ bool lcc_probe_lds(
+    const uint32_t* __restrict__ t, uint32_t mask, uint32_t key) {
+  uint32_t idx = cdlp_hash(key) & mask;
+  for (;;) {
+    uint32_t x = t[idx];
+    if (x == key) return true;
+    if (x == kCdlpEmpty) return false;
+    idx = (idx + 1) & mask;
+  }
+}
+
+__global__ void lcc_tri_staged_kernel(
+    const uint64_t* __restrict__ goff, const uint32_t* __restrict__ gdst,
+    const uint64_t* __restrict__ hoff, const uint32_t* __restrict__ htab,
+    const uint32_t* __restrict__ rows, uint64_t nrows, uint32_t v_begin,
+    unsigned long long* __restrict__ T, uint32_t heavy_thresh,
+    unsigned long long* __restrict__ heavy_q,
+    unsigned long long* __restrict__ heavy_n, bool skip_witness) {
+  __shared__ uint32_t s_tab[kLccStageSlots];
+  __shared__ unsigned long long s_hits;
+  for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
+    uint32_t u = v_begin + rows[i];
+    uint64_t ub = goff[u];
+    uint32_t un = static_cast<uint32_t>(goff[u + 1] - ub);
+    uint64_t hb = hoff[u];
+    uint32_t ucap = static_cast<uint32_t>(hoff[u + 1] - hb);
+    const bool stage = ucap <= kLccStageSlots;
+    if (stage)
+      for (uint32_t k = threadIdx.x; k < ucap; k += blockDim.x)
+        s_tab[k] = htab[hb + k];
+    if (threadIdx.x == 0) s_hits = 0;
+    __syncthreads();
+    const uint32_t umask = ucap - 1;
+    const uint32_t* ut = htab + hb;
+    unsigned long long my_u = 0;
+    for (uint32_t k = threadIdx.x; k < un; k += blockDim.x) {
+      uint32_t v = gdst[ub + k];
+      uint64_t vb = goff[v];
+      uint32_t vn = static_cast<uint32_t>(goff[v + 1] - vb);
+      if (vn == 0) continue;
+      uint32_t mn = vn < un ? vn : un;
+      if (mn > heavy_thresh) {
+        heavy_q[atomicAdd(heavy_n, 1ull)] =
+            (static_cast<unsigned long long>(u) << 32) | v;
+        continue;
+      }
+      unsigned long long hits = 0;
+      if (vn < un) {
+        // enumerate O(v), probe u's set (LDS when staged)
+        for (uint32_t e = 0; e < vn; ++e) {
+          uint32_t w = gdst[vb + e];
+          bool hit = stage ? lcc_probe_lds(s_tab, umask, w)
+                           : lcc_probe(ut, umask, w);
+          if (hit) {
+            ++hits;
+            if (!skip_witness) atomicAdd(&T[w], 1ull);
+          }
+        }
+      } else {
+        // enumerate O(u) (streamed, shared), probe v's global set
+        uint64_t vhb = hoff[v];
+        const uint64_t vmask = (hoff[v + 1] - vhb) - 1;
+        const uint32_t* vt = htab + vhb;
+        for (uint32_t e = 0; e < un; ++e) {
+          uint32_t w = gdst[ub + e];
+          if (lcc_probe(vt, vmask, w)) {
+            ++hits;
+            if (!skip_witness) atomicAdd(&T[w], 1ull);
+          }
+        }
+      }
+      if (hits) {
+        my_u += hits;
+        atomicAdd(&T[v], hits);
+      }
+    }
+    if (my_u) atomicAdd(&s_hits, my_u);
+    __syncthreads();
+    if (threadIdx.x == 0 && s_hits) atomicAdd(&T[u], s_hits);
+    __syncthreads();
+  }
+}
+
+// bucket rows for the two light tiers (skips empty rows)
+__global__ void lcc_tribucket_kernel(const uint64_t* __restrict__ goff,
+                                     uint32_t owned, uint32_t v_begin,
+                                     uint32_t* wee, unsigned long long* cw,
+                                     uint32_t* big, unsigned long long* cb) {
+  __shared__ uint32_t s_cnt[2];
+  __shared__ unsigned long long s_base[2];
+  uint32_t* lists[2] = {wee, big};
+  unsigned long long* gcnt[2] = {cw, cb};
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t base = blockIdx.x * blockDim.x; base < owned;
+       base += stride) {
+    if (threadIdx.x < 2) s_cnt[threadIdx.x] = 0;
+    __syncthreads();
+    uint32_t r = base + threadIdx.x;
+    int b = -1;
+    uint32_t loc = 0;
+    if (r < owned) {
+      uint64_t n = goff[v_begin + r + 1] - goff[v_begin + r];
+      if (n) {
+        b = n <= kLccWeeRows ? 0 : 1;
+        loc = atomicAdd(&s_cnt[b], 1u);
+      }
+    }
+    __syncthreads();
+    if (threadIdx.x < 2 && s_cnt[threadIdx.x])
+      s_base[threadIdx.x] =
+          atomicAdd(gcnt[threadIdx.x],
+                    static_cast<unsigned long long>(s_cnt[threadIdx.x]));
+    __syncthreads();
+    if (b >= 0) lists[b][s_base[b] + loc] = r;
+    __syncthreads();
   }
 }
 
@@ -3643,13 +3767,26 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
     HIP_CHECK(hipStreamSynchronize(s));
     heavy_q.resize(std::max<uint64_t>(local_oedges - base_off, 1));
   }
-  if (owned)
-    lcc_triangle_kernel<<<grid_for(static_cast<size_t>(owned) * kWave),
-                          kBlock, 0, s>>>(goff.data(), gdst.data(),
-                                          hoff.data(), htab.data(), owned,
-                                          g.v_begin, Tcnt.data(),
-                                          kHeavyThresh, heavy_q.data(),
-                                          heavy_n.data(), nowit);
+  if (owned) {
+    DeviceBuffer<uint32_t> wee_rows(owned), big_rows(owned);
+    DeviceBuffer<unsigned long long> tcnts(2);
+    tcnts.zero(s);
+    lcc_tribucket_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+        goff.data(), owned, g.v_begin, wee_rows.data(), tcnts.data() + 0,
+        big_rows.data(), tcnts.data() + 1);
+    auto tc = tcnts.download(s);
+    if (tc[0])
+      lcc_tri_wee_kernel<<<grid_for(tc[0] * kWave), kBlock, 0, s>>>(
+          goff.data(), gdst.data(), hoff.data(), htab.data(),
+          wee_rows.data(), tc[0], g.v_begin, Tcnt.data(), kHeavyThresh,
+          heavy_q.data(), heavy_n.data(), nowit);
+    if (tc[1])
+      lcc_tri_staged_kernel<<<std::min<uint64_t>(tc[1], kMaxGrid), kBlock,
+                              0, s>>>(
+          goff.data(), gdst.data(), hoff.data(), htab.data(),
+          big_rows.data(), tc[1], g.v_begin, Tcnt.data(), kHeavyThresh,
+          heavy_q.data(), heavy_n.data(), nowit);
+  }
   {
     unsigned long long hn = 0;
     HIP_CHECK(hipMemcpyAsync(&hn, heavy_n.data(), 8, hipMemcpyDeviceToHost,
