@@ -8,6 +8,7 @@
 // N/A. Output: double rank per vertex.
 #pragma once
 
+#include <cmath>
 #include <vector>
 
 #include "../core/fragment.hpp"
@@ -18,13 +19,16 @@ namespace grapehip {
 struct PageRankContext {
   double damping;
   int max_iters;
+  double tol = 0.0;  // >0: competitor-equivalent convergence (reference
+                     // pagerank_local.h) — stop at global L1 delta < tol
   int iter = 0;
   std::vector<double> rank;               // over ivnum
   std::vector<std::atomic<double>> acc;   // over tvnum: pushed contributions
 
-  void init(const Fragment& frag, double d, int mr) {
+  void init(const Fragment& frag, double d, int mr, double tol_ = 0.0) {
     damping = d;
     max_iters = mr;
+    tol = tol_;
     iter = 0;
     double r0 = 1.0 / static_cast<double>(frag.total_vertices());
     rank.assign(frag.ivnum(), r0);
@@ -100,14 +104,22 @@ class PageRankApp {
       }
     }
 
-    // apply
+    // apply (+ optional L1 delta for convergence mode)
     const double base = (1.0 - d) / N + d * dangling / N;
-    parallel_for(0, iv, [&](size_t v) {
-      ctx.rank[v] = base + d * ctx.acc[v].load(std::memory_order_relaxed);
+    std::vector<double> l1_t(mm.n_threads(), 0.0);
+    parallel_for_tid(0, iv, [&](int tid, size_t v) {
+      double nv = base + d * ctx.acc[v].load(std::memory_order_relaxed);
+      if (ctx.tol > 0) l1_t[tid] += std::fabs(nv - ctx.rank[v]);
+      ctx.rank[v] = nv;
       ctx.acc[v].store(0.0, std::memory_order_relaxed);
     }, 4096);
-
-    if (++ctx.iter < ctx.max_iters) mm.force_continue();
+    bool converged = false;
+    if (ctx.tol > 0) {
+      double l1 = 0;
+      for (double x : l1_t) l1 += x;
+      converged = mm.sum_double(l1) < ctx.tol;
+    }
+    if (!converged && ++ctx.iter < ctx.max_iters) mm.force_continue();
   }
 };
 

@@ -507,13 +507,13 @@ PYBIND11_MODULE(_core, m) {
            py::arg("values") = true)
       .def("pagerank",
            [](PyEngine& eng, PyGraph& g, double damping, int iters,
-              bool values) {
+              double tol, bool values) {
 #ifdef GRAPEHIP_WITH_HIP
              if (eng.use_gpu) {
                GpuRunResult r;
                {
                  py::gil_scoped_release rel;
-                 r = eng.gpu->pagerank(*g.dev, damping, iters, values);
+                 r = eng.gpu->pagerank(*g.dev, damping, iters, tol, values);
                }
                return gpu_dict(r, *g.dev, false, values);
              }
@@ -522,7 +522,7 @@ PYBIND11_MODULE(_core, m) {
              PageRankContext ctx;
              MessageManager mm;
              mm.init(eng.c(), g.frag.get(), eng.n_threads);
-             ctx.init(*g.frag, damping, iters);
+             ctx.init(*g.frag, damping, iters, tol);
              py::dict meta = run_timed(
                  eng, [&] { return RunWorker(app, ctx, *g.frag, mm); });
              meta["oids"] = inner_oids(*g.frag);
@@ -530,7 +530,7 @@ PYBIND11_MODULE(_core, m) {
              return meta;
            },
            py::arg("graph"), py::arg("damping") = 0.85, py::arg("iters") = 10,
-           py::arg("values") = true)
+           py::arg("tol") = 0.0, py::arg("values") = true)
       .def("wcc",
            [](PyEngine& eng, PyGraph& g, bool values) {
 #ifdef GRAPEHIP_WITH_HIP

@@ -60,6 +60,15 @@ class MessageManager {
 
   void force_continue() { force_continue_ = true; }
 
+  // Cooperative abort (reference MessageManagerBase::ForceTerminate +
+  // TerminateInfo, default_message_manager.h:156-166): the error string is
+  // gathered on every rank at the next round boundary and thrown there,
+  // so a failure on one fragment ends the whole query cleanly.
+  void force_terminate(const std::string& info) {
+    force_terminate_ = true;
+    terminate_info_ = info;
+  }
+
   // ---- round boundary --------------------------------------------------
   void finish_round() {
     // concatenate per-thread channels per destination
@@ -82,6 +91,20 @@ class MessageManager {
     uint64_t global_moved =
         comm_ ? comm_->allreduce_sum(moved + (force_continue_ ? 1 : 0))
               : moved + (force_continue_ ? 1 : 0);
+    bool any_abort = comm_ ? comm_->allreduce_or(force_terminate_)
+                           : force_terminate_;
+    if (any_abort) {
+      std::string info = terminate_info_;
+      if (comm_) {
+        std::vector<std::string> send(fnum_, terminate_info_);
+        auto all = comm_->exchange_all(send);
+        for (auto& blob : all)
+          if (info.empty() && !blob.empty()) info = blob;
+      }
+      throw std::runtime_error("query force-terminated: " +
+                               (info.empty() ? std::string("(no info)")
+                                             : info));
+    }
     terminated_ = (global_moved == 0);
     ++round_;
   }
@@ -174,6 +197,8 @@ class MessageManager {
   int fnum_ = 1;
   int n_threads_ = 1;
   bool force_continue_ = false;
+  bool force_terminate_ = false;
+  std::string terminate_info_;
   bool terminated_ = false;
   int round_ = 0;
   std::vector<std::vector<InArchive>> channels_;  // [tid][fid]

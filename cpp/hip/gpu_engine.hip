@@ -729,11 +729,28 @@ __global__ void pr_contrib_kernel(const double* __restrict__ rank,
 __global__ void pr_apply_kernel(double* __restrict__ rank,
                                 const double* __restrict__ acc, double base,
                                 double damping, uint32_t v_begin,
-                                uint32_t owned) {
+                                uint32_t owned,
+                                double* __restrict__ l1_delta) {
+  __shared__ double s_wave[kBlock / kWave];
+  double d1 = 0;
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < owned;
-       r += stride)
-    rank[v_begin + r] = base + damping * acc[v_begin + r];
+       r += stride) {
+    double nv = base + damping * acc[v_begin + r];
+    if (l1_delta) d1 += fabs(nv - rank[v_begin + r]);
+    rank[v_begin + r] = nv;
+  }
+  if (!l1_delta) return;
+#pragma unroll
+  for (int d = 32; d > 0; d >>= 1) d1 += __shfl_down(d1, d, 64);
+  if ((threadIdx.x & 63) == 0) s_wave[threadIdx.x >> 6] = d1;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double t = 0;
+#pragma unroll
+    for (int w = 0; w < kBlock / kWave; ++w) t += s_wave[w];
+    unsafeAtomicAdd(l1_delta, t);
+  }
 }
 
 __global__ void pr_dangling_kernel(const double* __restrict__ rank,
@@ -1739,7 +1756,7 @@ GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
 // PageRank (push + fp64 hw atomics; reduce-scatter/allgather over xGMI)
 // ---------------------------------------------------------------------------
 GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
-                                  int iters, bool fetch) {
+                                  int iters, double tol, bool fetch) {
   auto& I = *impl_;
   hipStream_t s = I.compute;
   uint32_t nv_pad = padded_nv(g, world_);
@@ -1761,6 +1778,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
   DeviceBuffer<double> rank_arr(nv_pad), acc(nv_pad);
   DeviceBuffer<float> contrib(nv_pad);
   DeviceBuffer<double> d_dangling(1);
+  DeviceBuffer<double> d_l1(tol > 0 ? 1 : 0);
   DevGraphView view = make_view(g, rank_, world_);
 
   if (comm_) comm_->barrier();
@@ -1811,8 +1829,10 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
                                kBlock, 0, s>>>(
             pull_off, pull_dst, contrib.data(), g.rows_large.data(),
             g.n_large, g.v_begin, acc.data());
+      if (tol > 0) d_l1.zero(s);
       pr_apply_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-          rank_arr.data(), acc.data(), base, damping, g.v_begin, owned);
+          rank_arr.data(), acc.data(), base, damping, g.v_begin, owned,
+          tol > 0 ? d_l1.data() : nullptr);
     } else {
       acc.zero(s);
       int nchunks = static_cast<int>((owned + kBlock - 1) / kBlock);
@@ -1824,10 +1844,27 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
         NCCL_CHECK(ncclReduceScatter(
             acc.data(), acc.data() + static_cast<uint64_t>(rank_) * slice,
             slice, ncclDouble, ncclSum, I.nccl, s));
+      if (tol > 0) d_l1.zero(s);
       pr_apply_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-          rank_arr.data(), acc.data(), base, damping, g.v_begin, owned);
+          rank_arr.data(), acc.data(), base, damping, g.v_begin, owned,
+          tol > 0 ? d_l1.data() : nullptr);
     }
     ++rounds;
+    if (tol > 0) {
+      // competitor-equivalent convergence (reference pagerank_local.h):
+      // stop when the global L1 delta drops under tol
+      double l1 = 0;
+      HIP_CHECK(hipMemcpyAsync(&l1, d_l1.data(), 8, hipMemcpyDeviceToHost,
+                               s));
+      HIP_CHECK(hipStreamSynchronize(s));
+      if (multi) {
+        std::vector<double> all(world_);
+        comm_->allgather(&l1, 8, all.data());
+        l1 = 0;
+        for (double x : all) l1 += x;
+      }
+      if (l1 < tol) break;
+    }
   }
   HIP_CHECK(hipDeviceSynchronize());
   if (comm_) comm_->barrier();

@@ -70,7 +70,7 @@ class GpuContext {
   GpuRunResult sssp(DeviceGraph& g, int64_t source, float delta,
                     bool fetch = true);
   GpuRunResult pagerank(DeviceGraph& g, double damping, int iters,
-                        bool fetch = true);
+                        double tol = 0.0, bool fetch = true);
   GpuRunResult wcc(DeviceGraph& g, bool fetch = true);
   GpuRunResult cdlp(DeviceGraph& g, int iters, bool fetch = true);
   GpuRunResult lcc(DeviceGraph& g, bool fetch = true);

@@ -205,3 +205,15 @@ def test_sssp_auto(eng):
     finite = b < 1e300
     assert np.allclose(a[finite], b[finite], rtol=1e-9)
     assert (a[~finite] > 1e300).all()
+
+
+def test_pagerank_convergence_mode(eng):
+    # pagerank_local parity: tol-based early stop reaches the fixpoint
+    src, dst, _ = random_graph(num_v=300, num_e=2000, seed=69)
+    g = eng.load_edges(src, dst, directed=True, num_vertices=300)
+    r_conv = eng.pagerank(g, 0.85, 1000, tol=1e-12)
+    r_long = eng.pagerank(g, 0.85, 200)
+    assert r_conv["rounds"] < 1000  # stopped early
+    o1, o2 = np.argsort(r_conv["oids"]), np.argsort(r_long["oids"])
+    assert np.allclose(r_conv["values"][o1], r_long["values"][o2],
+                       rtol=1e-9)
