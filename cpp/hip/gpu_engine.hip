@@ -2081,6 +2081,7 @@ __global__ void cdlp_small_kernel(const uint64_t* __restrict__ off1,
                                   const uint32_t* __restrict__ lab,
                                   const uint32_t* __restrict__ rows,
                                   uint64_t nrows, uint32_t v_begin,
+                                  const uint32_t* __restrict__ dirty,
                                   uint32_t* __restrict__ next) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -2089,6 +2090,10 @@ __global__ void cdlp_small_kernel(const uint64_t* __restrict__ off1,
   for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid; i < nrows;
        i += wstride) {
     uint32_t r = rows[i];
+    if (dirty) {
+      uint32_t v = v_begin + r;
+      if (!((dirty[v >> 5] >> (v & 31)) & 1)) continue;
+    }
     uint64_t b1 = off1[r], e1 = off1[r + 1];
     uint32_t d1 = static_cast<uint32_t>(e1 - b1);
     uint32_t deg = d1;
@@ -2153,11 +2158,16 @@ __global__ void cdlp_mid_kernel(const uint64_t* __restrict__ off1,
                                 const uint32_t* __restrict__ lab,
                                 const uint32_t* __restrict__ rows,
                                 uint64_t nrows, uint32_t v_begin,
+                                const uint32_t* __restrict__ dirty,
                                 uint32_t* __restrict__ next) {
   __shared__ uint32_t s_lab[kCdlpLdsSlots];
   __shared__ uint32_t s_cnt[kCdlpLdsSlots];
   for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
     uint32_t r = rows[i];
+    if (dirty) {
+      uint32_t vg = v_begin + r;
+      if (!((dirty[vg >> 5] >> (vg & 31)) & 1)) continue;
+    }
     uint64_t b1 = off1[r], e1 = off1[r + 1];
     uint32_t d1 = static_cast<uint32_t>(e1 - b1);
     uint32_t deg = d1;
@@ -2234,6 +2244,7 @@ __global__ void cdlp_large_kernel(const uint64_t* __restrict__ off1,
                                   const uint64_t* __restrict__ tbl_off,
                                   unsigned long long* __restrict__ tbl,
                                   uint32_t epoch, uint32_t v_begin,
+                                  const uint32_t* __restrict__ dirty,
                                   uint32_t* __restrict__ next) {
   // slot u64 = (epoch:16 | label:32 hashed into low? ) — layout:
   // high 16 bits epoch, next 32 bits label, low 16 bits... counts can
@@ -2244,6 +2255,10 @@ __global__ void cdlp_large_kernel(const uint64_t* __restrict__ off1,
   // clamp only at astronomically heavy rows).
   for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
     uint32_t r = rows[i];
+    if (dirty) {
+      uint32_t vg = v_begin + r;
+      if (!((dirty[vg >> 5] >> (vg & 31)) & 1)) continue;
+    }
     uint64_t b1 = off1[r], e1 = off1[r + 1];
     uint64_t d1 = e1 - b1;
     uint64_t deg = d1;
@@ -2301,12 +2316,49 @@ __global__ void cdlp_large_kernel(const uint64_t* __restrict__ off1,
 
 __global__ void cdlp_commit_kernel(const uint32_t* __restrict__ next,
                                    uint32_t owned, uint32_t v_begin,
-                                   uint32_t* __restrict__ lab) {
+                                   const uint32_t* __restrict__ dirty,
+                                   uint32_t* __restrict__ lab,
+                                   DevBitmap changed,
+                                   unsigned long long* __restrict__ nch) {
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < owned;
-       r += stride)
-    lab[v_begin + r] = next[r];
+       r += stride) {
+    uint32_t v = v_begin + r;
+    if (dirty && !((dirty[v >> 5] >> (v & 31)) & 1)) continue;
+    uint32_t nv = next[r];
+    if (nv != lab[v]) {
+      lab[v] = nv;
+      changed.set_once(r);
+      atomicAdd(nch, 1ull);
+    }
+  }
 }
+
+// mark rows whose label multiset can have changed: every neighbor of a
+// changed vertex (undirected: out-CSR covers it; directed: out + in)
+struct CdlpMarkOp {
+  DevBitmap dirty;
+  __device__ __forceinline__ void operator()(uint32_t v, uint32_t) const {
+    dirty.set_once(v);
+  }
+};
+
+struct CdlpDirtyOp {
+  DevBitmap dirty;  // global-vid bitmap
+  uint32_t v_begin, v_end;
+  bool multi;
+  DevHalo halo;
+  __device__ __forceinline__ void operator()(uint32_t, uint32_t d,
+                                             float) const {
+    if (d >= v_begin && d < v_end) {
+      dirty.set_once(d);
+    } else if (multi) {
+      halo.add(d);
+    } else {
+      dirty.set_once(d);
+    }
+  }
+};
 
 GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
   auto& I = *impl_;
@@ -2355,38 +2407,98 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
 
   DeviceBuffer<uint32_t> lab(nv_pad);
   DeviceBuffer<uint32_t> next(owned ? owned : 1);
+  // delta-convergence machinery: only rows with a changed neighborhood
+  // recompute their mode (exact: an unchanged multiset reproduces the
+  // same mode); fixpoint exits early
+  size_t dirty_words = (static_cast<size_t>(nv_pad) + 31) / 32 + 1;
+  DeviceBuffer<uint32_t> dirty(dirty_words);
+  DeviceBuffer<uint32_t> changed_bm((owned + 31) / 32 + 1);
+  DeviceBuffer<uint32_t> changed_q(owned ? owned : 1);
+  DeviceBuffer<unsigned long long> d_nch(1);
+  uint64_t halo_cap = nv_pad / (world_ ? world_ : 1);
+  if (multi) {
+    I.halo_idx.resize(static_cast<uint64_t>(world_) * halo_cap);
+    I.halo_cnt.resize(world_);
+    I.halo_bm.resize(dirty_words);
+    I.halo_cnt.zero(s);
+    I.halo_bm.zero(s);
+  }
+  DevGraphView view = make_view(g, rank_, world_);
+  DevGraphView view_in = view;
+  if (g.directed) {
+    view_in.oe_off = g.ie_off.data();
+    view_in.oe_dst = g.ie_dst.data();
+    view_in.oe_w = nullptr;
+  }
 
   if (comm_) comm_->barrier();
   HIP_CHECK(hipDeviceSynchronize());
   double t0 = wall_s();
 
   iota_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(lab.data(), 0, nv_pad);
+  changed_bm.zero(s);
   int rounds = 0;
   for (int it = 0; it < iters; ++it) {
+    const uint32_t* dw = it == 0 ? nullptr : dirty.data();
     if (n_small)
       cdlp_small_kernel<<<grid_for(n_small * kWave), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, lab.data(), t_small.data(), n_small,
-          g.v_begin, next.data());
+          g.v_begin, dw, next.data());
     if (n_mid)
       cdlp_mid_kernel<<<std::min<int>(n_mid, kMaxGrid), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, lab.data(), t_mid.data(), n_mid, g.v_begin,
-          next.data());
+          dw, next.data());
     if (n_large && it > 0 && it % 254 == 0)
       heavy_tbl.zero(s);  // epoch tag is 8-bit; re-zero on wrap
     if (n_large)
       cdlp_large_kernel<<<std::min<int>(n_large, kMaxGrid), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, lab.data(), t_large.data(), n_large,
           heavy_off.data(), heavy_tbl.data(),
-          static_cast<uint32_t>((it % 254) + 1), g.v_begin, next.data());
+          static_cast<uint32_t>((it % 254) + 1), g.v_begin, dw,
+          next.data());
+    d_nch.zero(s);
     if (owned)
       cdlp_commit_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-          next.data(), owned, g.v_begin, lab.data());
-    if (multi) {
+          next.data(), owned, g.v_begin, dw, lab.data(),
+          DevBitmap{changed_bm.data()}, d_nch.data());
+    ++rounds;
+    unsigned long long nch = 0;
+    HIP_CHECK(hipMemcpyAsync(&nch, d_nch.data(), 8, hipMemcpyDeviceToHost,
+                             s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    uint64_t g_nch = multi ? comm_->allreduce_sum(nch) : nch;
+    if (multi)
       NCCL_CHECK(ncclAllGather(
           lab.data() + static_cast<uint64_t>(rank_) * slice, lab.data(),
           slice, ncclUint32, I.nccl, s));
+    if (g_nch == 0) break;  // fixpoint: further rounds are no-ops
+    if (it + 1 == iters) break;
+    // rebuild the dirty set from this round's changed rows
+    dirty.zero(s);
+    uint64_t qn = compact_frontier(I, changed_bm.data(), owned, g.v_begin,
+                                   changed_q.data(), s);
+    if (qn) {
+      CdlpDirtyOp op{DevBitmap{dirty.data()}, g.v_begin, g.v_end, multi,
+                     DevHalo{I.halo_idx.data(), I.halo_cnt.data(),
+                             DevBitmap{I.halo_bm.data()}, halo_cap,
+                             view.slice, world_}};
+      expand_frontier(I, view, changed_q.data(),
+                      static_cast<uint32_t>(qn), op, s);
+      if (g.directed)
+        expand_frontier(I, view_in, changed_q.data(),
+                        static_cast<uint32_t>(qn), op, s);
     }
-    ++rounds;
+    if (multi) {
+      uint64_t nrecv = halo_flush<uint32_t>(I, comm_, rank_, world_,
+                                            lab.data(), halo_cap, s);
+      if (nrecv) {
+        DevBitmap db{dirty.data()};
+        halo_process_kernel<uint32_t, CdlpMarkOp>
+            <<<grid_for(nrecv), kBlock, 0, s>>>(
+                reinterpret_cast<HaloPair<uint32_t>*>(I.recvbuf.data()),
+                nrecv, CdlpMarkOp{db});
+      }
+    }
   }
   HIP_CHECK(hipDeviceSynchronize());
   if (comm_) comm_->barrier();
