@@ -9,6 +9,7 @@
 #include <pybind11/stl.h>
 
 #include <chrono>
+#include <fstream>
 #include <memory>
 #include <optional>
 #include <unordered_set>
@@ -288,6 +289,33 @@ PYBIND11_MODULE(_core, m) {
 #ifdef GRAPEHIP_WITH_HIP
              if (e.gpu) e.gpu->device_sync();
 #endif
+           })
+      .def("memory_info",
+           [](PyEngine& eng) {
+             // reference GetMemoryUsage (util.h:51) + TRACKING_MEMORY
+             py::dict out;
+             std::ifstream st("/proc/self/status");
+             std::string line;
+             while (std::getline(st, line)) {
+               if (line.rfind("VmHWM", 0) == 0 ||
+                   line.rfind("VmRSS", 0) == 0) {
+                 auto c = line.find(':');
+                 std::string key = line.substr(0, c);
+                 long kb = atol(line.c_str() + c + 1);
+                 out[key.c_str()] = kb * 1024L;
+               }
+             }
+#ifdef GRAPEHIP_WITH_HIP
+             if (eng.use_gpu) {
+               size_t free_b = 0, total_b = 0;
+               if (hipMemGetInfo(&free_b, &total_b) == hipSuccess) {
+                 out["hip_free"] = free_b;
+                 out["hip_total"] = total_b;
+                 out["hip_used"] = total_b - free_b;
+               }
+             }
+#endif
+             return out;
            })
       .def("_exchange_all",
            [](PyEngine& eng, std::vector<py::bytes> blobs) {

@@ -1449,6 +1449,7 @@ void expand_frontier(GpuContext::Impl& I, const DevGraphView& view,
 // ---------------------------------------------------------------------------
 GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source,
                              bool fetch) {
+  RangeMarker _mk("grapehip::bfs");
   auto& I = *impl_;
   hipStream_t s = I.compute;
   DevGraphView view = make_view(g, rank_, world_);
@@ -1592,6 +1593,7 @@ GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source,
 // ---------------------------------------------------------------------------
 GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
                               float delta, bool fetch) {
+  RangeMarker _mk("grapehip::sssp");
   auto& I = *impl_;
   hipStream_t s = I.compute;
   DevGraphView view = make_view(g, rank_, world_);
@@ -1757,6 +1759,7 @@ GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
 // ---------------------------------------------------------------------------
 GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
                                   int iters, double tol, bool fetch) {
+  RangeMarker _mk("grapehip::pagerank");
   auto& I = *impl_;
   hipStream_t s = I.compute;
   uint32_t nv_pad = padded_nv(g, world_);
@@ -1887,6 +1890,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
 // WCC (min-root union-find, replicated parent + allreduce-min merge)
 // ---------------------------------------------------------------------------
 GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
+  RangeMarker _mk("grapehip::wcc");
   auto& I = *impl_;
   hipStream_t s = I.compute;
   DevGraphView view = make_view(g, rank_, world_);
@@ -2361,6 +2365,7 @@ struct CdlpDirtyOp {
 };
 
 GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
+  RangeMarker _mk("grapehip::cdlp");
   auto& I = *impl_;
   hipStream_t s = I.compute;
   uint32_t nv_pad = padded_nv(g, world_);
@@ -3514,6 +3519,7 @@ void build_hash_sets(GpuContext::Impl& I, uint32_t nv_pad,
 // Directed LCC: build the union family U(v)=N(v) and the deduped OUT
 // family Nout(v), then count |U(v) ∩ Nout(u)| per (v, u∈U(v)) pair.
 GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
+  RangeMarker _mk("grapehip::lcc_directed");
   auto& I = *impl_;
   hipStream_t s = I.compute;
   uint32_t nv_pad = padded_nv(g, world_);
@@ -3699,6 +3705,7 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
 }
 
 GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
+  RangeMarker _mk("grapehip::lcc");
   auto& I = *impl_;
   hipStream_t s = I.compute;
   uint32_t nv_pad = padded_nv(g, world_);

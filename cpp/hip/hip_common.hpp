@@ -10,6 +10,44 @@
 #include <string>
 #include <vector>
 
+// rocTX range markers (reference NVTX RangeMarker, cuda/utils/markers.h):
+// enabled with GRAPEHIP_ROCTX=1; the library is dlopen'd so builds carry
+// no hard dependency. View ranges with `rocprofv3 --marker-trace`.
+#include <dlfcn.h>
+
+namespace grapehip {
+namespace roctx {
+using PushFn = int (*)(const char*);
+using PopFn = int (*)();
+inline PushFn& push_fn() {
+  static PushFn fn = [] {
+    if (!getenv("GRAPEHIP_ROCTX")) return static_cast<PushFn>(nullptr);
+    void* h = dlopen("libroctx64.so", RTLD_NOW | RTLD_GLOBAL);
+    return h ? reinterpret_cast<PushFn>(dlsym(h, "roctxRangePushA"))
+             : nullptr;
+  }();
+  return fn;
+}
+inline PopFn& pop_fn() {
+  static PopFn fn = [] {
+    if (!getenv("GRAPEHIP_ROCTX")) return static_cast<PopFn>(nullptr);
+    void* h = dlopen("libroctx64.so", RTLD_NOW | RTLD_GLOBAL);
+    return h ? reinterpret_cast<PopFn>(dlsym(h, "roctxRangePop")) : nullptr;
+  }();
+  return fn;
+}
+}  // namespace roctx
+
+struct RangeMarker {
+  explicit RangeMarker(const char* name) {
+    if (roctx::push_fn()) roctx::push_fn()(name);
+  }
+  ~RangeMarker() {
+    if (roctx::pop_fn()) roctx::pop_fn()();
+  }
+};
+}  // namespace grapehip
+
 #define HIP_CHECK(expr)                                                     \
   do {                                                                      \
     hipError_t _e = (expr);                                                 \
