@@ -99,7 +99,9 @@ class SamplerApp {
         ctx.paths[it - ctx.walk_ids.begin()][m.hop] = m.gid;
     }
     records_.clear();
-    ctx.live = std::move(arrivals);
+    // arrivals JOIN the locally-advancing walks (replacing them dropped
+    // every local walk after its first hop)
+    ctx.live.insert(ctx.live.end(), arrivals.begin(), arrivals.end());
     advance(frag, ctx, mm);
   }
 

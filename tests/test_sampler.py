@@ -34,13 +34,20 @@ def test_walks_are_paths(eng):
     starts = np.arange(50, dtype=np.int64)
     r = eng.sample(g, starts, hops=3, strategy="random", seed=11)
     assert len(r["walk_ids"]) == 50
+    completed = 0
     for wid, path in zip(r["walk_ids"], r["paths"]):
         assert path[0] == starts[wid]
+        if path[-1] >= 0:
+            completed += 1
         for h in range(3):
             a, b = int(path[h]), int(path[h + 1])
-            if a == -1 or b == -1:
+            if b == -1:
+                # a dead end must really be a dead end
+                assert a == -1 or not adj.get(a), (wid, h, a)
                 continue
             assert b in adj.get(a, set()), (wid, h, a, b)
+    # dense random graph: nearly every walk must run its full length
+    assert completed >= 45, completed
 
 
 def test_deterministic(eng):
