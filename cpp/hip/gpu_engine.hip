@@ -2443,8 +2443,9 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
   iota_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(lab.data(), 0, nv_pad);
   changed_bm.zero(s);
   int rounds = 0;
+  bool use_dirty = false;  // iteration 0 recomputes everything
   for (int it = 0; it < iters; ++it) {
-    const uint32_t* dw = it == 0 ? nullptr : dirty.data();
+    const uint32_t* dw = use_dirty ? dirty.data() : nullptr;
     if (n_small)
       cdlp_small_kernel<<<grid_for(n_small * kWave), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, lab.data(), t_small.data(), n_small,
@@ -2478,6 +2479,18 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
           slice, ncclUint32, I.nccl, s));
     if (g_nch == 0) break;  // fixpoint: further rounds are no-ops
     if (it + 1 == iters) break;
+    // adaptive: building the dirty set costs an edge expansion over the
+    // changed rows — only worth it once changes are sparse (converging
+    // graphs); with heavy churn (early rounds, RMAT oscillation) a full
+    // recompute is cheaper than sweep + marking
+    uint64_t g_owned =
+        multi ? comm_->allreduce_sum(static_cast<uint64_t>(owned))
+              : owned;
+    use_dirty = g_nch < g_owned / 8;
+    if (!use_dirty) {
+      changed_bm.zero(s);
+      continue;
+    }
     // rebuild the dirty set from this round's changed rows
     dirty.zero(s);
     uint64_t qn = compact_frontier(I, changed_bm.data(), owned, g.v_begin,
