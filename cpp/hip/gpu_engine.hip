@@ -330,6 +330,171 @@ __global__ void scatter_edges_kernel(const uint32_t* src, const uint32_t* dst,
 }
 
 // ===========================================================================
+// Segmented adjacency sort (ascending dst per CSR row). Applied once after
+// the device CSR is built: ascending neighbor ids turn the pull kernels'
+// random gathers into locally-increasing streams (L2-friendlier) and give
+// deterministic adjacency order. Weighted rows sort (dst,weight) pairs
+// packed into u64. LDS bitonic for rows <= 4096, pow2-padded global
+// scratch bitonic for heavier rows.
+// ===========================================================================
+
+constexpr uint32_t kSegSortLds = 4096;
+
+__device__ __forceinline__ void bitonic_stage_u64(unsigned long long* a,
+                                                  uint32_t n, uint32_t k,
+                                                  uint32_t j) {
+  for (uint32_t i = threadIdx.x; i < n; i += blockDim.x) {
+    uint32_t ij = i ^ j;
+    if (ij > i) {
+      bool up = (i & k) == 0;
+      unsigned long long x = a[i], y = a[ij];
+      if ((x > y) == up) {
+        a[i] = y;
+        a[ij] = x;
+      }
+    }
+  }
+}
+
+__global__ void seg_sortbucket_kernel(const uint64_t* __restrict__ off,
+                                      uint32_t rows_n, uint32_t* lds_rows,
+                                      unsigned long long* c_lds,
+                                      uint32_t* big_rows,
+                                      unsigned long long* c_big) {
+  __shared__ uint32_t s_cnt[2];
+  __shared__ unsigned long long s_base[2];
+  uint32_t* lists[2] = {lds_rows, big_rows};
+  unsigned long long* gcnt[2] = {c_lds, c_big};
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t base = blockIdx.x * blockDim.x; base < rows_n;
+       base += stride) {
+    if (threadIdx.x < 2) s_cnt[threadIdx.x] = 0;
+    __syncthreads();
+    uint32_t r = base + threadIdx.x;
+    int b = -1;
+    uint32_t loc = 0;
+    if (r < rows_n) {
+      uint64_t n = off[r + 1] - off[r];
+      if (n >= 2) {
+        b = n <= kSegSortLds ? 0 : 1;
+        loc = atomicAdd(&s_cnt[b], 1u);
+      }
+    }
+    __syncthreads();
+    if (threadIdx.x < 2 && s_cnt[threadIdx.x])
+      s_base[threadIdx.x] =
+          atomicAdd(gcnt[threadIdx.x],
+                    static_cast<unsigned long long>(s_cnt[threadIdx.x]));
+    __syncthreads();
+    if (b >= 0) lists[b][s_base[b] + loc] = r;
+    __syncthreads();
+  }
+}
+
+__global__ void seg_sort_lds_kernel(const uint64_t* __restrict__ off,
+                                    uint32_t* __restrict__ dst,
+                                    float* __restrict__ w,
+                                    const uint32_t* __restrict__ rows,
+                                    uint64_t nrows) {
+  __shared__ unsigned long long s_a[kSegSortLds];
+  for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
+    uint32_t r = rows[i];
+    uint64_t b = off[r];
+    uint32_t n = static_cast<uint32_t>(off[r + 1] - b);
+    uint32_t cap = 2;
+    while (cap < n) cap <<= 1;
+    for (uint32_t k = threadIdx.x; k < cap; k += blockDim.x)
+      s_a[k] = k < n
+                   ? ((static_cast<unsigned long long>(dst[b + k]) << 32) |
+                      (w ? __float_as_uint(w[b + k]) : 0u))
+                   : ~0ull;
+    __syncthreads();
+    for (uint32_t k = 2; k <= cap; k <<= 1)
+      for (uint32_t j = k >> 1; j > 0; j >>= 1) {
+        bitonic_stage_u64(s_a, cap, k, j);
+        __syncthreads();
+      }
+    for (uint32_t k = threadIdx.x; k < n; k += blockDim.x) {
+      dst[b + k] = static_cast<uint32_t>(s_a[k] >> 32);
+      if (w) w[b + k] = __uint_as_float(static_cast<uint32_t>(s_a[k]));
+    }
+    __syncthreads();
+  }
+}
+
+__global__ void seg_bigcap_kernel(const uint64_t* __restrict__ off,
+                                  const uint32_t* __restrict__ rows,
+                                  uint64_t nrows,
+                                  uint32_t* __restrict__ caps) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * blockDim.x +
+                    threadIdx.x;
+       i < nrows; i += stride) {
+    uint64_t n = off[rows[i] + 1] - off[rows[i]];
+    uint64_t cap = 2;
+    while (cap < n) cap <<= 1;
+    caps[i] = static_cast<uint32_t>(cap);
+  }
+}
+
+__global__ void seg_bigpad_kernel(const uint64_t* __restrict__ off,
+                                  const uint32_t* __restrict__ dst,
+                                  const float* __restrict__ w,
+                                  const uint32_t* __restrict__ rows,
+                                  uint64_t nrows,
+                                  const uint64_t* __restrict__ pad_off,
+                                  unsigned long long* __restrict__ scratch) {
+  for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
+    uint32_t r = rows[i];
+    uint64_t b = off[r];
+    uint64_t n = off[r + 1] - b;
+    uint64_t pb = pad_off[i];
+    uint64_t cap = pad_off[i + 1] - pb;
+    for (uint64_t k = threadIdx.x; k < cap; k += blockDim.x)
+      scratch[pb + k] =
+          k < n ? ((static_cast<unsigned long long>(dst[b + k]) << 32) |
+                   (w ? __float_as_uint(w[b + k]) : 0u))
+                : ~0ull;
+  }
+}
+
+__global__ void seg_bigsort_kernel(const uint64_t* __restrict__ pad_off,
+                                   uint64_t nrows,
+                                   unsigned long long* __restrict__ scratch) {
+  for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
+    uint64_t pb = pad_off[i];
+    uint32_t cap = static_cast<uint32_t>(pad_off[i + 1] - pb);
+    unsigned long long* a = scratch + pb;
+    for (uint32_t k = 2; k <= cap; k <<= 1)
+      for (uint32_t j = k >> 1; j > 0; j >>= 1) {
+        bitonic_stage_u64(a, cap, k, j);
+        __syncthreads();
+      }
+  }
+}
+
+__global__ void seg_bigunpad_kernel(const uint64_t* __restrict__ off,
+                                    uint32_t* __restrict__ dst,
+                                    float* __restrict__ w,
+                                    const uint32_t* __restrict__ rows,
+                                    uint64_t nrows,
+                                    const uint64_t* __restrict__ pad_off,
+                                    const unsigned long long* __restrict__
+                                        scratch) {
+  for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
+    uint32_t r = rows[i];
+    uint64_t b = off[r];
+    uint64_t n = off[r + 1] - b;
+    uint64_t pb = pad_off[i];
+    for (uint64_t k = threadIdx.x; k < n; k += blockDim.x) {
+      dst[b + k] = static_cast<uint32_t>(scratch[pb + k] >> 32);
+      if (w) w[b + k] = __uint_as_float(
+                 static_cast<uint32_t>(scratch[pb + k]));
+    }
+  }
+}
+
+// ===========================================================================
 // CM-style load-balanced edge expansion (LDS-staged owner search).
 // Each block claims 256 consecutive work items (frontier entries or rows),
 // stages their frontier-edge prefix + adjacency base in LDS, then its 256
@@ -1187,6 +1352,43 @@ void build_csr_from_coo(const DeviceBuffer<uint32_t>& src,
   HIP_CHECK(hipStreamSynchronize(s));
 }
 
+void sort_csr_rows(const DeviceBuffer<uint64_t>& off,
+                   DeviceBuffer<uint32_t>& dst, DeviceBuffer<float>* w,
+                   uint32_t rows_n, hipStream_t s, ScanTemp& scan) {
+  if (rows_n == 0) return;
+  DeviceBuffer<uint32_t> lds_rows(rows_n), big_rows(rows_n);
+  DeviceBuffer<unsigned long long> cnts(2);
+  cnts.zero(s);
+  seg_sortbucket_kernel<<<grid_for(rows_n), kBlock, 0, s>>>(
+      off.data(), rows_n, lds_rows.data(), cnts.data() + 0, big_rows.data(),
+      cnts.data() + 1);
+  auto h = cnts.download(s);
+  if (h[0])
+    seg_sort_lds_kernel<<<std::min<uint64_t>(h[0], kMaxGrid), kBlock, 0,
+                          s>>>(off.data(), dst.data(),
+                               w ? w->data() : nullptr, lds_rows.data(),
+                               h[0]);
+  if (h[1]) {
+    DeviceBuffer<uint32_t> caps(h[1]);
+    DeviceBuffer<uint64_t> pad_off(h[1] + 1);
+    seg_bigcap_kernel<<<grid_for(h[1]), kBlock, 0, s>>>(
+        off.data(), big_rows.data(), h[1], caps.data());
+    uint64_t pad_total =
+        exclusive_scan(caps.data(), pad_off.data(), h[1], s, scan);
+    DeviceBuffer<unsigned long long> scratch(pad_total);
+    seg_bigpad_kernel<<<std::min<uint64_t>(h[1], kMaxGrid), kBlock, 0, s>>>(
+        off.data(), dst.data(), w ? w->data() : nullptr, big_rows.data(),
+        h[1], pad_off.data(), scratch.data());
+    seg_bigsort_kernel<<<std::min<uint64_t>(h[1], kMaxGrid), kBlock, 0,
+                         s>>>(pad_off.data(), h[1], scratch.data());
+    seg_bigunpad_kernel<<<std::min<uint64_t>(h[1], kMaxGrid), kBlock, 0,
+                          s>>>(off.data(), dst.data(),
+                               w ? w->data() : nullptr, big_rows.data(),
+                               h[1], pad_off.data(), scratch.data());
+  }
+  HIP_CHECK(hipStreamSynchronize(s));
+}
+
 }  // namespace
 
 std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
@@ -1240,6 +1442,8 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
 
   build_csr_from_coo(e_src, e_dst, e_w, n_local, g->v_begin, owned, weighted,
                      g->oe_off, g->oe_dst, g->oe_w, s, impl_->scan);
+  sort_csr_rows(g->oe_off, g->oe_dst, weighted ? &g->oe_w : nullptr, owned,
+                s, impl_->scan);
   if (directed && build_in_csr) {
     // regenerate the SAME edge stream, keeping edges whose dst is owned,
     // reversed — builds the in-CSR without materializing the global list
@@ -1255,6 +1459,8 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
     if (n_in > est) throw std::runtime_error("gen_synthetic: in overflow");
     build_csr_from_coo(e_src, e_dst, e_w, n_in, g->v_begin, owned, weighted,
                        g->ie_off, g->ie_dst, g->ie_w, s, impl_->scan);
+    sort_csr_rows(g->ie_off, g->ie_dst, weighted ? &g->ie_w : nullptr,
+                  owned, s, impl_->scan);
     g->has_in = true;
   }
   e_src.free();
@@ -1302,6 +1508,12 @@ std::unique_ptr<DeviceGraph> GpuContext::upload(const Fragment& frag) {
     g->ie_dst.upload(convert(frag.ie_dsts()), s);
     if (g->weighted) g->ie_w.upload(frag.ie_weights(), s);
   }
+  uint32_t owned_rows = g->owned();
+  sort_csr_rows(g->oe_off, g->oe_dst, g->weighted ? &g->oe_w : nullptr,
+                owned_rows, s, impl_->scan);
+  if (g->has_in)
+    sort_csr_rows(g->ie_off, g->ie_dst, g->weighted ? &g->ie_w : nullptr,
+                  owned_rows, s, impl_->scan);
   HIP_CHECK(hipStreamSynchronize(s));
   return g;
 }
