@@ -3510,7 +3510,9 @@ __global__ void lcc_triangle_heavy_kernel(
     const uint32_t cap =
         static_cast<uint32_t>((vn < un ? hoff[u + 1] : hoff[v + 1]) - hb);
     const uint32_t mask = cap - 1;
-    const bool stage = cap <= kLccStageSlots;
+    // staging costs ~cap LDS fills; only worth it when the enumeration
+    // amortizes them (staging everything regressed 3.2s -> 3.9s at 405M)
+    const bool stage = cap <= kLccStageSlots && en >= (cap >> 2);
     if (stage)
       for (uint32_t k = threadIdx.x; k < cap; k += blockDim.x)
         s_tab[k] = htab[hb + k];
