@@ -3477,18 +3477,18 @@ __global__ void lcc_tribucket_kernel(const uint64_t* __restrict__ goff,
   }
 }
 
-// second pass: one BLOCK per heavy edge. The larger endpoint's hash set
-// is staged into LDS when it fits (the wave-per-edge global-probe version
-// ran 93% wave-parked, PMC r01); 256 threads split the smaller list's
-// enumeration against LDS.
+// second pass: one wave per heavy edge; lanes stride the smaller list
 __global__ void lcc_triangle_heavy_kernel(
     const uint64_t* __restrict__ goff, const uint32_t* __restrict__ gdst,
     const uint64_t* __restrict__ hoff, const uint32_t* __restrict__ htab,
     const unsigned long long* __restrict__ heavy_q, uint64_t heavy_n,
     unsigned long long* __restrict__ T) {
-  __shared__ uint32_t s_tab[kLccStageSlots];
-  __shared__ unsigned long long s_hits;
-  for (uint64_t i = blockIdx.x; i < heavy_n; i += gridDim.x) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wpb = kBlock / kWave;
+  size_t wstride = static_cast<size_t>(gridDim.x) * wpb;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid;
+       i < heavy_n; i += wstride) {
     unsigned long long pk = heavy_q[i];
     uint32_t u = static_cast<uint32_t>(pk >> 32);
     uint32_t v = static_cast<uint32_t>(pk);
@@ -3496,46 +3496,36 @@ __global__ void lcc_triangle_heavy_kernel(
     uint32_t un = static_cast<uint32_t>(goff[u + 1] - ub);
     uint64_t vb = goff[v];
     uint32_t vn = static_cast<uint32_t>(goff[v + 1] - vb);
-    uint64_t eb, hb;
+    uint64_t eb, hb_big, hcap_end;
     uint32_t en;
     if (vn < un) {
       eb = vb;
       en = vn;
-      hb = hoff[u];
+      hb_big = hoff[u];
+      hcap_end = hoff[u + 1];
     } else {
       eb = ub;
       en = un;
-      hb = hoff[v];
+      hb_big = hoff[v];
+      hcap_end = hoff[v + 1];
     }
-    const uint32_t cap =
-        static_cast<uint32_t>((vn < un ? hoff[u + 1] : hoff[v + 1]) - hb);
-    const uint32_t mask = cap - 1;
-    // staging costs ~cap LDS fills; only worth it when the enumeration
-    // amortizes them (staging everything regressed 3.2s -> 3.9s at 405M)
-    const bool stage = cap <= kLccStageSlots && en >= (cap >> 2);
-    if (stage)
-      for (uint32_t k = threadIdx.x; k < cap; k += blockDim.x)
-        s_tab[k] = htab[hb + k];
-    if (threadIdx.x == 0) s_hits = 0;
-    __syncthreads();
-    const uint32_t* gt = htab + hb;
+    const uint64_t mask = (hcap_end - hb_big) - 1;
+    const uint32_t* t = htab + hb_big;
     unsigned long long hits = 0;
-    for (uint32_t e = threadIdx.x; e < en; e += blockDim.x) {
+    for (uint32_t e = lane; e < en; e += kWave) {
       uint32_t w = gdst[eb + e];
-      bool hit = stage ? lcc_probe_lds(s_tab, mask, w)
-                       : lcc_probe(gt, mask, w);
-      if (hit) {
+      if (lcc_probe(t, mask, w)) {
         ++hits;
         atomicAdd(&T[w], 1ull);
       }
     }
-    if (hits) atomicAdd(&s_hits, hits);
-    __syncthreads();
-    if (threadIdx.x == 0 && s_hits) {
-      atomicAdd(&T[u], s_hits);
-      atomicAdd(&T[v], s_hits);
+#pragma unroll
+    for (int d = 32; d > 0; d >>= 1)
+      hits += __shfl_down(hits, d, 64);
+    if (lane == 0 && hits) {
+      atomicAdd(&T[u], hits);
+      atomicAdd(&T[v], hits);
     }
-    __syncthreads();
   }
 }
 
@@ -4188,8 +4178,7 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
               static_cast<unsigned long long>(oriented_total),
               static_cast<unsigned long long>(hn));
     if (hn)
-      lcc_triangle_heavy_kernel<<<std::min<uint64_t>(hn, kMaxGrid), kBlock,
-                                  0, s>>>(
+      lcc_triangle_heavy_kernel<<<grid_for(hn * kWave), kBlock, 0, s>>>(
           goff.data(), gdst.data(), hoff.data(), htab.data(),
           heavy_q.data(), hn, Tcnt.data());
   }
