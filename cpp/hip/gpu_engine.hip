@@ -284,23 +284,32 @@ __global__ void gen_edges_kernel(uint64_t ne, uint64_t seed, int scale,
     // materialize both orientations for undirected storage; keep owned src.
     // reverse=true keeps edges by OWNED DST as (d, s) pairs — the same
     // deterministic edge stream builds the in-CSR of a directed graph.
+    // out_src==nullptr: counting pass only (RMAT mass is skewed toward low
+    // vids, so per-rank shares are NOT ne/world — rank 0 of an 8-way run
+    // owns far more; exact counts size the buffers)
     if (!reverse && s >= my_begin && s < my_end) {
       uint64_t pos = atomicAdd(out_cnt, 1ull);
-      out_src[pos] = s;
-      out_dst[pos] = d;
-      if (weighted) out_w[pos] = w;
+      if (out_src) {
+        out_src[pos] = s;
+        out_dst[pos] = d;
+        if (weighted) out_w[pos] = w;
+      }
     }
     if (undirected && d >= my_begin && d < my_end && d != s) {
       uint64_t pos = atomicAdd(out_cnt, 1ull);
-      out_src[pos] = d;
-      out_dst[pos] = s;
-      if (weighted) out_w[pos] = w;
+      if (out_src) {
+        out_src[pos] = d;
+        out_dst[pos] = s;
+        if (weighted) out_w[pos] = w;
+      }
     }
     if (reverse && d >= my_begin && d < my_end) {
       uint64_t pos = atomicAdd(out_cnt, 1ull);
-      out_src[pos] = d;
-      out_dst[pos] = s;
-      if (weighted) out_w[pos] = w;
+      if (out_src) {
+        out_src[pos] = d;
+        out_dst[pos] = s;
+        if (weighted) out_w[pos] = w;
+      }
     }
   }
 }
@@ -1418,14 +1427,25 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
   uint32_t t_abc = static_cast<uint32_t>((a + b + c) * 65536.0);
 
   hipStream_t s = impl_->compute;
-  // stored-edge estimate: undirected stores both orientations
-  double orient = directed ? 1.0 : 2.0;
-  uint64_t est = static_cast<uint64_t>(orient * ne / world_ * 1.1) + (1 << 20);
-  if (world_ == 1) est = static_cast<uint64_t>(orient * ne) + 16;
-  DeviceBuffer<uint32_t> e_src(est), e_dst(est);
-  DeviceBuffer<float> e_w(weighted ? est : 0);
+  // exact per-rank edge count via a counting pass (see gen_edges_kernel)
   DeviceBuffer<unsigned long long> cnt(1);
   cnt.zero(s);
+  uint64_t est;
+  if (world_ == 1) {
+    est = static_cast<uint64_t>((directed ? 1.0 : 2.0) * ne) + 16;
+  } else {
+    gen_edges_kernel<<<kMaxGrid, kBlock, 0, s>>>(
+        ne, seed, scale, g->nv_global, t_a, t_ab, t_abc,
+        g->v_begin, g->v_end, !directed, false, false, nullptr, nullptr,
+        nullptr, cnt.data());
+    unsigned long long c = 0;
+    HIP_CHECK(hipMemcpyAsync(&c, cnt.data(), 8, hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    est = c + 16;
+    cnt.zero(s);
+  }
+  DeviceBuffer<uint32_t> e_src(est), e_dst(est);
+  DeviceBuffer<float> e_w(weighted ? est : 0);
   if (getenv("GRAPEHIP_DEBUG"))
     fprintf(stderr, "[gen] nv=%u ne=%lu est=%lu scale=%d\n", g->nv_global,
             (unsigned long)ne, (unsigned long)est, scale);
@@ -1447,6 +1467,27 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
   if (directed && build_in_csr) {
     // regenerate the SAME edge stream, keeping edges whose dst is owned,
     // reversed — builds the in-CSR without materializing the global list
+    if (world_ > 1) {
+      // in-edge ownership skews differently than out; recount exactly
+      cnt.zero(s);
+      gen_edges_kernel<<<kMaxGrid, kBlock, 0, s>>>(
+          ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, g->v_begin,
+          g->v_end, false, true, false, nullptr, nullptr, nullptr,
+          cnt.data());
+      unsigned long long c_in = 0;
+      HIP_CHECK(hipMemcpyAsync(&c_in, cnt.data(), 8, hipMemcpyDeviceToHost,
+                               s));
+      HIP_CHECK(hipStreamSynchronize(s));
+      if (c_in + 16 > est) {
+        est = c_in + 16;
+        e_src.free();
+        e_dst.free();
+        e_w.free();
+        e_src.resize(est);
+        e_dst.resize(est);
+        if (weighted) e_w.resize(est);
+      }
+    }
     cnt.zero(s);
     gen_edges_kernel<<<kMaxGrid, kBlock, 0, s>>>(
         ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, g->v_begin,
@@ -2536,6 +2577,10 @@ __global__ void cdlp_commit_kernel(const uint32_t* __restrict__ next,
                                    uint32_t* __restrict__ lab,
                                    DevBitmap changed,
                                    unsigned long long* __restrict__ nch) {
+  __shared__ unsigned long long s_nch;
+  if (threadIdx.x == 0) s_nch = 0;
+  __syncthreads();
+  unsigned long long my = 0;
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < owned;
        r += stride) {
@@ -2545,9 +2590,12 @@ __global__ void cdlp_commit_kernel(const uint32_t* __restrict__ next,
     if (nv != lab[v]) {
       lab[v] = nv;
       changed.set_once(r);
-      atomicAdd(nch, 1ull);
+      ++my;
     }
   }
+  if (my) atomicAdd(&s_nch, my);
+  __syncthreads();
+  if (threadIdx.x == 0 && s_nch) atomicAdd(nch, s_nch);
 }
 
 // mark rows whose label multiset can have changed: every neighbor of a
