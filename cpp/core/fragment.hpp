@@ -220,7 +220,9 @@ class Fragment {
       std::vector<std::atomic<eid_t>> deg(F.ivnum_ + 1);
       for (auto& d : deg) d.store(0, std::memory_order_relaxed);
       for (auto& blob : blobs) {
-        auto [p, n] = as_edges(blob);
+        auto pe = as_edges(blob);
+        const GidEdge* p = pe.first;
+        size_t n = pe.second;
         parallel_for(0, n, [&](size_t i) {
           deg[P.lid(p[i].src)].fetch_add(1, std::memory_order_relaxed);
         }, 8192);
@@ -236,7 +238,9 @@ class Fragment {
       dst.resize(run);
       if (weighted) wts.resize(run);
       for (auto& blob : blobs) {
-        auto [p, n] = as_edges(blob);
+        auto pe = as_edges(blob);
+        const GidEdge* p = pe.first;
+        size_t n = pe.second;
         parallel_for(0, n, [&](size_t i) {
           vid_t u = P.lid(p[i].src);
           eid_t slot = deg[u].fetch_add(1, std::memory_order_relaxed);

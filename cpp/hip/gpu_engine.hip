@@ -2818,7 +2818,6 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
 // ---------------------------------------------------------------------------
 
 constexpr uint32_t kLccLdsSlots = 8192;  // 32 KB label-only LDS hash
-constexpr uint32_t kLccMidDeg = kLccLdsSlots / 2;
 constexpr uint32_t kLccSortLds = 4096;   // 16 KB LDS bitonic bound
 
 // orientation: keep u in O(v) iff (D[u],u) > (D[v],v)

@@ -101,7 +101,7 @@ class DeviceBuffer {
     n_ = n;
   }
   void free() {
-    if (p_) hipFree(p_);
+    if (p_) (void)hipFree(p_);
     p_ = nullptr;
     n_ = 0;
   }
@@ -143,8 +143,8 @@ class DeviceCounter {
     HIP_CHECK(hipHostMalloc(&h_, sizeof(uint64_t)));
   }
   ~DeviceCounter() {
-    if (d_) hipFree(d_);
-    if (h_) hipHostFree(h_);
+    if (d_) (void)hipFree(d_);
+    if (h_) (void)hipHostFree(h_);
   }
   DeviceCounter(const DeviceCounter&) = delete;
   DeviceCounter& operator=(const DeviceCounter&) = delete;
@@ -169,7 +169,7 @@ class Stream {
  public:
   Stream() { HIP_CHECK(hipStreamCreateWithFlags(&s_, hipStreamNonBlocking)); }
   ~Stream() {
-    if (s_) hipStreamDestroy(s_);
+    if (s_) (void)hipStreamDestroy(s_);
   }
   Stream(const Stream&) = delete;
   Stream& operator=(const Stream&) = delete;
@@ -185,7 +185,7 @@ class Event {
  public:
   Event() { HIP_CHECK(hipEventCreateWithFlags(&e_, hipEventDisableTiming)); }
   ~Event() {
-    if (e_) hipEventDestroy(e_);
+    if (e_) (void)hipEventDestroy(e_);
   }
   Event(const Event&) = delete;
   Event& operator=(const Event&) = delete;
