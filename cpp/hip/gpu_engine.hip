@@ -3336,7 +3336,7 @@ __device__ __forceinline__ bool lcc_probe(const uint32_t* __restrict__ t,
 //   staged rows: block-per-row; the row's hash set is staged into LDS once
 //   and every thread's probes hit LDS (the global-probe version spent its
 //   time on scattered 4B HBM reads; witness atomics measured ~5%)
-constexpr uint32_t kLccWeeRows = 256;
+constexpr uint32_t kLccWeeRows = 768;  // swept 32/256/768/2048: 3206/2750/2607/2713 ms at datagen-9_0
 constexpr uint32_t kLccStageSlots = 8192;  // 32 KB LDS
 
 __global__ void lcc_tri_wee_kernel(const uint64_t* __restrict__ goff,
