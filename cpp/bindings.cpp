@@ -10,6 +10,7 @@
 
 #include <chrono>
 #include <fstream>
+#include <limits>
 #include <memory>
 #include <optional>
 #include <unordered_set>
@@ -205,21 +206,71 @@ py::dict run_timed(PyEngine& eng, RunFn&& run) {
 }
 
 #ifdef GRAPEHIP_WITH_HIP
-py::dict gpu_dict(const GpuRunResult& r, const DeviceGraph& g, bool is_i64,
-                  bool with_values) {
+// device id <-> oid translation: identity maps use oids as device ids;
+// hashmap maps are densely renumbered on upload (dev = fid*slice + lid)
+struct DevIdMap {
+  const Fragment* frag = nullptr;  // null => identity (dev id == oid)
+  uint32_t slice = 0;
+  int64_t to_oid(int64_t dev) const {
+    if (!frag) return dev;
+    const IdParser& P = frag->parser();
+    fid_t f = static_cast<fid_t>(dev / slice);
+    vid_t lid = static_cast<vid_t>(dev % slice);
+    return frag->vm().get_oid(P.gid(f, lid));
+  }
+  uint32_t to_dev(int64_t oid) const {
+    if (!frag) return static_cast<uint32_t>(oid);
+    vid_t gid;
+    if (!frag->vm().get_gid(oid, &gid))
+      throw std::runtime_error("unknown vertex id " + std::to_string(oid));
+    const IdParser& P = frag->parser();
+    return static_cast<uint32_t>(
+        static_cast<uint64_t>(P.fid(gid)) * slice + P.lid(gid));
+  }
+};
+
+DevIdMap dev_id_map(const PyGraph& g) {
+  DevIdMap m;
+  if (g.frag && g.frag->vm().idxer() != IdxerKind::kIdentity) {
+    m.frag = g.frag.get();
+    m.slice = g.dev->seg_host.size() > 1
+                  ? g.dev->seg_host[1] - g.dev->seg_host[0]
+                  : g.dev->nv_global;
+  }
+  return m;
+}
+
+py::dict gpu_dict(const GpuRunResult& r, const PyGraph& pg, bool is_i64,
+                  bool with_values, bool labels_are_ids = false) {
+  const DeviceGraph& g = *pg.dev;
   py::dict out;
   out["rounds"] = r.rounds;
   out["seconds"] = r.seconds;
   out["traversed_edges"] = r.traversed_edges;
   if (with_values) {
-    uint32_t owned = g.owned();
-    py::array_t<int64_t> oids(owned);
+    DevIdMap m = dev_id_map(pg);
+    // hashmap uploads pad the owned range; emit only real vertices
+    uint32_t n = m.frag ? m.frag->ivnum() : g.owned();
+    py::array_t<int64_t> oids(n);
     auto* p = oids.mutable_data();
-    for (uint32_t i = 0; i < owned; ++i)
-      p[i] = static_cast<int64_t>(g.v_begin) + i;
+    for (uint32_t i = 0; i < n; ++i)
+      p[i] = m.frag ? m.frag->lid2oid(i)
+                    : static_cast<int64_t>(g.v_begin) + i;
     out["oids"] = oids;
-    out["values"] =
-        is_i64 ? py::object(to_np(r.i64)) : py::object(to_np(r.f64));
+    if (is_i64) {
+      py::array_t<int64_t> vals(n);
+      auto* q = vals.mutable_data();
+      for (uint32_t i = 0; i < n; ++i)
+        q[i] = (labels_are_ids && m.frag &&
+                r.i64[i] != std::numeric_limits<int64_t>::max())
+                   ? m.to_oid(r.i64[i])
+                   : r.i64[i];
+      out["values"] = vals;
+    } else {
+      py::array_t<double> vals(n);
+      std::memcpy(vals.mutable_data(), r.f64.data(), n * 8);
+      out["values"] = vals;
+    }
   }
   return out;
 }
@@ -483,9 +534,10 @@ PYBIND11_MODULE(_core, m) {
                GpuRunResult r;
                {
                  py::gil_scoped_release rel;
-                 r = eng.gpu->bfs(*g.dev, source, values);
+                 r = eng.gpu->bfs(*g.dev, dev_id_map(g).to_dev(source),
+                                  values);
                }
-               return gpu_dict(r, *g.dev, true, values);
+               return gpu_dict(r, g, true, values);
              }
 #endif
              BFSApp app;
@@ -512,9 +564,10 @@ PYBIND11_MODULE(_core, m) {
                GpuRunResult r;
                {
                  py::gil_scoped_release rel;
-                 r = eng.gpu->sssp(*g.dev, source, delta, values);
+                 r = eng.gpu->sssp(*g.dev, dev_id_map(g).to_dev(source),
+                                   delta, values);
                }
-               return gpu_dict(r, *g.dev, false, values);
+               return gpu_dict(r, g, false, values);
              }
 #endif
              SSSPApp app;
@@ -543,7 +596,7 @@ PYBIND11_MODULE(_core, m) {
                  py::gil_scoped_release rel;
                  r = eng.gpu->pagerank(*g.dev, damping, iters, tol, values);
                }
-               return gpu_dict(r, *g.dev, false, values);
+               return gpu_dict(r, g, false, values);
              }
 #endif
              PageRankApp app;
@@ -568,7 +621,7 @@ PYBIND11_MODULE(_core, m) {
                  py::gil_scoped_release rel;
                  r = eng.gpu->wcc(*g.dev, values);
                }
-               return gpu_dict(r, *g.dev, true, values);
+               return gpu_dict(r, g, true, values, /*labels_are_ids=*/true);
              }
 #endif
              WCCApp app;
@@ -595,7 +648,7 @@ PYBIND11_MODULE(_core, m) {
                  py::gil_scoped_release rel;
                  r = eng.gpu->cdlp(*g.dev, iters, values);
                }
-               return gpu_dict(r, *g.dev, true, values);
+               return gpu_dict(r, g, true, values, /*labels_are_ids=*/true);
              }
 #endif
              CDLPApp app;
@@ -622,7 +675,7 @@ PYBIND11_MODULE(_core, m) {
                  py::gil_scoped_release rel;
                  r = eng.gpu->lcc(*g.dev, values);
                }
-               return gpu_dict(r, *g.dev, false, values);
+               return gpu_dict(r, g, false, values);
              }
 #endif
              LCCApp app;

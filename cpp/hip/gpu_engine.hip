@@ -16,6 +16,7 @@
 
 #include <algorithm>
 #include <cmath>
+#include <functional>
 #include <cstring>
 #include <unordered_map>
 
@@ -1407,6 +1408,7 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
     throw std::runtime_error("gen_synthetic: nv must fit in 32 bits");
   auto g = std::make_unique<DeviceGraph>();
   g->nv_global = static_cast<uint32_t>(nv);
+  g->nv_real = nv;
   g->directed = directed;
   g->weighted = weighted;
   g->input_edges = ne;
@@ -1419,6 +1421,7 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
   g->v_begin = g->seg_host[rank_];
   g->v_end = g->seg_host[rank_ + 1];
   uint32_t owned = g->owned();
+  g->owned_real = owned;
 
   int scale = 0;
   while ((1ull << scale) < nv) ++scale;
@@ -1514,38 +1517,79 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
 }
 
 std::unique_ptr<DeviceGraph> GpuContext::upload(const Fragment& frag) {
-  if (frag.vm().idxer() != IdxerKind::kIdentity)
-    throw std::runtime_error(
-        "GPU path requires the identity vertex map (dense 0..V-1 oids); "
-        "renumber the graph or use the CPU engine");
   auto g = std::make_unique<DeviceGraph>();
-  g->nv_global = static_cast<uint32_t>(frag.total_vertices());
   g->directed = frag.directed();
   g->weighted = frag.has_weights();
   g->has_in = frag.has_in_csr();
   g->input_edges = frag.input_edges();
   g->total_edges = frag.total_edges();
   g->local_edges = frag.local_edges();
-  const auto& seg64 = frag.vm().segments();
-  g->seg_host.assign(seg64.begin(), seg64.end());
-  g->v_begin = g->seg_host[frag.fid()];
-  g->v_end = g->seg_host[frag.fid() + 1];
+  g->nv_real = frag.total_vertices();
   hipStream_t s = impl_->compute;
+  const bool identity = frag.vm().idxer() == IdxerKind::kIdentity;
+
+  // device id mapping. identity maps use oids directly (segmented
+  // ownership from the vertex map). Arbitrary-oid (hashmap) maps are
+  // DENSELY RENUMBERED on upload: device id = fid * slice + lid with one
+  // uniform slice = max fragment vnum, so owner lookup stays v/slice and
+  // RCCL slices stay uniform (the reference needs a device hashmap,
+  // cuda/vertex_map/device_vertex_map.h, for the same capability).
+  // Padding rows [ivnum, slice) get empty adjacency.
+  std::function<uint32_t(vid_t)> lid2dev;
+  uint32_t slice;
+  if (identity) {
+    const auto& seg64 = frag.vm().segments();
+    g->seg_host.assign(seg64.begin(), seg64.end());
+    g->nv_global = static_cast<uint32_t>(frag.total_vertices());
+    g->v_begin = g->seg_host[frag.fid()];
+    g->v_end = g->seg_host[frag.fid() + 1];
+    slice = g->seg_host.size() > 1
+                ? static_cast<uint32_t>(g->seg_host[1] - g->seg_host[0])
+                : g->nv_global;
+    lid2dev = [&frag](vid_t lid) {
+      return static_cast<uint32_t>(frag.lid2oid(lid));
+    };
+  } else {
+    uint64_t vmax = frag.ivnum();
+    if (comm_ && world_ > 1) vmax = comm_->allreduce_max(vmax);
+    slice = static_cast<uint32_t>(vmax);
+    g->nv_global = static_cast<uint32_t>(
+        static_cast<uint64_t>(slice) * world_);
+    g->seg_host.resize(world_ + 1);
+    for (int f = 0; f <= world_; ++f)
+      g->seg_host[f] = static_cast<uint32_t>(
+          static_cast<uint64_t>(f) * slice);
+    g->v_begin = g->seg_host[frag.fid()];
+    g->v_end = g->seg_host[frag.fid() + 1];
+    const IdParser& P = frag.parser();
+    lid2dev = [&frag, &P, slice](vid_t lid) {
+      vid_t gid = frag.lid2gid(lid);
+      return static_cast<uint32_t>(
+          static_cast<uint64_t>(P.fid(gid)) * slice + P.lid(gid));
+    };
+  }
+  g->owned_real = identity ? g->owned() : frag.ivnum();
   g->seg.upload(g->seg_host, s);
 
-  // rewrite dst lids -> global vids on host (parallel), then upload
   auto convert = [&](const std::vector<vid_t>& dst_lid) {
     std::vector<uint32_t> out(dst_lid.size());
     parallel_for(0, dst_lid.size(), [&](size_t i) {
-      out[i] = static_cast<uint32_t>(frag.lid2oid(dst_lid[i]));
+      out[i] = lid2dev(dst_lid[i]);
     }, 8192);
     return out;
   };
-  g->oe_off.upload(frag.oe_offsets(), s);
+  auto pad_offsets = [&](const std::vector<eid_t>& off) {
+    // extend to the padded owned range with empty rows
+    uint32_t rows = g->v_end - g->v_begin;
+    std::vector<eid_t> out(off);
+    out.resize(rows + 1, off.empty() ? 0 : off.back());
+    return out;
+  };
+  g->oe_off.upload(pad_offsets(frag.oe_offsets()), s);
   g->oe_dst.upload(convert(frag.oe_dsts()), s);
   if (g->weighted) g->oe_w.upload(frag.oe_weights(), s);
   if (g->has_in) {
-    g->ie_off.upload(frag.ie_offsets(), s);
+    g->ie_off.upload(pad_offsets(frag.ie_offsets()), s);
     g->ie_dst.upload(convert(frag.ie_dsts()), s);
     if (g->weighted) g->ie_w.upload(frag.ie_weights(), s);
   }
@@ -2020,7 +2064,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
   uint32_t slice = g.seg_host.size() > 1 ? g.seg_host[1] - g.seg_host[0]
                                          : g.nv_global;
   bool multi = world_ > 1;
-  const double N = static_cast<double>(g.nv_global);
+  const double N = static_cast<double>(g.nv_real);
 
   // pull path: undirected symmetric storage (out-CSR == neighborhood) or
   // directed with an in-CSR. Fallback: fp64-atomic push.
@@ -2046,8 +2090,8 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
   for (int it = 0; it < iters; ++it) {
     // dangling mass (vertices with zero out-degree), own slice
     d_dangling.zero(s);
-    pr_dangling_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-        rank_arr.data(), g.oe_off.data(), owned, g.v_begin,
+    pr_dangling_kernel<<<grid_for(g.owned_real), kBlock, 0, s>>>(
+        rank_arr.data(), g.oe_off.data(), g.owned_real, g.v_begin,
         d_dangling.data());
     double local_dangling = 0;
     HIP_CHECK(hipMemcpyAsync(&local_dangling, d_dangling.data(), 8,
@@ -2062,8 +2106,8 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
     }
 
     // contributions of owned rows (rank/outdeg), written into the own slice
-    pr_contrib_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-        rank_arr.data(), g.oe_off.data(), owned, g.v_begin,
+    pr_contrib_kernel<<<grid_for(g.owned_real), kBlock, 0, s>>>(
+        rank_arr.data(), g.oe_off.data(), g.owned_real, g.v_begin,
         contrib.data() + g.v_begin);
     double base = (1.0 - damping) / N + damping * dangling / N;
     if (pull) {

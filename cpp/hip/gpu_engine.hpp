@@ -20,8 +20,11 @@ struct ncclComm;  // fwd (rccl.h included in the .hip TU)
 namespace grapehip {
 
 struct DeviceGraph {
-  uint32_t nv_global = 0;
+  uint32_t nv_global = 0;   // device id space (padded for dense renumber)
+  uint64_t nv_real = 0;     // true vertex count (PageRank N etc.)
   uint32_t v_begin = 0, v_end = 0;
+  uint32_t owned_real = 0;  // rows with real vertices (<= owned(); dense
+                            // renumbering pads the tail of each slice)
   bool directed = false, weighted = false, has_in = false;
   uint64_t local_edges = 0, total_edges = 0, input_edges = 0;
   std::vector<uint32_t> seg_host;

@@ -198,3 +198,54 @@ def test_gpu_synthetic_directed_incsr(eng):
     assert len(r["values"]) == 50000
     r2 = eng.pagerank(g, 0.85, 5)
     assert abs(r2["values"].sum() - 1.0) < 1e-5
+
+
+def test_gpu_arbitrary_oids(eng):
+    # hashmap vertex map: sparse non-dense oids, densely renumbered on
+    # upload (reference needs cuda_hashmap's DeviceVertexMap for this)
+    rng = np.random.default_rng(71)
+    nv = 3000
+    oids = np.sort(rng.choice(10**9, size=nv, replace=False)).astype(np.int64)
+    si = rng.integers(0, nv, 25000)
+    di = rng.integers(0, nv, 25000)
+    keep = si != di
+    si, di = si[keep], di[keep]
+    w = (rng.random(len(si), dtype=np.float32) * 9 + 1)
+    g = eng.load_edges(oids[si], oids[di], weights=w, directed=False,
+                       vertex_oids=oids)
+    src_oid = int(oids[7])
+    r = eng.bfs(g, src_oid)
+    got = dict(zip(r["oids"].tolist(), r["values"].tolist()))
+    exp = bfs_oracle(nv, si, di, 7, directed=False)
+    for j in range(nv):
+        assert got[int(oids[j])] == exp[j], j
+    # pagerank sums to 1 over REAL vertices (padding must not leak mass)
+    rp = eng.pagerank(g, 0.85, 10)
+    assert abs(rp["values"].sum() - 1.0) < 1e-5
+    expect_pr = pagerank_oracle(nv, si, di, 0.85, 10, directed=False)
+    gotp = dict(zip(rp["oids"].tolist(), rp["values"].tolist()))
+    for j in range(0, nv, 7):
+        assert abs(gotp[int(oids[j])] - expect_pr[j]) < 3e-5 * max(
+            expect_pr[j], 1e-9), j
+    # wcc labels come back as REAL oids forming consistent components
+    rw = eng.wcc(g)
+    labs = dict(zip(rw["oids"].tolist(), rw["values"].tolist()))
+    exp_w = wcc_oracle(nv, si, di)
+    fwd = {}
+    for j in range(nv):
+        assert labs[int(oids[j])] in labs  # label is a real vertex oid
+        assert fwd.setdefault(exp_w[j], labs[int(oids[j])]) == \
+            labs[int(oids[j])], j
+    # cdlp labels are oids too
+    rc = eng.cdlp(g, 4)
+    labc = dict(zip(rc["oids"].tolist(), rc["values"].tolist()))
+    oid_to_dense = {int(o): j for j, o in enumerate(oids)}
+    exp_c = cdlp_oracle(nv, si, di, 4, directed=False)
+    for j in range(nv):
+        assert labc[int(oids[j])] == int(oids[exp_c[j]]), j
+    # lcc values
+    rl = eng.lcc(g)
+    gotl = dict(zip(rl["oids"].tolist(), rl["values"].tolist()))
+    exp_l = lcc_oracle(nv, si, di, directed=False)
+    for j in range(0, nv, 11):
+        assert abs(gotl[int(oids[j])] - exp_l[j]) < 1e-9, j
