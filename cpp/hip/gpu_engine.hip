@@ -902,11 +902,15 @@ __global__ void pr_contrib_kernel(const double* __restrict__ rank,
 }
 
 __global__ void pr_apply_kernel(double* __restrict__ rank,
-                                const double* __restrict__ acc, double base,
-                                double damping, uint32_t v_begin,
-                                uint32_t owned,
+                                const double* __restrict__ acc,
+                                const double* __restrict__ dangling,
+                                double inv_n, double damping,
+                                uint32_t v_begin, uint32_t owned,
                                 double* __restrict__ l1_delta) {
   __shared__ double s_wave[kBlock / kWave];
+  // base recomputed on device so the whole iteration is hipGraph-capturable
+  const double base =
+      (1.0 - damping) * inv_n + damping * (*dangling) * inv_n;
   double d1 = 0;
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < owned;
@@ -2087,29 +2091,34 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
 
   fill(rank_arr.data(), 1.0 / N, nv_pad, s);
   int rounds = 0;
-  for (int it = 0; it < iters; ++it) {
-    // dangling mass (vertices with zero out-degree), own slice
+  const double inv_n = 1.0 / N;
+
+  // one iteration, recorded as a stream of kernels. Single-GPU fixed-iter
+  // runs capture it into a hipGraph once and replay (the iteration is
+  // fully device-side: the dangling sum feeds pr_apply through a device
+  // scalar); multi-GPU and tol-convergence runs stay on the plain path
+  // (RCCL ops + per-iteration host reads).
+  auto record_iteration = [&]() {
     d_dangling.zero(s);
     pr_dangling_kernel<<<grid_for(g.owned_real), kBlock, 0, s>>>(
         rank_arr.data(), g.oe_off.data(), g.owned_real, g.v_begin,
         d_dangling.data());
-    double local_dangling = 0;
-    HIP_CHECK(hipMemcpyAsync(&local_dangling, d_dangling.data(), 8,
-                             hipMemcpyDeviceToHost, s));
-    HIP_CHECK(hipStreamSynchronize(s));
-    double dangling = local_dangling;
     if (multi) {
+      // fold the global dangling sum on the host (outside captures)
+      double local_dangling = 0;
+      HIP_CHECK(hipMemcpyAsync(&local_dangling, d_dangling.data(), 8,
+                               hipMemcpyDeviceToHost, s));
+      HIP_CHECK(hipStreamSynchronize(s));
       std::vector<double> all(world_);
       comm_->allgather(&local_dangling, 8, all.data());
-      dangling = 0;
+      double dangling = 0;
       for (double x : all) dangling += x;
+      HIP_CHECK(hipMemcpyAsync(d_dangling.data(), &dangling, 8,
+                               hipMemcpyHostToDevice, s));
     }
-
-    // contributions of owned rows (rank/outdeg), written into the own slice
     pr_contrib_kernel<<<grid_for(g.owned_real), kBlock, 0, s>>>(
         rank_arr.data(), g.oe_off.data(), g.owned_real, g.v_begin,
         contrib.data() + g.v_begin);
-    double base = (1.0 - damping) / N + damping * dangling / N;
     if (pull) {
       if (multi)
         NCCL_CHECK(ncclAllGather(contrib.data() + static_cast<uint64_t>(
@@ -2129,10 +2138,6 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
                                kBlock, 0, s>>>(
             pull_off, pull_dst, contrib.data(), g.rows_large.data(),
             g.n_large, g.v_begin, acc.data());
-      if (tol > 0) d_l1.zero(s);
-      pr_apply_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-          rank_arr.data(), acc.data(), base, damping, g.v_begin, owned,
-          tol > 0 ? d_l1.data() : nullptr);
     } else {
       acc.zero(s);
       int nchunks = static_cast<int>((owned + kBlock - 1) / kBlock);
@@ -2144,11 +2149,29 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
         NCCL_CHECK(ncclReduceScatter(
             acc.data(), acc.data() + static_cast<uint64_t>(rank_) * slice,
             slice, ncclDouble, ncclSum, I.nccl, s));
-      if (tol > 0) d_l1.zero(s);
-      pr_apply_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-          rank_arr.data(), acc.data(), base, damping, g.v_begin, owned,
-          tol > 0 ? d_l1.data() : nullptr);
     }
+    if (tol > 0) d_l1.zero(s);
+    pr_apply_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+        rank_arr.data(), acc.data(), d_dangling.data(), inv_n, damping,
+        g.v_begin, owned, tol > 0 ? d_l1.data() : nullptr);
+  };
+
+  hipGraphExec_t graph_exec = nullptr;
+  const bool capture = !multi && tol <= 0;
+  if (capture) {
+    hipGraph_t graph = nullptr;
+    HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal));
+    record_iteration();
+    HIP_CHECK(hipStreamEndCapture(s, &graph));
+    HIP_CHECK(hipGraphInstantiate(&graph_exec, graph, nullptr, nullptr, 0));
+    HIP_CHECK(hipGraphDestroy(graph));
+  }
+
+  for (int it = 0; it < iters; ++it) {
+    if (capture)
+      HIP_CHECK(hipGraphLaunch(graph_exec, s));
+    else
+      record_iteration();
     ++rounds;
     if (tol > 0) {
       // competitor-equivalent convergence (reference pagerank_local.h):
@@ -2167,6 +2190,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
     }
   }
   HIP_CHECK(hipDeviceSynchronize());
+  if (graph_exec) HIP_CHECK(hipGraphExecDestroy(graph_exec));
   if (comm_) comm_->barrier();
   double t1 = wall_s();
 
