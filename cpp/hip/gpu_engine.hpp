@@ -41,6 +41,10 @@ struct DeviceGraph {
   uint64_t n_small = 0, n_mid = 0, n_large = 0;
   bool buckets_built = false;
   uint32_t owned() const { return v_end - v_begin; }
+  // cached hipGraph of one PageRank iteration (single-GPU fixed-iter path)
+  void* pr_graph_exec = nullptr;
+  double pr_graph_damping = 0.0;
+  ~DeviceGraph();
 };
 
 struct GpuRunResult {
