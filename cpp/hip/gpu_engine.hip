@@ -2085,9 +2085,18 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
       (!g.directed ? g.oe_dst.data() : g.ie_dst.data());
   if (pull) ensure_buckets(g, s);
 
-  DeviceBuffer<double> rank_arr(nv_pad), acc(nv_pad);
-  DeviceBuffer<float> contrib(nv_pad);
-  DeviceBuffer<double> d_dangling(1);
+  // working set lives on the DeviceGraph: the cached hipGraph bakes these
+  // pointers (locals would dangle across calls — GPU fault under realloc)
+  if (g.pr_rank.size() < nv_pad) {
+    g.pr_rank.resize(nv_pad);
+    g.pr_acc.resize(nv_pad);
+    g.pr_contrib.resize(nv_pad);
+    g.pr_dangling.resize(1);
+  }
+  DeviceBuffer<double>& rank_arr = g.pr_rank;
+  DeviceBuffer<double>& acc = g.pr_acc;
+  DeviceBuffer<float>& contrib = g.pr_contrib;
+  DeviceBuffer<double>& d_dangling = g.pr_dangling;
   DeviceBuffer<double> d_l1(tol > 0 ? 1 : 0);
   DevGraphView view = make_view(g, rank_, world_);
 

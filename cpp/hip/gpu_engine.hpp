@@ -41,9 +41,13 @@ struct DeviceGraph {
   uint64_t n_small = 0, n_mid = 0, n_large = 0;
   bool buckets_built = false;
   uint32_t owned() const { return v_end - v_begin; }
-  // cached hipGraph of one PageRank iteration (single-GPU fixed-iter path)
+  // cached hipGraph of one PageRank iteration (single-GPU fixed-iter
+  // path) + the working set it references: the capture bakes device
+  // pointers, so these buffers must live as long as the graph
   void* pr_graph_exec = nullptr;
   double pr_graph_damping = 0.0;
+  DeviceBuffer<double> pr_rank, pr_acc, pr_dangling;
+  DeviceBuffer<float> pr_contrib;
   ~DeviceGraph();
 };
 
