@@ -150,7 +150,10 @@ class Fragment {
         if (!V.get_gid(e.src, &sg) || !V.get_gid(e.dst, &dg)) continue;
         out_bins[P.fid(sg)].push_back({sg, dg, e.w});
         if (!directed) {
-          if (sg != dg) out_bins[P.fid(dg)].push_back({dg, sg, e.w});
+          // reference parity: undirected storage holds BOTH orientations,
+          // including self loops twice (csr_edgecut_fragment_base.h
+          // parse_iter_out_undirected increments both endpoints)
+          out_bins[P.fid(dg)].push_back({dg, sg, e.w});
         } else if (want_in) {
           in_bins[P.fid(dg)].push_back({dg, sg, e.w});
         }
@@ -296,12 +299,16 @@ class Fragment {
     for (vid_t v = 0; v < ivnum_; ++v) {
       oid_t so = lid2oid(v);
       auto adj = out_edges(v);
+      bool drop_self = false;  // self loops stored twice: emit every other
       for (size_t i = 0; i < adj.n; ++i) {
         oid_t dg = lid2oid(adj.dst[i]);
         if (!directed_) {
-          // keep one orientation per input edge; ties (self loops) kept
           vid_t sg = lid2gid(v), dgid = lid2gid(adj.dst[i]);
-          if (sg > dgid) continue;
+          if (sg > dgid) continue;  // one orientation per input edge
+          if (sg == dgid) {
+            drop_self = !drop_self;
+            if (drop_self) continue;
+          }
         }
         out.push_back({so, dg, adj.w ? adj.w[i] : 1.0f});
       }

@@ -296,7 +296,8 @@ __global__ void gen_edges_kernel(uint64_t ne, uint64_t seed, int scale,
         if (weighted) out_w[pos] = w;
       }
     }
-    if (undirected && d >= my_begin && d < my_end && d != s) {
+    if (undirected && d >= my_begin && d < my_end) {
+      // self loops stored twice, like the reference's undirected CSR
       uint64_t pos = atomicAdd(out_cnt, 1ull);
       if (out_src) {
         out_src[pos] = d;

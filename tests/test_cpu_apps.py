@@ -217,3 +217,17 @@ def test_pagerank_convergence_mode(eng):
     o1, o2 = np.argsort(r_conv["oids"]), np.argsort(r_long["oids"])
     assert np.allclose(r_conv["values"][o1], r_long["values"][o2],
                        rtol=1e-9)
+
+
+def test_undirected_self_loop_semantics(eng):
+    # reference parity: undirected storage holds self loops twice (both
+    # orientations), which flips CDLP's mode tie here
+    src = np.array([0, 0, 0], dtype=np.int64)
+    dst = np.array([0, 1, 1], dtype=np.int64)
+    g = eng.load_edges(src, dst, directed=False, num_vertices=2)
+    r = eng.cdlp(g, 1)
+    got = r["values"][np.argsort(r["oids"])]
+    assert np.array_equal(got, cdlp_oracle(2, src, dst, 1, directed=False))
+    rp = eng.pagerank(g, 0.85, 6)
+    exp = pagerank_oracle(2, src, dst, 0.85, 6, directed=False)
+    assert np.allclose(rp["values"][np.argsort(rp["oids"])], exp, rtol=1e-9)
