@@ -3670,26 +3670,62 @@ __global__ void lcc_triangle_heavy_kernel(
 
 // --- directed LCC (reference cuda/lcc/lcc_directed*.h semantics) ---------
 // numerator(v) = sum over u in N(v) of |N(v) ∩ Nout(u)| computed directly
-// (no orientation trick applies to the directed definition): enumerate the
-// smaller of U(v) / Nout(u) and probe the other family's hash set.
-__global__ void lcc_dir_count_kernel(
+// (no orientation trick applies to the directed definition). Same light
+// tiers as the undirected counter: wee rows wave-per-row with global
+// probes; bigger rows block-per-row with U(v)'s hash set staged in LDS.
+__global__ void lcc_dir_bucket_kernel(const uint64_t* __restrict__ goffU,
+                                      uint32_t owned, uint32_t v_begin,
+                                      uint32_t* wee, unsigned long long* cw,
+                                      uint32_t* big,
+                                      unsigned long long* cb) {
+  __shared__ uint32_t s_cnt[2];
+  __shared__ unsigned long long s_base[2];
+  uint32_t* lists[2] = {wee, big};
+  unsigned long long* gcnt[2] = {cw, cb};
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t base = blockIdx.x * blockDim.x; base < owned;
+       base += stride) {
+    if (threadIdx.x < 2) s_cnt[threadIdx.x] = 0;
+    __syncthreads();
+    uint32_t r = base + threadIdx.x;
+    int b = -1;
+    uint32_t loc = 0;
+    if (r < owned) {
+      uint64_t n = goffU[v_begin + r + 1] - goffU[v_begin + r];
+      if (n >= 2) {
+        b = n <= kLccWeeRows ? 0 : 1;
+        loc = atomicAdd(&s_cnt[b], 1u);
+      }
+    }
+    __syncthreads();
+    if (threadIdx.x < 2 && s_cnt[threadIdx.x])
+      s_base[threadIdx.x] =
+          atomicAdd(gcnt[threadIdx.x],
+                    static_cast<unsigned long long>(s_cnt[threadIdx.x]));
+    __syncthreads();
+    if (b >= 0) lists[b][s_base[b] + loc] = r;
+    __syncthreads();
+  }
+}
+
+__global__ void lcc_dir_count_wee_kernel(
     const uint64_t* __restrict__ goffU, const uint32_t* __restrict__ gdstU,
     const uint64_t* __restrict__ hoffU, const uint32_t* __restrict__ htabU,
     const uint64_t* __restrict__ goffO, const uint32_t* __restrict__ gdstO,
     const uint64_t* __restrict__ hoffO, const uint32_t* __restrict__ htabO,
-    uint32_t owned, uint32_t v_begin, unsigned long long* __restrict__ T,
-    uint32_t heavy_thresh, unsigned long long* __restrict__ heavy_q,
+    const uint32_t* __restrict__ rows, uint64_t nrows, uint32_t v_begin,
+    unsigned long long* __restrict__ T, uint32_t heavy_thresh,
+    unsigned long long* __restrict__ heavy_q,
     unsigned long long* __restrict__ heavy_n) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int wpb = kBlock / kWave;
   size_t wstride = static_cast<size_t>(gridDim.x) * wpb;
-  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid; i < owned;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid; i < nrows;
        i += wstride) {
-    uint32_t v = v_begin + i;
+    uint32_t v = v_begin + rows[i];
     uint64_t ub = goffU[v];
     uint32_t un = static_cast<uint32_t>(goffU[v + 1] - ub);
-    if (un < 2) continue;
     unsigned long long my_v = 0;
     for (uint32_t k = lane; k < un; k += kWave) {
       uint32_t u = gdstU[ub + k];
@@ -3723,6 +3759,68 @@ __global__ void lcc_dir_count_kernel(
   }
 }
 
+__global__ void lcc_dir_count_staged_kernel(
+    const uint64_t* __restrict__ goffU, const uint32_t* __restrict__ gdstU,
+    const uint64_t* __restrict__ hoffU, const uint32_t* __restrict__ htabU,
+    const uint64_t* __restrict__ goffO, const uint32_t* __restrict__ gdstO,
+    const uint64_t* __restrict__ hoffO, const uint32_t* __restrict__ htabO,
+    const uint32_t* __restrict__ rows, uint64_t nrows, uint32_t v_begin,
+    unsigned long long* __restrict__ T, uint32_t heavy_thresh,
+    unsigned long long* __restrict__ heavy_q,
+    unsigned long long* __restrict__ heavy_n) {
+  __shared__ uint32_t s_tab[kLccStageSlots];
+  __shared__ unsigned long long s_hits;
+  for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
+    uint32_t v = v_begin + rows[i];
+    uint64_t ub = goffU[v];
+    uint32_t un = static_cast<uint32_t>(goffU[v + 1] - ub);
+    uint64_t hb = hoffU[v];
+    uint32_t ucap = static_cast<uint32_t>(hoffU[v + 1] - hb);
+    const bool stage = ucap <= kLccStageSlots;
+    if (stage)
+      for (uint32_t k = threadIdx.x; k < ucap; k += blockDim.x)
+        s_tab[k] = htabU[hb + k];
+    if (threadIdx.x == 0) s_hits = 0;
+    __syncthreads();
+    const uint32_t umask = ucap - 1;
+    const uint32_t* ut = htabU + hb;
+    unsigned long long my_v = 0;
+    for (uint32_t k = threadIdx.x; k < un; k += blockDim.x) {
+      uint32_t u = gdstU[ub + k];
+      uint64_t ob = goffO[u];
+      uint32_t on = static_cast<uint32_t>(goffO[u + 1] - ob);
+      if (on == 0) continue;
+      uint32_t mn = on < un ? on : un;
+      if (mn > heavy_thresh) {
+        heavy_q[atomicAdd(heavy_n, 1ull)] =
+            (static_cast<unsigned long long>(v) << 32) | u;
+        continue;
+      }
+      unsigned long long hits = 0;
+      if (on <= un) {
+        for (uint32_t e = 0; e < on; ++e) {
+          uint32_t w = gdstO[ob + e];
+          bool hit = stage ? lcc_probe_lds(s_tab, umask, w)
+                           : lcc_probe(ut, umask, w);
+          if (hit) ++hits;
+        }
+      } else {
+        uint64_t ohb = hoffO[u];
+        const uint64_t omask = (hoffO[u + 1] - ohb) - 1;
+        const uint32_t* ot = htabO + ohb;
+        for (uint32_t e = 0; e < un; ++e)
+          if (lcc_probe(ot, omask, gdstU[ub + e])) ++hits;
+      }
+      my_v += hits;
+    }
+    if (my_v) atomicAdd(&s_hits, my_v);
+    __syncthreads();
+    if (threadIdx.x == 0 && s_hits) atomicAdd(&T[v], s_hits);
+    __syncthreads();
+  }
+}
+
+// heavy (wave-per-edge, lanes split the enumeration)
 __global__ void lcc_dir_heavy_kernel(
     const uint64_t* __restrict__ goffU, const uint32_t* __restrict__ gdstU,
     const uint64_t* __restrict__ hoffU, const uint32_t* __restrict__ htabU,
@@ -4029,13 +4127,28 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
     HIP_CHECK(hipStreamSynchronize(s));
     heavy_q.resize(std::max<uint64_t>(hi - lo, 1));
   }
-  if (owned)
-    lcc_dir_count_kernel<<<grid_for(static_cast<size_t>(owned) * kWave),
-                           kBlock, 0, s>>>(
-        U.goff.data(), U.gdst.data(), hoffU.data(), htabU.data(),
-        O.goff.data(), O.gdst.data(), hoffO.data(), htabO.data(), owned,
-        g.v_begin, Tcnt.data(), kHeavyThresh, heavy_q.data(),
-        heavy_n.data());
+  if (owned) {
+    DeviceBuffer<uint32_t> wee_rows(owned), big_rows(owned);
+    DeviceBuffer<unsigned long long> tcnts(2);
+    tcnts.zero(s);
+    lcc_dir_bucket_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+        U.goff.data(), owned, g.v_begin, wee_rows.data(), tcnts.data() + 0,
+        big_rows.data(), tcnts.data() + 1);
+    auto tc = tcnts.download(s);
+    if (tc[0])
+      lcc_dir_count_wee_kernel<<<grid_for(tc[0] * kWave), kBlock, 0, s>>>(
+          U.goff.data(), U.gdst.data(), hoffU.data(), htabU.data(),
+          O.goff.data(), O.gdst.data(), hoffO.data(), htabO.data(),
+          wee_rows.data(), tc[0], g.v_begin, Tcnt.data(), kHeavyThresh,
+          heavy_q.data(), heavy_n.data());
+    if (tc[1])
+      lcc_dir_count_staged_kernel<<<std::min<uint64_t>(tc[1], kMaxGrid),
+                                    kBlock, 0, s>>>(
+          U.goff.data(), U.gdst.data(), hoffU.data(), htabU.data(),
+          O.goff.data(), O.gdst.data(), hoffO.data(), htabO.data(),
+          big_rows.data(), tc[1], g.v_begin, Tcnt.data(), kHeavyThresh,
+          heavy_q.data(), heavy_n.data());
+  }
   {
     unsigned long long hn = 0;
     HIP_CHECK(hipMemcpyAsync(&hn, heavy_n.data(), 8, hipMemcpyDeviceToHost,
