@@ -3616,89 +3616,6 @@ __global__ void lcc_tribucket_kernel(const uint64_t* __restrict__ goff,
   }
 }
 
-// split heavy edges by enumeration length: short ones stay wave-per-edge;
-// long ones get a block each with the big-side table staged in LDS (the
-// all-block variant lost on short edges, the all-wave one stalls on long)
-__global__ void lcc_heavy_split_kernel(
-    const uint64_t* __restrict__ goff,
-    const unsigned long long* __restrict__ heavy_q, uint64_t heavy_n,
-    uint32_t en_thresh, unsigned long long* __restrict__ q_long,
-    unsigned long long* __restrict__ n_long,
-    unsigned long long* __restrict__ q_short,
-    unsigned long long* __restrict__ n_short) {
-  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
-  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * blockDim.x +
-                    threadIdx.x;
-       i < heavy_n; i += stride) {
-    unsigned long long pk = heavy_q[i];
-    uint32_t u = static_cast<uint32_t>(pk >> 32);
-    uint32_t v = static_cast<uint32_t>(pk);
-    uint32_t un = static_cast<uint32_t>(goff[u + 1] - goff[u]);
-    uint32_t vn = static_cast<uint32_t>(goff[v + 1] - goff[v]);
-    uint32_t mn = vn < un ? vn : un;
-    if (mn > en_thresh)
-      q_long[atomicAdd(n_long, 1ull)] = pk;
-    else
-      q_short[atomicAdd(n_short, 1ull)] = pk;
-  }
-}
-
-__global__ void lcc_triangle_heavy_block_kernel(
-    const uint64_t* __restrict__ goff, const uint32_t* __restrict__ gdst,
-    const uint64_t* __restrict__ hoff, const uint32_t* __restrict__ htab,
-    const unsigned long long* __restrict__ heavy_q, uint64_t heavy_n,
-    unsigned long long* __restrict__ T) {
-  __shared__ uint32_t s_tab[kLccStageSlots];
-  __shared__ unsigned long long s_hits;
-  for (uint64_t i = blockIdx.x; i < heavy_n; i += gridDim.x) {
-    unsigned long long pk = heavy_q[i];
-    uint32_t u = static_cast<uint32_t>(pk >> 32);
-    uint32_t v = static_cast<uint32_t>(pk);
-    uint64_t ub = goff[u];
-    uint32_t un = static_cast<uint32_t>(goff[u + 1] - ub);
-    uint64_t vb = goff[v];
-    uint32_t vn = static_cast<uint32_t>(goff[v + 1] - vb);
-    uint64_t eb, hb;
-    uint32_t en;
-    if (vn < un) {
-      eb = vb;
-      en = vn;
-      hb = hoff[u];
-    } else {
-      eb = ub;
-      en = un;
-      hb = hoff[v];
-    }
-    const uint32_t cap =
-        static_cast<uint32_t>((vn < un ? hoff[u + 1] : hoff[v + 1]) - hb);
-    const uint32_t mask = cap - 1;
-    const bool stage = cap <= kLccStageSlots && en >= (cap >> 3);
-    if (stage)
-      for (uint32_t k = threadIdx.x; k < cap; k += blockDim.x)
-        s_tab[k] = htab[hb + k];
-    if (threadIdx.x == 0) s_hits = 0;
-    __syncthreads();
-    const uint32_t* gt = htab + hb;
-    unsigned long long hits = 0;
-    for (uint32_t e = threadIdx.x; e < en; e += blockDim.x) {
-      uint32_t w = gdst[eb + e];
-      bool hit = stage ? lcc_probe_lds(s_tab, mask, w)
-                       : lcc_probe(gt, mask, w);
-      if (hit) {
-        ++hits;
-        atomicAdd(&T[w], 1ull);
-      }
-    }
-    if (hits) atomicAdd(&s_hits, hits);
-    __syncthreads();
-    if (threadIdx.x == 0 && s_hits) {
-      atomicAdd(&T[u], s_hits);
-      atomicAdd(&T[v], s_hits);
-    }
-    __syncthreads();
-  }
-}
-
 // second pass: one wave per heavy edge; lanes stride the smaller list
 __global__ void lcc_triangle_heavy_kernel(
     const uint64_t* __restrict__ goff, const uint32_t* __restrict__ gdst,
@@ -4512,28 +4429,10 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
       fprintf(stderr, "[lcc] oriented_total=%llu heavy_edges=%llu\n",
               static_cast<unsigned long long>(oriented_total),
               static_cast<unsigned long long>(hn));
-    if (hn) {
-      constexpr uint32_t kEnLong = 1024;
-      DeviceBuffer<unsigned long long> q_long(hn), q_short(hn);
-      DeviceBuffer<unsigned long long> ncnt(2);
-      ncnt.zero(s);
-      lcc_heavy_split_kernel<<<grid_for(hn), kBlock, 0, s>>>(
-          goff.data(), heavy_q.data(), hn, kEnLong, q_long.data(),
-          ncnt.data() + 0, q_short.data(), ncnt.data() + 1);
-      auto hc = ncnt.download(s);
-      if (hc[1])
-        lcc_triangle_heavy_kernel<<<grid_for(hc[1] * kWave), kBlock, 0,
-                                    s>>>(goff.data(), gdst.data(),
-                                         hoff.data(), htab.data(),
-                                         q_short.data(), hc[1],
-                                         Tcnt.data());
-      if (hc[0])
-        lcc_triangle_heavy_block_kernel<<<std::min<uint64_t>(hc[0],
-                                                             kMaxGrid),
-                                          kBlock, 0, s>>>(
-            goff.data(), gdst.data(), hoff.data(), htab.data(),
-            q_long.data(), hc[0], Tcnt.data());
-    }
+    if (hn)
+      lcc_triangle_heavy_kernel<<<grid_for(hn * kWave), kBlock, 0, s>>>(
+          goff.data(), gdst.data(), hoff.data(), htab.data(),
+          heavy_q.data(), hn, Tcnt.data());
   }
   if (multi)
     NCCL_CHECK(ncclAllReduce(Tcnt.data(), Tcnt.data(), nv_pad, ncclUint64,
