@@ -47,6 +47,10 @@ def build_parser():
     ap.add_argument("--cdlp_mr", type=int, default=10)
     ap.add_argument("--kcore_k", type=int, default=3)
     ap.add_argument("--kclique_k", type=int, default=3)
+    # delta mutation (LoadGraphAndMutate, loader.h:55-68 +
+    # ev_fragment_mutator.h: base graph + additions/removals files)
+    ap.add_argument("--efile_add", default="")
+    ap.add_argument("--efile_remove", default="")
     # checkpoint (ev_fragment_loader.h:75-93)
     ap.add_argument("--serialize", action="store_true")
     ap.add_argument("--deserialize", action="store_true")
@@ -90,6 +94,20 @@ def main(argv=None):
         if args.serialize:
             os.makedirs(args.serialization_prefix, exist_ok=True)
             eng.save_graph(g, args.serialization_prefix)
+    if args.efile_add or args.efile_remove:
+        empty = np.array([], dtype=np.int64)
+        a_s = a_d = r_s = r_d = empty
+        a_w = None
+        if args.efile_add:
+            a_s, a_d, a_w = read_ldbc_edges(args.efile_add,
+                                            weighted=weighted,
+                                            rank=rank, world=world)
+        if args.efile_remove:
+            r_s, r_d, _ = read_ldbc_edges(args.efile_remove, rank=rank,
+                                          world=world)
+        g = eng.mutate_graph(g, add_src=a_s, add_dst=a_d, add_weights=a_w,
+                             remove_src=r_s, remove_dst=r_d,
+                             remove_vertices=empty)
     timers.append(("load graph", time.time() - t))
 
     t = time.time()
