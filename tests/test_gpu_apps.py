@@ -249,3 +249,19 @@ def test_gpu_arbitrary_oids(eng):
     exp_l = lcc_oracle(nv, si, di, directed=False)
     for j in range(0, nv, 11):
         assert abs(gotl[int(oids[j])] - exp_l[j]) < 1e-9, j
+
+
+def test_gpu_serialized_roundtrip(eng, tmp_path):
+    # checkpoint -> reload -> upload on the GPU engine
+    src, dst, w = random_graph(num_v=3000, num_e=25000, seed=83,
+                               weighted=True)
+    g = eng.load_edges(src, dst, weights=w, directed=False,
+                       num_vertices=3000)
+    before = eng.sssp(g, 5)
+    eng.save_graph(g, str(tmp_path))
+    g2 = eng.load_serialized(str(tmp_path))
+    after = eng.sssp(g2, 5)
+    o1, o2 = np.argsort(before["oids"]), np.argsort(after["oids"])
+    assert np.array_equal(before["oids"][o1], after["oids"][o2])
+    assert np.allclose(before["values"][o1], after["values"][o2],
+                       rtol=1e-6)

@@ -58,7 +58,7 @@ BASE = dict(num_v=300, num_e=1800, seed=101, weighted=False, directed=True,
             in_csr=False, source=3, vertex_oids=False)
 
 
-@pytest.mark.parametrize("world", [2, 3])
+@pytest.mark.parametrize("world", [2, 3, 4])
 def test_bfs_mp(world, free_port, tmp_path):
     cfg = dict(BASE, app="bfs")
     oids, vals = run_world(world, cfg, free_port, tmp_path)
@@ -66,7 +66,7 @@ def test_bfs_mp(world, free_port, tmp_path):
     assert np.array_equal(vals, bfs_oracle(cfg["num_v"], src, dst, 3))
 
 
-@pytest.mark.parametrize("world", [2])
+@pytest.mark.parametrize("world", [2, 4])
 def test_sssp_mp(world, free_port, tmp_path):
     cfg = dict(BASE, app="sssp", weighted=True)
     oids, vals = run_world(world, cfg, free_port, tmp_path)
@@ -75,7 +75,7 @@ def test_sssp_mp(world, free_port, tmp_path):
                        rtol=1e-9)
 
 
-@pytest.mark.parametrize("world", [2, 3])
+@pytest.mark.parametrize("world", [2, 4])
 def test_pagerank_mp(world, free_port, tmp_path):
     cfg = dict(BASE, app="pagerank")
     oids, vals = run_world(world, cfg, free_port, tmp_path)
