@@ -108,7 +108,8 @@ using arr_f32 = py::array_t<float, py::array::c_style | py::array::forcecast>;
 std::shared_ptr<PyGraph> load_edges(
     PyEngine& eng, arr_i64 src, arr_i64 dst, std::optional<arr_f32> weights,
     bool directed, int64_t num_vertices, std::optional<arr_i64> vertex_oids,
-    bool build_in_csr, const std::string& partitioner) {
+    bool build_in_csr, const std::string& partitioner,
+    const std::string& idxer) {
   size_t n = src.size();
   if (static_cast<size_t>(dst.size()) != n)
     throw std::runtime_error("src/dst size mismatch");
@@ -129,6 +130,9 @@ std::shared_ptr<PyGraph> load_edges(
     const int64_t* po = vertex_oids->data();
     size_t nv = vertex_oids->size();
     std::vector<oid_t> owned;
+    if (idxer == "mph" && partitioner == "map")
+      throw std::runtime_error(
+          "idxer=mph requires hash ownership (partitioner=hash)");
     if (partitioner == "map") {
       // MapPartitioner semantics (reference partitioner.h:103): ownership
       // = the rank that supplied the oid (e.g. a rebalanced range)
@@ -154,8 +158,11 @@ std::shared_ptr<PyGraph> load_edges(
       std::sort(owned.begin(), owned.end());
       owned.erase(std::unique(owned.begin(), owned.end()), owned.end());
     }
-    g->vm->init_hashmap(eng.world, PartitionerKind::kHash, eng.c(),
-                        std::move(owned));
+    if (idxer == "mph")
+      g->vm->init_mph(eng.world, eng.c(), std::move(owned));
+    else
+      g->vm->init_hashmap(eng.world, PartitionerKind::kHash, eng.c(),
+                          std::move(owned));
   } else {
     if (num_vertices <= 0)
       throw std::runtime_error("num_vertices required for identity mapping");
@@ -392,7 +399,8 @@ PYBIND11_MODULE(_core, m) {
            py::arg("weights") = std::nullopt, py::arg("directed") = false,
            py::arg("num_vertices") = -1, py::arg("vertex_oids") = std::nullopt,
            py::arg("build_in_csr") = false,
-           py::arg("partitioner") = "segmented")
+           py::arg("partitioner") = "segmented",
+           py::arg("idxer") = "hashmap")
       .def("mutate_graph",
            [](PyEngine& eng, PyGraph& g, arr_i64 add_src, arr_i64 add_dst,
               std::optional<arr_f32> add_w, arr_i64 rm_src, arr_i64 rm_dst,

@@ -91,7 +91,7 @@ inline void serialize_graph(const Fragment& frag,
   w.pod(static_cast<uint8_t>(vm.partitioner()));
   w.pod(vm.total_vertices());
   w.vec(vm.segments());
-  if (vm.idxer() == IdxerKind::kHashmap) {
+  if (vm.idxer() == IdxerKind::kHashmap || vm.idxer() == IdxerKind::kMph) {
     for (int f = 0; f < vm.fnum(); ++f) w.vec(vm.frag_oids(f));
   }
   // fragment
@@ -134,7 +134,10 @@ deserialize_graph(const std::string& prefix, fid_t fid) {
   } else {
     std::vector<std::vector<oid_t>> oids(fnum);
     for (uint32_t f = 0; f < fnum; ++f) r.vec(oids[f]);
-    vm->init_hashmap_local(fnum, pk, std::move(oids));
+    if (idx == IdxerKind::kMph)
+      vm->init_mph_local(fnum, std::move(oids));
+    else
+      vm->init_hashmap_local(fnum, pk, std::move(oids));
   }
   auto frag = Fragment::FromParts(vm, r, fnum);
   return {vm, std::move(frag)};

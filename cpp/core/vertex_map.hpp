@@ -19,6 +19,7 @@
 
 #include "net.hpp"
 #include "parallel.hpp"
+#include "mph.hpp"
 #include "types.hpp"
 
 namespace grapehip {
@@ -48,7 +49,7 @@ class IdParser {
   vid_t lid_mask_ = ~0u;
 };
 
-enum class IdxerKind : uint8_t { kIdentity = 0, kHashmap = 1 };
+enum class IdxerKind : uint8_t { kIdentity = 0, kHashmap = 1, kMph = 2 };
 
 class VertexMap {
  public:
@@ -122,6 +123,13 @@ class VertexMap {
       *gid = parser_.gid(f, static_cast<vid_t>(oid - seg_[f]));
       return true;
     }
+    if (idxer_ == IdxerKind::kMph) {
+      fid_t f = static_cast<fid_t>(hash_oid(oid) % fnum_);
+      uint64_t lid = mph_[f].lookup(oid);
+      if (lid >= l2o_[f].size() || l2o_[f][lid] != oid) return false;
+      *gid = parser_.gid(f, static_cast<vid_t>(lid));
+      return true;
+    }
     auto it = o2g_.find(oid);
     if (it == o2g_.end()) return false;
     *gid = it->second;
@@ -133,13 +141,61 @@ class VertexMap {
     vid_t l = parser_.lid(gid);
     if (idxer_ == IdxerKind::kIdentity)
       return static_cast<oid_t>(seg_[f] + l);
-    return l2o_[f][l];
+    return l2o_[f][l];  // hashmap and mph both keep lid->oid dense
   }
 
   vid_t frag_vnum(fid_t f) const {
     if (idxer_ == IdxerKind::kIdentity)
       return static_cast<vid_t>(seg_[f + 1] - seg_[f]);
     return static_cast<vid_t>(l2o_[f].size());
+  }
+
+  // MPH mode (reference PTHashIdxer parity): hash-partitioned ownership +
+  // per-fragment minimal perfect hash for oid->lid, ~5 bits/key of index
+  // structure instead of an open-addressing map. Deterministic build, so
+  // every rank derives identical lids from the replicated oid lists.
+  void init_mph(int fnum, TcpComm* comm, std::vector<oid_t> owned_oids) {
+    fnum_ = fnum;
+    parser_.init(fnum);
+    idxer_ = IdxerKind::kMph;
+    pkind_ = PartitionerKind::kHash;
+    l2o_.assign(fnum, {});
+    // replicate per-fragment oid lists (same exchange as hashmap mode)
+    std::vector<std::string> send(fnum);
+    std::string mine(reinterpret_cast<const char*>(owned_oids.data()),
+                     owned_oids.size() * sizeof(oid_t));
+    for (int f = 0; f < fnum; ++f) send[f] = mine;
+    std::vector<std::string> recv =
+        comm ? comm->exchange_all(send) : std::vector<std::string>{mine};
+    total_vertices_ = 0;
+    mph_.assign(fnum, {});
+    for (int f = 0; f < fnum; ++f) {
+      const std::string& blob = recv[f];
+      size_t n = blob.size() / sizeof(oid_t);
+      std::vector<oid_t> keys(n);
+      std::memcpy(keys.data(), blob.data(), blob.size());
+      mph_[f].build(keys);
+      // place keys at their MPH index: lid = mph(oid)
+      l2o_[f].assign(n, 0);
+      for (oid_t k : keys) l2o_[f][mph_[f].lookup(k)] = k;
+      total_vertices_ += n;
+    }
+  }
+  void init_mph_local(int fnum, std::vector<std::vector<oid_t>> l2o) {
+    // deserialization path: rebuild the (deterministic) MPH per fragment
+    fnum_ = fnum;
+    parser_.init(fnum);
+    idxer_ = IdxerKind::kMph;
+    pkind_ = PartitionerKind::kHash;
+    l2o_.assign(fnum, {});
+    mph_.assign(fnum, {});
+    total_vertices_ = 0;
+    for (int f = 0; f < fnum; ++f) {
+      mph_[f].build(l2o[f]);
+      l2o_[f].assign(l2o[f].size(), 0);
+      for (oid_t k : l2o[f]) l2o_[f][mph_[f].lookup(k)] = k;
+      total_vertices_ += l2o[f].size();
+    }
   }
 
   // Rebuild hashmap mode from already-replicated per-fragment oid lists
@@ -176,8 +232,9 @@ class VertexMap {
   PartitionerKind pkind_ = PartitionerKind::kSegmented;
   uint64_t total_vertices_ = 0;
   std::vector<uint64_t> seg_;                   // identity/segmented
-  std::vector<std::vector<oid_t>> l2o_;         // hashmap: per-fid lid->oid
+  std::vector<std::vector<oid_t>> l2o_;         // hashmap/mph: lid->oid
   std::unordered_map<oid_t, vid_t> o2g_;        // hashmap: oid->gid
+  std::vector<MinimalPerfectHash> mph_;         // mph: per-fid oid->lid
 };
 
 }  // namespace grapehip

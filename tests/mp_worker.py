@@ -41,6 +41,8 @@ def main():
         # split vertex list round-robin too
         all_oids = np.arange(cfg["num_v"], dtype=np.int64) * 3 + 1
         kw["vertex_oids"] = all_oids[sl]
+        if cfg.get("idxer"):
+            kw["idxer"] = cfg["idxer"]
         src = all_oids[src]
         dst = all_oids[dst]
     else:

@@ -205,3 +205,13 @@ def test_sssp_auto_mp(world, free_port, tmp_path):
     expect = sssp_oracle(cfg["num_v"], src, dst, w, 3)
     finite = expect < 1e300
     assert np.allclose(vals[finite], expect[finite], rtol=1e-9)
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_bfs_mp_mph_idxer(world, free_port, tmp_path):
+    cfg = dict(BASE, app="bfs", vertex_oids=True, idxer="mph",
+               source=3 * 3 + 1)
+    oids, vals = run_world(world, cfg, free_port, tmp_path)
+    src, dst, _ = graph_arrays(cfg)
+    assert np.array_equal(oids, np.arange(cfg["num_v"]) * 3 + 1)
+    assert np.array_equal(vals, bfs_oracle(cfg["num_v"], src, dst, 3))

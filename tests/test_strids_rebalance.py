@@ -47,3 +47,49 @@ def test_rebalance_partition_single(eng):
     assert np.array_equal(oids, np.arange(nv))  # world=1: everything
     g = eng.load_edges(src, dst, directed=True, vertex_oids=oids)
     assert g.num_vertices == nv
+
+
+def test_mph_idxer(eng):
+    # minimal-perfect-hash idxer (pthash parity): same results as the
+    # hashmap idxer on sparse oids, including checkpoint round-trip
+    rng = np.random.default_rng(223)
+    nv = 4000
+    oids = np.sort(rng.choice(10**12, size=nv, replace=False)).astype(
+        np.int64)
+    si = rng.integers(0, nv, 30000)
+    di = rng.integers(0, nv, 30000)
+    keep = si != di
+    si, di = si[keep], di[keep]
+    g_h = eng.load_edges(oids[si], oids[di], directed=False,
+                         vertex_oids=oids, idxer="hashmap")
+    g_m = eng.load_edges(oids[si], oids[di], directed=False,
+                         vertex_oids=oids, idxer="mph")
+    src_oid = int(oids[17])
+    rh = eng.bfs(g_h, src_oid)
+    rm = eng.bfs(g_m, src_oid)
+    oh, om = np.argsort(rh["oids"]), np.argsort(rm["oids"])
+    assert np.array_equal(rh["oids"][oh], rm["oids"][om])
+    assert np.array_equal(rh["values"][oh], rm["values"][om])
+    rh = eng.pagerank(g_h, 0.85, 5)
+    rm = eng.pagerank(g_m, 0.85, 5)
+    oh, om = np.argsort(rh["oids"]), np.argsort(rm["oids"])
+    assert np.allclose(rh["values"][oh], rm["values"][om], rtol=1e-12)
+
+
+def test_mph_idxer_roundtrip(eng, tmp_path):
+    rng = np.random.default_rng(227)
+    nv = 800
+    oids = np.sort(rng.choice(10**9, size=nv, replace=False)).astype(
+        np.int64)
+    si = rng.integers(0, nv, 5000)
+    di = rng.integers(0, nv, 5000)
+    keep = si != di
+    g = eng.load_edges(oids[si[keep]], oids[di[keep]], directed=False,
+                       vertex_oids=oids, idxer="mph")
+    before = eng.wcc(g)
+    eng.save_graph(g, str(tmp_path))
+    g2 = eng.load_serialized(str(tmp_path))
+    after = eng.wcc(g2)
+    o1, o2 = np.argsort(before["oids"]), np.argsort(after["oids"])
+    assert np.array_equal(before["oids"][o1], after["oids"][o2])
+    assert np.array_equal(before["values"][o1], after["values"][o2])
