@@ -23,11 +23,55 @@ namespace grapehip {
 
 enum class SampleStrategy : int { kRandom = 0, kEdgeWeight = 1, kTopK = 2 };
 
+// Per-vertex alias tables for O(1) weighted draws (reference
+// fragment_indices.h builds the same structure per vertex). Flat over the
+// fragment's edge-id space; built once per query in parallel.
+struct AliasTables {
+  std::vector<float> prob;     // [local eid]
+  std::vector<uint32_t> alias; // [local eid] (intra-row index)
+
+  void build(const Fragment& frag) {
+    size_t ne = frag.local_edges();
+    prob.assign(ne, 1.0f);
+    alias.assign(ne, 0);
+    parallel_for(0, frag.ivnum(), [&](size_t vs) {
+      vid_t v = static_cast<vid_t>(vs);
+      auto adj = frag.out_edges(v);
+      if (adj.n == 0) return;
+      size_t base = adj.dst - frag.oe_dsts().data();
+      // Vose's method over the row's weights
+      double total = 0;
+      for (size_t i = 0; i < adj.n; ++i)
+        total += adj.w ? adj.w[i] : 1.0;
+      std::vector<double> scaled(adj.n);
+      std::vector<uint32_t> small, large;
+      for (size_t i = 0; i < adj.n; ++i) {
+        scaled[i] = (adj.w ? adj.w[i] : 1.0) * adj.n / total;
+        (scaled[i] < 1.0 ? small : large).push_back(
+            static_cast<uint32_t>(i));
+      }
+      while (!small.empty() && !large.empty()) {
+        uint32_t s = small.back();
+        small.pop_back();
+        uint32_t l = large.back();
+        large.pop_back();
+        prob[base + s] = static_cast<float>(scaled[s]);
+        alias[base + s] = l;
+        scaled[l] = scaled[l] + scaled[s] - 1.0;
+        (scaled[l] < 1.0 ? small : large).push_back(l);
+      }
+      for (uint32_t i : large) prob[base + i] = 1.0f;
+      for (uint32_t i : small) prob[base + i] = 1.0f;
+    }, 256);
+  }
+};
+
 struct SamplerContext {
   SampleStrategy strategy = SampleStrategy::kRandom;
   int hops = 2;
   int top_k = 4;
   uint64_t seed = 7;
+  AliasTables alias;  // built when strategy == kEdgeWeight
   // walks this fragment OWNS (origin here); path[w][h], gid space,
   // kInvalidVid where the walk died (dead end)
   std::vector<uint64_t> walk_ids;            // global walk index
@@ -73,6 +117,8 @@ class SamplerApp {
   };
 
   void PEval(const Fragment& frag, SamplerContext& ctx, MessageManager& mm) {
+    if (ctx.strategy == SampleStrategy::kEdgeWeight)
+      ctx.alias.build(frag);
     advance(frag, ctx, mm);
   }
 
@@ -120,21 +166,14 @@ class SamplerApp {
         pick = rng() % adj.n;
         break;
       case SampleStrategy::kEdgeWeight: {
-        double total = 0;
-        for (size_t i = 0; i < adj.n; ++i)
-          total += adj.w ? adj.w[i] : 1.0;
-        double r = (static_cast<double>(rng() & 0xFFFFFFFFFFFFull) /
-                    double(0x1000000000000ull)) *
-                   total;
-        double acc = 0;
-        pick = adj.n - 1;
-        for (size_t i = 0; i < adj.n; ++i) {
-          acc += adj.w ? adj.w[i] : 1.0;
-          if (r < acc) {
-            pick = i;
-            break;
-          }
-        }
+        // O(1) alias draw (Vose): pick a slot, then flip against its prob
+        size_t base = adj.dst - frag.oe_dsts().data();
+        size_t slot = rng() % adj.n;
+        double coin = static_cast<double>(rng() & 0xFFFFFFFFFFFFull) /
+                      double(0x1000000000000ull);
+        pick = coin < ctx.alias.prob[base + slot]
+                   ? slot
+                   : ctx.alias.alias[base + slot];
         break;
       }
       case SampleStrategy::kTopK: {
