@@ -265,3 +265,14 @@ def test_gpu_serialized_roundtrip(eng, tmp_path):
     assert np.array_equal(before["oids"][o1], after["oids"][o2])
     assert np.allclose(before["values"][o1], after["values"][o2],
                        rtol=1e-6)
+
+
+def test_gpu_exclusive_scan_sizes(eng):
+    # hand-rolled hierarchical scan vs numpy at boundary sizes (block
+    # chunk = 2048; recursion levels at multiples)
+    rng = np.random.default_rng(251)
+    for n in (0, 1, 2, 2047, 2048, 2049, 4096, 100000, 4194304, 4194305):
+        a = rng.integers(0, 50, size=n, dtype=np.uint32)
+        got = np.array(eng._debug_scan(a.tolist()), dtype=np.uint64)
+        exp = np.concatenate([[0], np.cumsum(a, dtype=np.uint64)])
+        assert np.array_equal(got, exp), n
