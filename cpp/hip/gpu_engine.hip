@@ -2174,7 +2174,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
 
   // capture once per graph (cached on the DeviceGraph: the instantiate
   // cost would otherwise swamp small runs; warmup absorbs it)
-  const bool capture = !multi && tol <= 0;
+  const bool capture = !multi && tol <= 0 && g.pr_calls++ > 0;
   hipGraphExec_t graph_exec =
       reinterpret_cast<hipGraphExec_t>(g.pr_graph_exec);
   if (capture && (!graph_exec || g.pr_graph_damping != damping)) {

@@ -46,6 +46,8 @@ struct DeviceGraph {
   // pointers, so these buffers must live as long as the graph
   void* pr_graph_exec = nullptr;
   double pr_graph_damping = 0.0;
+  int pr_calls = 0;  // capture lazily: one-shot runs skip the ~10ms
+                     // instantiate; repeated runs (bench warmup) get it
   DeviceBuffer<double> pr_rank, pr_acc, pr_dangling;
   DeviceBuffer<float> pr_contrib;
   ~DeviceGraph();
