@@ -2372,6 +2372,7 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
 // Directed graphs use the in+out multiset (both CSRs), like the CPU app.
 // ---------------------------------------------------------------------------
 
+constexpr uint32_t kCdlpTinyDeg = 8;       // thread-per-row register tier
 constexpr uint32_t kCdlpSmallDeg = 64;     // wave tier bound
 constexpr uint32_t kCdlpLdsSlots = 8192;   // 64 KB LDS table (mid tier)
 constexpr uint32_t kCdlpMidDeg = kCdlpLdsSlots / 2;  // load factor <= 0.5
@@ -2386,21 +2387,24 @@ __device__ __forceinline__ uint32_t cdlp_hash(uint32_t x) {
   return x;
 }
 
-// combined-degree bucketing over (out [+ in]) adjacency
+// combined-degree bucketing over (out [+ in]) adjacency.
+// tiny rows get thread-per-row register-mode treatment (tn nullptr folds
+// them into the small/wave tier — the LCC reuse keeps three tiers).
 __global__ void cdlp_bucket_kernel(const uint64_t* __restrict__ off1,
                                    const uint64_t* __restrict__ off2,
-                                   uint32_t owned, uint32_t* sm,
+                                   uint32_t owned, uint32_t* tn,
+                                   unsigned long long* ct, uint32_t* sm,
                                    unsigned long long* cs, uint32_t* md,
                                    unsigned long long* cm, uint32_t* lg,
                                    unsigned long long* cl) {
-  __shared__ uint32_t s_cnt[3];
-  __shared__ unsigned long long s_base[3];
-  uint32_t* lists[3] = {sm, md, lg};
-  unsigned long long* gcnt[3] = {cs, cm, cl};
+  __shared__ uint32_t s_cnt[4];
+  __shared__ unsigned long long s_base[4];
+  uint32_t* lists[4] = {tn, sm, md, lg};
+  unsigned long long* gcnt[4] = {ct, cs, cm, cl};
   const uint32_t stride = gridDim.x * blockDim.x;
   for (uint32_t base = blockIdx.x * blockDim.x; base < owned;
        base += stride) {
-    if (threadIdx.x < 3) s_cnt[threadIdx.x] = 0;
+    if (threadIdx.x < 4) s_cnt[threadIdx.x] = 0;
     __syncthreads();
     uint32_t r = base + threadIdx.x;
     int b = -1;
@@ -2408,17 +2412,74 @@ __global__ void cdlp_bucket_kernel(const uint64_t* __restrict__ off1,
     if (r < owned) {
       uint64_t deg = off1[r + 1] - off1[r];
       if (off2) deg += off2[r + 1] - off2[r];
-      b = deg <= kCdlpSmallDeg ? 0 : (deg <= kCdlpMidDeg ? 1 : 2);
+      b = (tn && deg <= kCdlpTinyDeg)
+              ? 0
+              : (deg <= kCdlpSmallDeg ? 1 : (deg <= kCdlpMidDeg ? 2 : 3));
       loc = atomicAdd(&s_cnt[b], 1u);
     }
     __syncthreads();
-    if (threadIdx.x < 3 && s_cnt[threadIdx.x])
+    if (threadIdx.x < 4 && s_cnt[threadIdx.x])
       s_base[threadIdx.x] =
           atomicAdd(gcnt[threadIdx.x],
                     static_cast<unsigned long long>(s_cnt[threadIdx.x]));
     __syncthreads();
     if (b >= 0) lists[b][s_base[b] + loc] = r;
     __syncthreads();
+  }
+}
+
+// thread-per-row register mode for deg <= 8 (most power-law rows):
+// a wave-per-row ballot on a deg-3 row idles 61 lanes
+__global__ void cdlp_tiny_kernel(const uint64_t* __restrict__ off1,
+                                 const uint32_t* __restrict__ dst1,
+                                 const uint64_t* __restrict__ off2,
+                                 const uint32_t* __restrict__ dst2,
+                                 const uint32_t* __restrict__ lab,
+                                 const uint32_t* __restrict__ rows,
+                                 uint64_t nrows, uint32_t v_begin,
+                                 const uint32_t* __restrict__ dirty,
+                                 uint32_t* __restrict__ next) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * blockDim.x +
+                    threadIdx.x;
+       i < nrows; i += stride) {
+    uint32_t r = rows[i];
+    if (dirty) {
+      uint32_t vg = v_begin + r;
+      if (!((dirty[vg >> 5] >> (vg & 31)) & 1)) continue;
+    }
+    uint64_t b1 = off1[r], e1 = off1[r + 1];
+    uint32_t d1 = static_cast<uint32_t>(e1 - b1);
+    uint32_t deg = d1;
+    uint64_t b2 = 0;
+    if (off2) {
+      b2 = off2[r];
+      deg += static_cast<uint32_t>(off2[r + 1] - b2);
+    }
+    if (deg == 0) {
+      next[r] = lab[v_begin + r];
+      continue;
+    }
+    uint32_t l[kCdlpTinyDeg];
+#pragma unroll
+    for (uint32_t k = 0; k < kCdlpTinyDeg; ++k)
+      if (k < deg)
+        l[k] = k < d1 ? lab[dst1[b1 + k]] : lab[dst2[b2 + (k - d1)]];
+    uint32_t best_lab = kCdlpEmpty, best_cnt = 0;
+    for (uint32_t k = 0; k < deg; ++k) {
+      bool first = true;
+      for (uint32_t j = 0; j < k; ++j)
+        if (l[j] == l[k]) first = false;
+      if (!first) continue;
+      uint32_t cnt = 1;
+      for (uint32_t j = k + 1; j < deg; ++j)
+        if (l[j] == l[k]) ++cnt;
+      if (cnt > best_cnt || (cnt == best_cnt && l[k] < best_lab)) {
+        best_cnt = cnt;
+        best_lab = l[k];
+      }
+    }
+    next[r] = best_lab;
   }
 }
 
@@ -2735,15 +2796,17 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
   const uint32_t* dst2 = g.directed ? g.ie_dst.data() : nullptr;
 
   // tier the rows once (degrees don't change across iterations)
-  DeviceBuffer<uint32_t> t_small(owned), t_mid(owned), t_large(owned);
-  DeviceBuffer<unsigned long long> cnts(3);
+  DeviceBuffer<uint32_t> t_tiny(owned), t_small(owned), t_mid(owned),
+      t_large(owned);
+  DeviceBuffer<unsigned long long> cnts(4);
   cnts.zero(s);
   if (owned)
     cdlp_bucket_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-        off1, off2, owned, t_small.data(), cnts.data() + 0, t_mid.data(),
-        cnts.data() + 1, t_large.data(), cnts.data() + 2);
+        off1, off2, owned, t_tiny.data(), cnts.data() + 0, t_small.data(),
+        cnts.data() + 1, t_mid.data(), cnts.data() + 2, t_large.data(),
+        cnts.data() + 3);
   auto hc = cnts.download(s);
-  uint64_t n_small = hc[0], n_mid = hc[1], n_large = hc[2];
+  uint64_t n_tiny = hc[0], n_small = hc[1], n_mid = hc[2], n_large = hc[3];
 
   // heavy-row global hash pool (epoch-tagged u64 slots: no per-iteration
   // clears; epoch 0 == the zeroed virgin state)
@@ -2798,6 +2861,10 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
   bool use_dirty = false;  // iteration 0 recomputes everything
   for (int it = 0; it < iters; ++it) {
     const uint32_t* dw = use_dirty ? dirty.data() : nullptr;
+    if (n_tiny)
+      cdlp_tiny_kernel<<<grid_for(n_tiny), kBlock, 0, s>>>(
+          off1, dst1, off2, dst2, lab.data(), t_tiny.data(), n_tiny,
+          g.v_begin, dw, next.data());
     if (n_small)
       cdlp_small_kernel<<<grid_for(n_small * kWave), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, lab.data(), t_small.data(), n_small,
@@ -4036,12 +4103,13 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
   {
     DeviceBuffer<uint32_t> t_small(owned ? owned : 1),
         t_mid(owned ? owned : 1), t_large(owned ? owned : 1);
-    DeviceBuffer<unsigned long long> cnts(3);
+    DeviceBuffer<unsigned long long> cnts(4);
     cnts.zero(s);
     if (owned)
       cdlp_bucket_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-          off1, off2, owned, t_small.data(), cnts.data() + 0, t_mid.data(),
-          cnts.data() + 1, t_large.data(), cnts.data() + 2);
+          off1, off2, owned, nullptr, cnts.data() + 3, t_small.data(),
+          cnts.data() + 0, t_mid.data(), cnts.data() + 1, t_large.data(),
+          cnts.data() + 2);
     auto hc = cnts.download(s);
     uint64_t n_small = hc[0], n_mid = hc[1], n_large = hc[2];
 
@@ -4241,12 +4309,13 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
   // tier rows by combined degree (reuses the CDLP bucketer)
   DeviceBuffer<uint32_t> t_small(owned), t_mid(owned), t_large(owned);
   {
-    DeviceBuffer<unsigned long long> cnts(3);
+    DeviceBuffer<unsigned long long> cnts(4);
     cnts.zero(s);
     if (owned)
       cdlp_bucket_kernel<<<grid_for(owned), kBlock, 0, s>>>(
-          off1, off2, owned, t_small.data(), cnts.data() + 0, t_mid.data(),
-          cnts.data() + 1, t_large.data(), cnts.data() + 2);
+          off1, off2, owned, nullptr, cnts.data() + 3, t_small.data(),
+          cnts.data() + 0, t_mid.data(), cnts.data() + 1, t_large.data(),
+          cnts.data() + 2);
     auto hc = cnts.download(s);
     uint64_t n_small = hc[0], n_mid = hc[1], n_large = hc[2];
 
