@@ -2372,7 +2372,7 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
 // Directed graphs use the in+out multiset (both CSRs), like the CPU app.
 // ---------------------------------------------------------------------------
 
-constexpr uint32_t kCdlpTinyDeg = 8;       // thread-per-row register tier
+constexpr uint32_t kCdlpTinyDeg = 16;      // thread-per-row register tier
 constexpr uint32_t kCdlpSmallDeg = 64;     // wave tier bound
 constexpr uint32_t kCdlpLdsSlots = 8192;   // 64 KB LDS table (mid tier)
 constexpr uint32_t kCdlpMidDeg = kCdlpLdsSlots / 2;  // load factor <= 0.5
