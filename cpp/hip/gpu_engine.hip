@@ -3546,11 +3546,39 @@ __global__ void lcc_tri_wee_kernel(const uint64_t* __restrict__ goff,
       }
       const uint32_t* t = htab + hb;
       unsigned long long hits = 0;
-      for (uint32_t e = 0; e < en; ++e) {
-        uint32_t w = gdst[eb + e];
-        if (lcc_probe(t, mask, w)) {
-          ++hits;
-          if (!skip_witness) atomicAdd(&T[w], 1ull);
+      // 4 interleaved probe chains (see the heavy kernel): the serial
+      // dependent probe left the wave parked on latency
+      for (uint32_t e0 = 0; e0 < en; e0 += 4) {
+        uint32_t key[4];
+        uint64_t idx[4];
+        bool live[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          live[j] = e0 + j < en;
+          key[j] = live[j] ? gdst[eb + e0 + j] : 0;
+          idx[j] = cdlp_hash(key[j]) & mask;
+        }
+        for (;;) {
+          uint32_t x[4];
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            if (live[j]) x[j] = t[idx[j]];
+          bool any = false;
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            if (!live[j]) continue;
+            if (x[j] == key[j]) {
+              ++hits;
+              if (!skip_witness) atomicAdd(&T[key[j]], 1ull);
+              live[j] = false;
+            } else if (x[j] == kCdlpEmpty) {
+              live[j] = false;
+            } else {
+              idx[j] = (idx[j] + 1) & mask;
+              any = true;
+            }
+          }
+          if (!any) break;
         }
       }
       if (hits) {
