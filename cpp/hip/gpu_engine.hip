@@ -2373,7 +2373,9 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
 // ---------------------------------------------------------------------------
 
 constexpr uint32_t kCdlpTinyDeg = 16;      // thread-per-row register tier
-constexpr uint32_t kCdlpSmallDeg = 64;     // wave tier bound
+constexpr uint32_t kCdlpSmallDeg = 64;     // wave-ballot tier bound
+constexpr uint32_t kCdlpWaveSlots = 1024;  // wave-hash tier (4 rows/block)
+constexpr uint32_t kCdlpWaveDeg = kCdlpWaveSlots / 2;
 constexpr uint32_t kCdlpLdsSlots = 8192;   // 64 KB LDS table (mid tier)
 constexpr uint32_t kCdlpMidDeg = kCdlpLdsSlots / 2;  // load factor <= 0.5
 constexpr uint32_t kCdlpEmpty = 0xFFFFFFFFu;
@@ -2394,17 +2396,18 @@ __global__ void cdlp_bucket_kernel(const uint64_t* __restrict__ off1,
                                    const uint64_t* __restrict__ off2,
                                    uint32_t owned, uint32_t* tn,
                                    unsigned long long* ct, uint32_t* sm,
-                                   unsigned long long* cs, uint32_t* md,
+                                   unsigned long long* cs, uint32_t* wv,
+                                   unsigned long long* cw, uint32_t* md,
                                    unsigned long long* cm, uint32_t* lg,
                                    unsigned long long* cl) {
-  __shared__ uint32_t s_cnt[4];
-  __shared__ unsigned long long s_base[4];
-  uint32_t* lists[4] = {tn, sm, md, lg};
-  unsigned long long* gcnt[4] = {ct, cs, cm, cl};
+  __shared__ uint32_t s_cnt[5];
+  __shared__ unsigned long long s_base[5];
+  uint32_t* lists[5] = {tn, sm, wv, md, lg};
+  unsigned long long* gcnt[5] = {ct, cs, cw, cm, cl};
   const uint32_t stride = gridDim.x * blockDim.x;
   for (uint32_t base = blockIdx.x * blockDim.x; base < owned;
        base += stride) {
-    if (threadIdx.x < 4) s_cnt[threadIdx.x] = 0;
+    if (threadIdx.x < 5) s_cnt[threadIdx.x] = 0;
     __syncthreads();
     uint32_t r = base + threadIdx.x;
     int b = -1;
@@ -2412,13 +2415,20 @@ __global__ void cdlp_bucket_kernel(const uint64_t* __restrict__ off1,
     if (r < owned) {
       uint64_t deg = off1[r + 1] - off1[r];
       if (off2) deg += off2[r + 1] - off2[r];
-      b = (tn && deg <= kCdlpTinyDeg)
-              ? 0
-              : (deg <= kCdlpSmallDeg ? 1 : (deg <= kCdlpMidDeg ? 2 : 3));
+      if (tn && deg <= kCdlpTinyDeg)
+        b = 0;
+      else if (deg <= kCdlpSmallDeg)
+        b = 1;
+      else if (wv && deg <= kCdlpWaveDeg)
+        b = 2;
+      else if (deg <= kCdlpMidDeg)
+        b = 3;
+      else
+        b = 4;
       loc = atomicAdd(&s_cnt[b], 1u);
     }
     __syncthreads();
-    if (threadIdx.x < 4 && s_cnt[threadIdx.x])
+    if (threadIdx.x < 5 && s_cnt[threadIdx.x])
       s_base[threadIdx.x] =
           atomicAdd(gcnt[threadIdx.x],
                     static_cast<unsigned long long>(s_cnt[threadIdx.x]));
@@ -2558,6 +2568,91 @@ __device__ __forceinline__ void cdlp_block_argmax(uint64_t key,
     *out = best ? ~static_cast<uint32_t>(best) : fallback;
   }
   __syncthreads();
+}
+
+// 4-rows-per-block wave hash for 64 < deg <= kCdlpWaveDeg: a block-per-row
+// table on a deg-100 row leaves 3 of 4 waves idle; here each wave owns a
+// 1024-slot LDS region and the group loop is uniform so plain
+// __syncthreads stages the clear/insert/reduce phases.
+__global__ void cdlp_wave_kernel(const uint64_t* __restrict__ off1,
+                                 const uint32_t* __restrict__ dst1,
+                                 const uint64_t* __restrict__ off2,
+                                 const uint32_t* __restrict__ dst2,
+                                 const uint32_t* __restrict__ lab,
+                                 const uint32_t* __restrict__ rows,
+                                 uint64_t nrows, uint32_t v_begin,
+                                 const uint32_t* __restrict__ dirty,
+                                 uint32_t* __restrict__ next) {
+  __shared__ uint32_t s_lab[4][kCdlpWaveSlots];
+  __shared__ uint32_t s_cnt[4][kCdlpWaveSlots];
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  uint64_t groups = (nrows + 3) / 4;
+  for (uint64_t grp = blockIdx.x; grp < groups; grp += gridDim.x) {
+    uint64_t i = grp * 4 + wid;
+    bool have = i < nrows;
+    uint32_t r = have ? rows[i] : 0;
+    if (have && dirty) {
+      uint32_t vg = v_begin + r;
+      if (!((dirty[vg >> 5] >> (vg & 31)) & 1)) have = false;
+    }
+    uint32_t d1 = 0, deg = 0;
+    uint64_t b1 = 0, b2 = 0;
+    uint32_t cap = 64, mask = 63;
+    if (have) {
+      b1 = off1[r];
+      d1 = static_cast<uint32_t>(off1[r + 1] - b1);
+      deg = d1;
+      if (off2) {
+        b2 = off2[r];
+        deg += static_cast<uint32_t>(off2[r + 1] - b2);
+      }
+      while (cap < 2 * deg) cap <<= 1;
+      if (cap > kCdlpWaveSlots) cap = kCdlpWaveSlots;
+      mask = cap - 1;
+      for (uint32_t k = lane; k < cap; k += kWave) {
+        s_lab[wid][k] = kCdlpEmpty;
+        s_cnt[wid][k] = 0;
+      }
+    }
+    __syncthreads();
+    if (have) {
+      for (uint32_t k = lane; k < deg; k += kWave) {
+        uint32_t l = k < d1 ? lab[dst1[b1 + k]] : lab[dst2[b2 + (k - d1)]];
+        uint32_t idx = cdlp_hash(l) & mask;
+        for (;;) {
+          uint32_t old = atomicCAS(&s_lab[wid][idx], kCdlpEmpty, l);
+          if (old == kCdlpEmpty || old == l) {
+            atomicAdd(&s_cnt[wid][idx], 1u);
+            break;
+          }
+          idx = (idx + 1) & mask;
+        }
+      }
+    }
+    __syncthreads();
+    if (have) {
+      uint64_t key = 0;
+      for (uint32_t k = lane; k < cap; k += kWave) {
+        uint32_t c = s_cnt[wid][k];
+        if (c) {
+          uint64_t cand = (static_cast<uint64_t>(c) << 32) |
+                          static_cast<uint32_t>(~s_lab[wid][k]);
+          if (cand > key) key = cand;
+        }
+      }
+#pragma unroll
+      for (int d = 32; d > 0; d >>= 1) {
+        uint64_t o = __shfl_down(static_cast<unsigned long long>(key), d,
+                                 64);
+        if (o > key) key = o;
+      }
+      if (lane == 0)
+        next[r] = key ? ~static_cast<uint32_t>(key)
+                      : (deg ? 0 : lab[v_begin + r]);
+    }
+    __syncthreads();
+  }
 }
 
 // block-per-row LDS hash, deg <= kCdlpMidDeg
@@ -2796,17 +2891,18 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
   const uint32_t* dst2 = g.directed ? g.ie_dst.data() : nullptr;
 
   // tier the rows once (degrees don't change across iterations)
-  DeviceBuffer<uint32_t> t_tiny(owned), t_small(owned), t_mid(owned),
-      t_large(owned);
-  DeviceBuffer<unsigned long long> cnts(4);
+  DeviceBuffer<uint32_t> t_tiny(owned), t_small(owned), t_wave(owned),
+      t_mid(owned), t_large(owned);
+  DeviceBuffer<unsigned long long> cnts(5);
   cnts.zero(s);
   if (owned)
     cdlp_bucket_kernel<<<grid_for(owned), kBlock, 0, s>>>(
         off1, off2, owned, t_tiny.data(), cnts.data() + 0, t_small.data(),
-        cnts.data() + 1, t_mid.data(), cnts.data() + 2, t_large.data(),
-        cnts.data() + 3);
+        cnts.data() + 1, t_wave.data(), cnts.data() + 2, t_mid.data(),
+        cnts.data() + 3, t_large.data(), cnts.data() + 4);
   auto hc = cnts.download(s);
-  uint64_t n_tiny = hc[0], n_small = hc[1], n_mid = hc[2], n_large = hc[3];
+  uint64_t n_tiny = hc[0], n_small = hc[1], n_wave = hc[2], n_mid = hc[3],
+           n_large = hc[4];
 
   // heavy-row global hash pool (epoch-tagged u64 slots: no per-iteration
   // clears; epoch 0 == the zeroed virgin state)
@@ -2869,6 +2965,11 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
       cdlp_small_kernel<<<grid_for(n_small * kWave), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, lab.data(), t_small.data(), n_small,
           g.v_begin, dw, next.data());
+    if (n_wave)
+      cdlp_wave_kernel<<<grid_for(((n_wave + 3) / 4) * kBlock), kBlock, 0,
+                         s>>>(off1, dst1, off2, dst2, lab.data(),
+                              t_wave.data(), n_wave, g.v_begin, dw,
+                              next.data());
     if (n_mid)
       cdlp_mid_kernel<<<std::min<int>(n_mid, kMaxGrid), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, lab.data(), t_mid.data(), n_mid, g.v_begin,
@@ -4131,13 +4232,13 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
   {
     DeviceBuffer<uint32_t> t_small(owned ? owned : 1),
         t_mid(owned ? owned : 1), t_large(owned ? owned : 1);
-    DeviceBuffer<unsigned long long> cnts(4);
+    DeviceBuffer<unsigned long long> cnts(5);
     cnts.zero(s);
     if (owned)
       cdlp_bucket_kernel<<<grid_for(owned), kBlock, 0, s>>>(
           off1, off2, owned, nullptr, cnts.data() + 3, t_small.data(),
-          cnts.data() + 0, t_mid.data(), cnts.data() + 1, t_large.data(),
-          cnts.data() + 2);
+          cnts.data() + 0, nullptr, cnts.data() + 4, t_mid.data(),
+          cnts.data() + 1, t_large.data(), cnts.data() + 2);
     auto hc = cnts.download(s);
     uint64_t n_small = hc[0], n_mid = hc[1], n_large = hc[2];
 
@@ -4337,13 +4438,13 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
   // tier rows by combined degree (reuses the CDLP bucketer)
   DeviceBuffer<uint32_t> t_small(owned), t_mid(owned), t_large(owned);
   {
-    DeviceBuffer<unsigned long long> cnts(4);
+    DeviceBuffer<unsigned long long> cnts(5);
     cnts.zero(s);
     if (owned)
       cdlp_bucket_kernel<<<grid_for(owned), kBlock, 0, s>>>(
           off1, off2, owned, nullptr, cnts.data() + 3, t_small.data(),
-          cnts.data() + 0, t_mid.data(), cnts.data() + 1, t_large.data(),
-          cnts.data() + 2);
+          cnts.data() + 0, nullptr, cnts.data() + 4, t_mid.data(),
+          cnts.data() + 1, t_large.data(), cnts.data() + 2);
     auto hc = cnts.download(s);
     uint64_t n_small = hc[0], n_mid = hc[1], n_large = hc[2];
 

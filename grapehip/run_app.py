@@ -38,6 +38,10 @@ def build_parser():
                     help="parse a third edge column (required by sssp)")
     ap.add_argument("--gpu", action="store_true")
     ap.add_argument("--app_concurrency", type=int, default=0)
+    ap.add_argument("--idxer", default="hashmap",
+                    choices=("hashmap", "mph"),
+                    help="oid->lid index for --vfile graphs "
+                         "(reference --idxer_type)")
     # per-app parameters (flags.cc:23-67)
     ap.add_argument("--bfs_source", type=int, default=0)
     ap.add_argument("--sssp_source", type=int, default=0)
@@ -86,6 +90,7 @@ def main(argv=None):
         if args.vfile:
             oids = read_ldbc_vertices(args.vfile, rank=rank, world=world)
             kw["vertex_oids"] = oids
+            kw["idxer"] = args.idxer
         else:
             hi = max(src.max(initial=0), dst.max(initial=0)) + 1
             kw["num_vertices"] = int(hi)

@@ -122,16 +122,27 @@ def test_gpu_cdlp_directed(eng):
 
 
 def test_gpu_cdlp_hub_tiers(eng):
-    # star hubs exercise the LDS-hash (mid) and global-hash (large) tiers
+    # exercises every CDLP tier: tiny (<=16), ballot (<=64), wave-hash
+    # (<=512), block LDS hash (<=4096), global hash (hub)
     rng = np.random.default_rng(31)
     nv = 30000
     hub_e = np.stack([np.zeros(12000, np.int64),
                       rng.integers(1, nv, 12000)], 1)
     mid_e = np.stack([np.ones(2000, np.int64),
                       rng.integers(2, nv, 2000)], 1)
+    wave_rows = []
+    for v in range(2, 12):  # ten rows with degree ~100-400 (wave tier)
+        d = int(rng.integers(80, 400))
+        wave_rows.append(np.stack([np.full(d, v, np.int64),
+                                   rng.integers(0, nv, d)], 1))
+    ballot_rows = []
+    for v in range(12, 30):  # rows with degree ~20-60 (ballot tier)
+        d = int(rng.integers(18, 60))
+        ballot_rows.append(np.stack([np.full(d, v, np.int64),
+                                     rng.integers(0, nv, d)], 1))
     rest = np.stack([rng.integers(0, nv, 40000),
                      rng.integers(0, nv, 40000)], 1)
-    e = np.concatenate([hub_e, mid_e, rest])
+    e = np.concatenate([hub_e, mid_e, *wave_rows, *ballot_rows, rest])
     keep = e[:, 0] != e[:, 1]
     src, dst = e[keep, 0], e[keep, 1]
     g = eng.load_edges(src, dst, directed=False, num_vertices=nv)
