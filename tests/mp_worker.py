@@ -53,6 +53,17 @@ def main():
                        build_in_csr=cfg["in_csr"], **kw)
 
     app = cfg["app"]
+    if cfg.get("mutate"):
+        # collective delta: each rank contributes a slice of adds/removes
+        rng2 = np.random.default_rng(cfg["seed"] + 1)
+        ad_s = rng2.integers(0, cfg["num_v"], 300).astype(np.int64)
+        ad_d = rng2.integers(0, cfg["num_v"], 300).astype(np.int64)
+        k = ad_s != ad_d
+        ad_s, ad_d = ad_s[k], ad_d[k]
+        msl = slice(rank, None, world)
+        g = eng.mutate_graph(g, add_src=ad_s[msl], add_dst=ad_d[msl],
+                             remove_src=src[sl][:50], remove_dst=dst[sl][:50],
+                             remove_vertices=np.array([], dtype=np.int64))
     if app == "sssp_auto":
         res = eng.sssp_auto(g, cfg["source"])
     elif app == "bfs":

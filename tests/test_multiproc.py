@@ -215,3 +215,35 @@ def test_bfs_mp_mph_idxer(world, free_port, tmp_path):
     src, dst, _ = graph_arrays(cfg)
     assert np.array_equal(oids, np.arange(cfg["num_v"]) * 3 + 1)
     assert np.array_equal(vals, bfs_oracle(cfg["num_v"], src, dst, 3))
+
+
+@pytest.mark.parametrize("world", [8])
+def test_bfs_mp_world8(world, free_port, tmp_path):
+    # the SCALE tier's rank count: full 8-way TCP mesh + shuffle + halo
+    cfg = dict(BASE, app="bfs", num_v=500, num_e=3000)
+    oids, vals = run_world(world, cfg, free_port, tmp_path, timeout=240)
+    src, dst, _ = graph_arrays(cfg)
+    assert np.array_equal(vals, bfs_oracle(cfg["num_v"], src, dst, 3))
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_mutation_mp(world, free_port, tmp_path):
+    # collective mutate: per-rank delta slices must merge identically
+    cfg = dict(BASE, app="bfs", mutate=True, directed=True)
+    oids, vals = run_world(world, cfg, free_port, tmp_path)
+    src, dst, _ = graph_arrays(cfg)
+    rng2 = np.random.default_rng(cfg["seed"] + 1)
+    ad_s = rng2.integers(0, cfg["num_v"], 300).astype(np.int64)
+    ad_d = rng2.integers(0, cfg["num_v"], 300).astype(np.int64)
+    k = ad_s != ad_d
+    ad_s, ad_d = ad_s[k], ad_d[k]
+    # removals: every rank removed the first 50 of ITS slice = a global set
+    rm = set()
+    for r in range(world):
+        sl = slice(r, None, world)
+        for a, b in zip(src[sl][:50], dst[sl][:50]):
+            rm.add((a, b))
+    keep = np.array([(a, b) not in rm for a, b in zip(src, dst)])
+    e_src = np.concatenate([src[keep], ad_s])
+    e_dst = np.concatenate([dst[keep], ad_d])
+    assert np.array_equal(vals, bfs_oracle(cfg["num_v"], e_src, e_dst, 3))
