@@ -3215,6 +3215,123 @@ __global__ void lcc_distinct_large_kernel(
   }
 }
 
+// 4-rows-per-block wave variants for 64 < deg <= kCdlpWaveDeg (a block-
+// per-row hash on a deg-100 row idles 3 of 4 waves; same uniform-barrier
+// staging as cdlp_wave_kernel)
+__global__ void lcc_distinct_wave_kernel(
+    const uint64_t* __restrict__ off1, const uint32_t* __restrict__ dst1,
+    const uint64_t* __restrict__ off2, const uint32_t* __restrict__ dst2,
+    const uint32_t* __restrict__ rows, uint64_t nrows, uint32_t v_begin,
+    uint32_t* __restrict__ D) {
+  __shared__ uint32_t s_lab[4][kCdlpWaveSlots];
+  __shared__ uint32_t s_distinct[4];
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  uint64_t groups = (nrows + 3) / 4;
+  for (uint64_t grp = blockIdx.x; grp < groups; grp += gridDim.x) {
+    uint64_t i = grp * 4 + wid;
+    const bool have = i < nrows;
+    uint32_t r = have ? rows[i] : 0;
+    uint32_t v = v_begin + r;
+    uint32_t d1 = 0, deg = 0, cap = 64, mask = 63;
+    uint64_t b1 = 0, b2 = 0;
+    if (have) {
+      b1 = off1[r];
+      d1 = static_cast<uint32_t>(off1[r + 1] - b1);
+      deg = d1;
+      if (off2) {
+        b2 = off2[r];
+        deg += static_cast<uint32_t>(off2[r + 1] - b2);
+      }
+      while (cap < 2 * deg) cap <<= 1;
+      if (cap > kCdlpWaveSlots) cap = kCdlpWaveSlots;
+      mask = cap - 1;
+      for (uint32_t k = lane; k < cap; k += kWave)
+        s_lab[wid][k] = kCdlpEmpty;
+      if (lane == 0) s_distinct[wid] = 0;
+    }
+    __syncthreads();
+    if (have) {
+      for (uint32_t k = lane; k < deg; k += kWave) {
+        uint32_t u = k < d1 ? dst1[b1 + k] : dst2[b2 + (k - d1)];
+        if (u == v) continue;
+        uint32_t idx = cdlp_hash(u) & mask;
+        for (;;) {
+          uint32_t old = atomicCAS(&s_lab[wid][idx], kCdlpEmpty, u);
+          if (old == kCdlpEmpty) {
+            atomicAdd(&s_distinct[wid], 1u);
+            break;
+          }
+          if (old == u) break;
+          idx = (idx + 1) & mask;
+        }
+      }
+    }
+    __syncthreads();
+    if (have && lane == 0) D[v] = s_distinct[wid];
+    __syncthreads();
+  }
+}
+
+__global__ void lcc_orient_wave_kernel(
+    const uint64_t* __restrict__ off1, const uint32_t* __restrict__ dst1,
+    const uint64_t* __restrict__ off2, const uint32_t* __restrict__ dst2,
+    const uint32_t* __restrict__ rows, uint64_t nrows, uint32_t v_begin,
+    const uint32_t* __restrict__ D, const uint64_t* __restrict__ ooff,
+    uint32_t* __restrict__ oadj, uint32_t* __restrict__ ocnt, bool filter) {
+  __shared__ uint32_t s_lab[4][kCdlpWaveSlots];
+  __shared__ uint32_t s_cursor[4];
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  uint64_t groups = (nrows + 3) / 4;
+  for (uint64_t grp = blockIdx.x; grp < groups; grp += gridDim.x) {
+    uint64_t i = grp * 4 + wid;
+    const bool have = i < nrows;
+    uint32_t r = have ? rows[i] : 0;
+    uint32_t v = v_begin + r;
+    uint32_t d1 = 0, deg = 0, cap = 64, mask = 63;
+    uint64_t b1 = 0, b2 = 0;
+    if (have) {
+      b1 = off1[r];
+      d1 = static_cast<uint32_t>(off1[r + 1] - b1);
+      deg = d1;
+      if (off2) {
+        b2 = off2[r];
+        deg += static_cast<uint32_t>(off2[r + 1] - b2);
+      }
+      while (cap < 2 * deg) cap <<= 1;
+      if (cap > kCdlpWaveSlots) cap = kCdlpWaveSlots;
+      mask = cap - 1;
+      for (uint32_t k = lane; k < cap; k += kWave)
+        s_lab[wid][k] = kCdlpEmpty;
+      if (lane == 0) s_cursor[wid] = 0;
+    }
+    __syncthreads();
+    if (have) {
+      uint32_t dv = D[v];
+      uint64_t base = ooff[r];
+      for (uint32_t k = lane; k < deg; k += kWave) {
+        uint32_t u = k < d1 ? dst1[b1 + k] : dst2[b2 + (k - d1)];
+        if (u == v) continue;
+        uint32_t idx = cdlp_hash(u) & mask;
+        for (;;) {
+          uint32_t old = atomicCAS(&s_lab[wid][idx], kCdlpEmpty, u);
+          if (old == kCdlpEmpty) {
+            if (!filter || lcc_keep(D[u], u, dv, v))
+              oadj[base + atomicAdd(&s_cursor[wid], 1u)] = u;
+            break;
+          }
+          if (old == u) break;
+          idx = (idx + 1) & mask;
+        }
+      }
+    }
+    __syncthreads();
+    if (have && lane == 0) ocnt[r] = s_cursor[wid];
+    __syncthreads();
+  }
+}
+
 // --- pass 2: build oriented adjacency (dedup + orientation filter) --------
 
 __global__ void lcc_orient_small_kernel(
@@ -4232,15 +4349,17 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
   {
     DeviceBuffer<uint32_t> t_small(owned ? owned : 1),
         t_mid(owned ? owned : 1), t_large(owned ? owned : 1);
+    DeviceBuffer<uint32_t> t_wavelds(owned ? owned : 1);
     DeviceBuffer<unsigned long long> cnts(5);
     cnts.zero(s);
     if (owned)
       cdlp_bucket_kernel<<<grid_for(owned), kBlock, 0, s>>>(
           off1, off2, owned, nullptr, cnts.data() + 3, t_small.data(),
-          cnts.data() + 0, nullptr, cnts.data() + 4, t_mid.data(),
+          cnts.data() + 0, t_wavelds.data(), cnts.data() + 4, t_mid.data(),
           cnts.data() + 1, t_large.data(), cnts.data() + 2);
     auto hc = cnts.download(s);
-    uint64_t n_small = hc[0], n_mid = hc[1], n_large = hc[2];
+    uint64_t n_small = hc[0], n_mid = hc[1], n_large = hc[2],
+             n_wavelds = hc[4];
 
     DeviceBuffer<uint32_t> heavy_caps;
     DeviceBuffer<uint64_t> heavy_off;
@@ -4262,6 +4381,12 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
                                   s>>>(off1, dst1, off2, dst2,
                                        t_small.data(), n_small, g.v_begin,
                                        Dv.data());
+    if (n_wavelds)
+      lcc_distinct_wave_kernel<<<grid_for(((n_wavelds + 3) / 4) * kBlock),
+                                 kBlock, 0, s>>>(off1, dst1, off2, dst2,
+                                                 t_wavelds.data(),
+                                                 n_wavelds, g.v_begin,
+                                                 Dv.data());
     if (n_mid)
       lcc_distinct_mid_kernel<<<std::min<int>(n_mid, kMaxGrid), kBlock, 0,
                                 s>>>(off1, dst1, off2, dst2, t_mid.data(),
@@ -4302,6 +4427,11 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
                                 s>>>(o1, d1, o2, d2, t_mid.data(), n_mid,
                                      g.v_begin, Dv.data(), ooff.data(),
                                      oadj.data(), ocnt.data(), false);
+      if (n_wavelds)
+        lcc_orient_wave_kernel<<<grid_for(((n_wavelds + 3) / 4) * kBlock),
+                                 kBlock, 0, s>>>(
+            o1, d1, o2, d2, t_wavelds.data(), n_wavelds, g.v_begin,
+            Dv.data(), ooff.data(), oadj.data(), ocnt.data(), false);
       if (n_large) {
         fill(tbl_lab.data(), kCdlpEmpty, heavy_total, s);
         lcc_orient_large_kernel<<<std::min<int>(n_large, kMaxGrid), kBlock,
@@ -4438,15 +4568,17 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
   // tier rows by combined degree (reuses the CDLP bucketer)
   DeviceBuffer<uint32_t> t_small(owned), t_mid(owned), t_large(owned);
   {
+    DeviceBuffer<uint32_t> t_wavelds(owned ? owned : 1);
     DeviceBuffer<unsigned long long> cnts(5);
     cnts.zero(s);
     if (owned)
       cdlp_bucket_kernel<<<grid_for(owned), kBlock, 0, s>>>(
           off1, off2, owned, nullptr, cnts.data() + 3, t_small.data(),
-          cnts.data() + 0, nullptr, cnts.data() + 4, t_mid.data(),
+          cnts.data() + 0, t_wavelds.data(), cnts.data() + 4, t_mid.data(),
           cnts.data() + 1, t_large.data(), cnts.data() + 2);
     auto hc = cnts.download(s);
-    uint64_t n_small = hc[0], n_mid = hc[1], n_large = hc[2];
+    uint64_t n_small = hc[0], n_mid = hc[1], n_large = hc[2],
+             n_wavelds = hc[4];
 
     // heavy-row hash pool (labels only)
     DeviceBuffer<uint32_t> heavy_caps;
@@ -4470,6 +4602,12 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
       lcc_distinct_small_kernel<<<grid_for(n_small * kWave), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, t_small.data(), n_small, g.v_begin,
           Dv.data());
+    if (n_wavelds)
+      lcc_distinct_wave_kernel<<<grid_for(((n_wavelds + 3) / 4) * kBlock),
+                                 kBlock, 0, s>>>(off1, dst1, off2, dst2,
+                                                 t_wavelds.data(),
+                                                 n_wavelds, g.v_begin,
+                                                 Dv.data());
     if (n_mid)
       lcc_distinct_mid_kernel<<<std::min<int>(n_mid, kMaxGrid), kBlock, 0,
                                 s>>>(off1, dst1, off2, dst2, t_mid.data(),
@@ -4504,6 +4642,11 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
       lcc_orient_mid_kernel<<<std::min<int>(n_mid, kMaxGrid), kBlock, 0, s>>>(
           off1, dst1, off2, dst2, t_mid.data(), n_mid, g.v_begin, Dv.data(),
           ooff.data(), oadj.data(), ocnt.data(), true);
+    if (n_wavelds)
+      lcc_orient_wave_kernel<<<grid_for(((n_wavelds + 3) / 4) * kBlock),
+                               kBlock, 0, s>>>(
+          off1, dst1, off2, dst2, t_wavelds.data(), n_wavelds, g.v_begin,
+          Dv.data(), ooff.data(), oadj.data(), ocnt.data(), true);
     if (n_large) {
       fill(tbl_lab.data(), kCdlpEmpty, heavy_total, s);
       lcc_orient_large_kernel<<<std::min<int>(n_large, kMaxGrid), kBlock, 0,
