@@ -53,6 +53,14 @@ def main():
                        build_in_csr=cfg["in_csr"], **kw)
 
     app = cfg["app"]
+    if cfg.get("serialize_roundtrip"):
+        # per-fragment checkpoint files in a shared dir; reload replaces g
+        ckpt = os.path.join(cfg["out_dir"], "ckpt")
+        if rank == 0:
+            os.makedirs(ckpt, exist_ok=True)
+        eng.barrier()
+        eng.save_graph(g, ckpt)
+        g = eng.load_serialized(ckpt)
     if cfg.get("mutate"):
         # collective delta: each rank contributes a slice of adds/removes
         rng2 = np.random.default_rng(cfg["seed"] + 1)

@@ -247,3 +247,21 @@ def test_mutation_mp(world, free_port, tmp_path):
     e_src = np.concatenate([src[keep], ad_s])
     e_dst = np.concatenate([dst[keep], ad_d])
     assert np.array_equal(vals, bfs_oracle(cfg["num_v"], e_src, e_dst, 3))
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_serialize_roundtrip_mp(world, free_port, tmp_path):
+    # per-rank checkpoint + reload mid-run (reference app_tests.sh:53-66)
+    cfg = dict(BASE, app="bfs", serialize_roundtrip=True)
+    oids, vals = run_world(world, cfg, free_port, tmp_path)
+    src, dst, _ = graph_arrays(cfg)
+    assert np.array_equal(vals, bfs_oracle(cfg["num_v"], src, dst, 3))
+
+
+@pytest.mark.parametrize("world", [2])
+def test_serialize_roundtrip_mp_hashmap(world, free_port, tmp_path):
+    cfg = dict(BASE, app="bfs", serialize_roundtrip=True, vertex_oids=True,
+               source=3 * 3 + 1)
+    oids, vals = run_world(world, cfg, free_port, tmp_path)
+    src, dst, _ = graph_arrays(cfg)
+    assert np.array_equal(vals, bfs_oracle(cfg["num_v"], src, dst, 3))
