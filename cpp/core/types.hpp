@@ -35,7 +35,8 @@ enum class PartitionerKind : uint8_t {
   kSegmented = 1, // contiguous oid ranges (natural for synthetic/renumbered)
 };
 
-// Load-balancing strategy for GPU neighbor expansion (see hip/engine.hpp).
+// Load-balancing strategy for GPU neighbor expansion (hip/gpu_engine.hip,
+// runtime-selected via GRAPEHIP_LB like the reference --lb flag).
 enum class LB : uint8_t {
   kNone = 0,    // thread-per-vertex grid stride
   kCM = 1,      // per-block shared-prefix owner search (global scan)
