@@ -28,6 +28,7 @@ DATASETS = {
     "datagen-9_0-fb": (404_817_003, 1_010_447_118),
     "graph500-26": (67_108_864, 1_073_741_824),
     "com-friendster-shaped": (65_608_366, 1_806_067_135),
+    "graph500-27": (134_217_728, 2_147_483_648),
     "p2p-31-shaped": (6_300, 148_000),
     # reference GPU-table graphs (Performance.md:80-97, 8x V100)
     "soc-LiveJournal1-shaped": (4_847_571, 68_993_773),
