@@ -114,6 +114,8 @@ std::shared_ptr<PyGraph> load_edges(
   if (static_cast<size_t>(dst.size()) != n)
     throw std::runtime_error("src/dst size mismatch");
   bool weighted = weights.has_value();
+  if (weighted && static_cast<size_t>(weights->size()) != n)
+    throw std::runtime_error("weights size mismatch (expected one per edge)");
   std::vector<EdgeTriple> edges(n);
   {
     auto s = src.unchecked<1>();

@@ -51,3 +51,27 @@ def test_roundtrip_hashmap_oids(eng, tmp_path):
     o2 = np.argsort(after["oids"])
     assert np.array_equal(before["oids"][o1], after["oids"][o2])
     assert np.allclose(before["values"][o1], after["values"][o2], rtol=1e-14)
+
+
+def test_missing_checkpoint_raises(eng, tmp_path):
+    with pytest.raises(RuntimeError, match="cannot open"):
+        eng.load_serialized(str(tmp_path / "nope"))
+
+
+def test_bad_magic_raises(eng, tmp_path):
+    p = tmp_path / "frag_0.s"
+    p.write_bytes(b"\x00" * 64)
+    with pytest.raises(RuntimeError, match="bad magic"):
+        eng.load_serialized(str(tmp_path))
+
+
+def test_truncated_checkpoint_raises(eng, tmp_path):
+    src, dst, w = random_graph(num_v=300, num_e=2000, seed=77)
+    g = eng.load_edges(src, dst, weights=w, directed=False,
+                       num_vertices=300)
+    eng.save_graph(g, str(tmp_path))
+    p = tmp_path / "frag_0.s"
+    data = p.read_bytes()
+    p.write_bytes(data[: len(data) // 2])
+    with pytest.raises(RuntimeError, match="short read"):
+        eng.load_serialized(str(tmp_path))
