@@ -4469,7 +4469,9 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
 
   DeviceBuffer<unsigned long long> Tcnt(nv_pad);
   Tcnt.zero(s);
-  constexpr uint32_t kHeavyThresh = 96;
+  uint32_t kHeavyThresh = 96;  // min-side size above which an edge defers
+  if (const char* ht = getenv("GRAPEHIP_LCC_HEAVY"))
+    kHeavyThresh = static_cast<uint32_t>(atoi(ht));
   DeviceBuffer<unsigned long long> heavy_q;
   DeviceBuffer<unsigned long long> heavy_n(1);
   heavy_n.zero(s);
@@ -4753,7 +4755,9 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
   // wave-per-edge pass (power-law tail otherwise stalls single lanes)
   Tcnt.resize(nv_pad);
   Tcnt.zero(s);
-  constexpr uint32_t kHeavyThresh = 96;
+  uint32_t kHeavyThresh = 96;  // min-side size above which an edge defers
+  if (const char* ht = getenv("GRAPEHIP_LCC_HEAVY"))
+    kHeavyThresh = static_cast<uint32_t>(atoi(ht));
   const bool nowit = getenv("GRAPEHIP_LCC_NOWIT") != nullptr;
   DeviceBuffer<unsigned long long> heavy_q;
   DeviceBuffer<unsigned long long> heavy_n(1);
