@@ -17,6 +17,14 @@ else
 HIP_OBJS :=
 endif
 
+# Address-sanitized host build for the CPU engine (reference WITH_ASAN,
+# CMakeLists:15,87-90): make clean && make WITH_ASAN=1 WITH_HIP=0
+WITH_ASAN ?= 0
+ifeq ($(WITH_ASAN),1)
+CXXFLAGS += -fsanitize=address -fno-omit-frame-pointer -g
+LDFLAGS += -fsanitize=address
+endif
+
 TARGET := grapehip/_core.so
 
 all: $(TARGET)

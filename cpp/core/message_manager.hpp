@@ -14,6 +14,9 @@
 // cpp/hip/.
 #pragma once
 
+#include <chrono>
+#include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <string>
 #include <vector>
@@ -107,6 +110,15 @@ class MessageManager {
     }
     terminated_ = (global_moved == 0);
     ++round_;
+  }
+
+  // bytes queued in channels this round (message-size accounting parity:
+  // reference GetMsgSize, default_message_manager.h:146)
+  uint64_t bytes_sent() const {
+    uint64_t n = 0;
+    for (auto& per_tid : channels_)
+      for (auto& a : per_tid) n += a.size();
+    return n;
   }
 
   bool to_terminate() const { return terminated_; }
@@ -205,20 +217,42 @@ class MessageManager {
   std::vector<OutArchive> recv_;                  // [src fid]
 };
 
-// BSP superstep driver (reference worker.h:105-146).
+// BSP superstep driver (reference worker.h:105-146). GRAPEHIP_TRACE=1
+// prints per-superstep wall time and bytes shipped (the reference's
+// coordinator VLOG(1) round log, worker.h:120-139).
 template <typename APP, typename CTX>
 inline int RunWorker(APP& app, CTX& ctx, const Fragment& frag,
                      MessageManager& mm) {
+  const bool trace = getenv("GRAPEHIP_TRACE") != nullptr;
+  auto now = [] {
+    return std::chrono::duration<double>(
+               std::chrono::steady_clock::now().time_since_epoch())
+        .count();
+  };
+  double t0 = trace ? now() : 0.0;
   mm.start_round();
   app.PEval(frag, ctx, mm);
+  uint64_t sent = trace ? mm.bytes_sent() : 0;
   mm.finish_round();
+  if (trace)
+    fprintf(stderr, "[trace f%u] PEval  %8.2f ms  sent %zu B\n",
+            frag.fid(), (now() - t0) * 1e3, static_cast<size_t>(sent));
   int rounds = 1;
   while (!mm.to_terminate()) {
+    double tr = trace ? now() : 0.0;
     mm.start_round();
     app.IncEval(frag, ctx, mm);
+    sent = trace ? mm.bytes_sent() : 0;
     mm.finish_round();
+    if (trace)
+      fprintf(stderr, "[trace f%u] round %-3d %6.2f ms  sent %zu B\n",
+              frag.fid(), rounds, (now() - tr) * 1e3,
+              static_cast<size_t>(sent));
     ++rounds;
   }
+  if (trace)
+    fprintf(stderr, "[trace f%u] done: %d supersteps, %.2f ms total\n",
+            frag.fid(), rounds, (now() - t0) * 1e3);
   return rounds;
 }
 
