@@ -397,6 +397,30 @@ PYBIND11_MODULE(_core, m) {
              return out;
            },
            py::arg("blobs"))
+      // test hook: validates the cooperative-abort path (reference
+      // ForceTerminate/TerminateInfo, default_message_manager.h:156-166) —
+      // one rank aborts in PEval, EVERY rank must raise with its info
+      .def("_test_force_terminate",
+           [](PyEngine& eng, std::shared_ptr<PyGraph> g, int fail_rank) {
+             struct AbortApp {
+               int fail_rank, rank;
+               void PEval(const Fragment&, int&, MessageManager& mm) {
+                 if (rank == fail_rank)
+                   mm.force_terminate("boom from rank " +
+                                      std::to_string(rank));
+                 else
+                   mm.force_continue();
+               }
+               void IncEval(const Fragment&, int&, MessageManager&) {}
+             };
+             AbortApp app{fail_rank, eng.rank};
+             int ctx = 0;
+             MessageManager mm;
+             mm.init(eng.c(), g->frag.get(), eng.n_threads);
+             py::gil_scoped_release rel;
+             return RunWorker(app, ctx, *g->frag, mm);
+           },
+           py::arg("graph"), py::arg("fail_rank"))
       .def("load_edges", &load_edges, py::arg("src"), py::arg("dst"),
            py::arg("weights") = std::nullopt, py::arg("directed") = false,
            py::arg("num_vertices") = -1, py::arg("vertex_oids") = std::nullopt,

@@ -100,6 +100,17 @@ def main():
         r = eng.sample(g, np.arange(cfg.get("n_walks", 20), dtype=np.int64),
                        hops=cfg.get("hops", 3), seed=5)
         res = {"oids": r["walk_ids"], "values": r["paths"]}
+    elif app == "force_terminate":
+        # cooperative abort: rank 1 aborts; EVERY rank must raise with
+        # the aborting rank's info string at the round boundary
+        try:
+            eng._test_force_terminate(g, 1)
+            res = {"oids": np.array([rank], dtype=np.int64),
+                   "values": np.array([0], dtype=np.int64)}
+        except RuntimeError as ex:
+            ok = "boom from rank 1" in str(ex)
+            res = {"oids": np.array([rank], dtype=np.int64),
+                   "values": np.array([1 if ok else -1], dtype=np.int64)}
     elif app == "kclique":
         res = eng.kclique(g, cfg.get("k", 3))
         res = dict(res, oids=np.array([0], dtype=np.int64),

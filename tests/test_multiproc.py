@@ -265,3 +265,14 @@ def test_serialize_roundtrip_mp_hashmap(world, free_port, tmp_path):
     oids, vals = run_world(world, cfg, free_port, tmp_path)
     src, dst, _ = graph_arrays(cfg)
     assert np.array_equal(vals, bfs_oracle(cfg["num_v"], src, dst, 3))
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_force_terminate_mp(world, free_port, tmp_path):
+    """Cooperative abort (reference ForceTerminate + TerminateInfo,
+    default_message_manager.h:156-166): rank 1 aborts in PEval; every
+    rank raises at the round boundary carrying rank 1's info string."""
+    cfg = dict(BASE, app="force_terminate")
+    oids, vals = run_world(world, cfg, free_port, tmp_path)
+    assert len(vals) == world
+    assert (vals == 1).all()  # 1 = raised with the right info on that rank
