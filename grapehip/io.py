@@ -11,26 +11,30 @@ import os
 import numpy as np
 
 
+def _line_start(f, pos: int, size: int) -> int:
+    """First line-start byte offset at or after pos."""
+    if pos <= 0:
+        return 0
+    if pos >= size:
+        return size
+    f.seek(pos - 1)
+    if f.read(1) != b"\n":
+        f.readline()  # skip the rest of the straddling line
+    return f.tell()
+
+
 def _byte_slice(path: str, rank: int, world: int):
+    # a rank owns the lines whose first byte falls in its raw byte range;
+    # aligning BOTH ends with the same rule gives no gaps, overlaps, or
+    # duplicates even when a slice is smaller than one line
     size = os.path.getsize(path)
-    lo = size * rank // world
-    hi = size * (rank + 1) // world
     with open(path, "rb") as f:
-        if lo > 0:
-            f.seek(lo - 1)
-            # advance to the start of the next full line
-            chunk = f.read(1)
-            if chunk != b"\n":
-                f.readline()
-            lo = f.tell()
+        lo = _line_start(f, size * rank // world, size)
+        hi = _line_start(f, size * (rank + 1) // world, size)
+        if lo >= hi:
+            return b""
         f.seek(lo)
-        data = f.read(hi - lo)
-        if hi < size and (not data or data[-1:] != b"\n"):
-            # finish the straddling line
-            with open(path, "rb") as g:
-                g.seek(lo + len(data))
-                data += g.readline()
-    return data
+        return f.read(hi - lo)
 
 
 def _ncols(path: str) -> int:

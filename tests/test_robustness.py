@@ -67,3 +67,34 @@ def test_unknown_oid_source_unreached(eng):
                        directed=False, vertex_oids=oids)
     r = eng.sssp(g, 77)
     assert (r["values"] == np.finfo(np.float64).max).all()
+
+
+def test_byte_slice_partition_exact(tmp_path):
+    """Every line lands on exactly one rank for any world size, including
+    worlds where a slice is smaller than one line (regression: a negative
+    read count returned the rest of the file, duplicating edges)."""
+    from grapehip.io import _byte_slice
+    lines = [f"{i} {i*7} 1.25\n" for i in range(17)]
+    p = tmp_path / "t.e"
+    p.write_text("".join(lines))
+    for world in (1, 2, 3, 5, 8, 16, 64):
+        got = []
+        for r in range(world):
+            got.append(_byte_slice(str(p), r, world).decode())
+        assert "".join(got) == "".join(lines), world
+        for g in got:  # each slice is whole lines
+            assert g == "" or g.endswith("\n")
+
+
+def test_read_ldbc_edges_many_ranks(tmp_path):
+    import grapehip.io as io
+    lines = [f"{i} {i+1} 0.5\n" for i in range(10)]
+    p = tmp_path / "w.e"
+    p.write_text("".join(lines))
+    seen = []
+    for r in range(32):
+        src, dst, w = io.read_ldbc_edges(str(p), weighted=True, rank=r,
+                                         world=32)
+        assert len(src) == len(dst) == len(w)
+        seen.extend(src.tolist())
+    assert sorted(seen) == list(range(10))
