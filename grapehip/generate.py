@@ -15,9 +15,8 @@ def uniform_edges(num_vertices: int, num_edges: int, seed: int = 42,
     """Uniform random directed edges; rank gets a contiguous slice."""
     lo = num_edges * rank // world
     hi = num_edges * (rank + 1) // world
-    rng = np.random.default_rng(seed)
-    # advance cheaply: draw per-slice with a per-rank child seed derived from
-    # (seed, slice) so ranks are independent but the union is deterministic.
+    # per-rank child seed derived from (seed, rank, world): ranks are
+    # independent but the union is deterministic for a fixed world size
     rng = np.random.default_rng([seed, rank, world])
     n = hi - lo
     src = rng.integers(0, num_vertices, size=n, dtype=np.int64)
