@@ -6,6 +6,8 @@ import sys
 
 
 def main():
+    if len(sys.argv) < 2:
+        sys.exit("usage: pmc_summary.py <results.db> [out.md]")
     db = sqlite3.connect(sys.argv[1])
     cur = db.cursor()
     tabs = [r[0] for r in cur.execute(

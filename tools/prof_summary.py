@@ -34,6 +34,8 @@ def summarize(db_path: str):
 
 
 def main():
+    if len(sys.argv) < 2:
+        sys.exit("usage: prof_summary.py <results.db> [out.md]")
     db_path = sys.argv[1]
     rows = summarize(db_path)
     total = sum(r[2] for r in rows) or 1

@@ -34,6 +34,8 @@ def load_file(p):
 
 
 def main():
+    if len(sys.argv) < 4:
+        sys.exit("usage: verify_output.py {exact,eps,wcc} <out_dir> <golden> [rtol]")
     mode, out_dir, golden = sys.argv[1], sys.argv[2], sys.argv[3]
     o1, v1 = load_dir(out_dir)
     o2, v2 = load_file(golden)
