@@ -55,9 +55,12 @@ class ThreadPool {
   void start(int n) {
     n_ = n;
     stop_ = false;
+    // workers start at the CURRENT epoch: a pool restarted after earlier
+    // runs (set_num_threads) must not treat a stale epoch as a pending
+    // task — a spurious wakeup would otherwise call a null fn_
     for (int t = 1; t < n_; ++t) {
-      workers_.emplace_back([this, t] {
-        uint64_t seen = 0;
+      workers_.emplace_back([this, t, e = epoch_] {
+        uint64_t seen = e;
         for (;;) {
           const std::function<void(int)>* fn;
           {

@@ -98,3 +98,18 @@ def test_read_ldbc_edges_many_ranks(tmp_path):
         assert len(src) == len(dst) == len(w)
         seen.extend(src.tolist())
     assert sorted(seen) == list(range(10))
+
+
+def test_engine_thread_count_restart(tmp_path):
+    """Recreating engines with different thread counts in one process
+    restarts the shared pool; work after a restart must be correct
+    (regression: restarted workers started at epoch 0 and could call a
+    null task on a spurious wakeup)."""
+    src = np.array([0, 1, 2, 3], dtype=np.int64)
+    dst = np.array([1, 2, 3, 0], dtype=np.int64)
+    for i, nt in enumerate((2, 4, 1, 3)):
+        eng2 = grapehip.Engine(rank=0, world=1, master_port=29770 + i,
+                               n_threads=nt)
+        g = eng2.load_edges(src, dst, directed=False, num_vertices=4)
+        r = eng2.bfs(g, 0)
+        assert sorted(r["values"].tolist()) == [0, 1, 1, 2]
