@@ -75,3 +75,17 @@ def test_truncated_checkpoint_raises(eng, tmp_path):
     p.write_bytes(data[: len(data) // 2])
     with pytest.raises(RuntimeError, match="short read"):
         eng.load_serialized(str(tmp_path))
+
+
+def test_roundtrip_mph_oids(eng, tmp_path):
+    src, dst, w = random_graph(num_v=400, num_e=2500, seed=79)
+    oids = np.arange(400, dtype=np.int64) * 11 + 5
+    g = eng.load_edges(oids[src], oids[dst], weights=w, directed=False,
+                       vertex_oids=oids, idxer="mph")
+    before = eng.sssp(g, oids[3])
+    eng.save_graph(g, str(tmp_path))
+    g2 = eng.load_serialized(str(tmp_path))
+    after = eng.sssp(g2, oids[3])
+    o1, o2 = np.argsort(before["oids"]), np.argsort(after["oids"])
+    assert np.array_equal(before["oids"][o1], after["oids"][o2])
+    assert np.array_equal(before["values"][o1], after["values"][o2])
