@@ -78,7 +78,7 @@ class KCliqueApp {
     send[frag.fid()].clear();
     auto recv = mm.comm()->exchange_all(send);
     for (int f = 0; f < fnum; ++f) {
-      if (f == frag.fid()) continue;
+      if (f == static_cast<int>(frag.fid())) continue;
       const char* p = recv[f].data();
       uint64_t n;
       std::memcpy(&n, p, 8);
