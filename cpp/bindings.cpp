@@ -163,8 +163,10 @@ std::shared_ptr<PyGraph> load_edges(
     if (idxer == "mph")
       g->vm->init_mph(eng.world, eng.c(), std::move(owned));
     else
-      g->vm->init_hashmap(eng.world, PartitionerKind::kHash, eng.c(),
-                          std::move(owned));
+      g->vm->init_hashmap(eng.world,
+                          partitioner == "map" ? PartitionerKind::kMap
+                                               : PartitionerKind::kHash,
+                          eng.c(), std::move(owned));
   } else {
     if (num_vertices <= 0)
       throw std::runtime_error("num_vertices required for identity mapping");
@@ -531,7 +533,8 @@ PYBIND11_MODULE(_core, m) {
            [](PyEngine& eng, const std::string& prefix) {
              auto g = std::make_shared<PyGraph>();
              py::gil_scoped_release rel;
-             auto [vm, frag] = deserialize_graph(prefix, eng.rank);
+             auto [vm, frag] = deserialize_graph(
+                 prefix, eng.rank, static_cast<uint32_t>(eng.world));
              g->vm = vm;
              g->frag = std::move(frag);
 #ifdef GRAPEHIP_WITH_HIP

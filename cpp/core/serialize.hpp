@@ -115,8 +115,13 @@ inline void serialize_graph(const Fragment& frag,
   }
 }
 
+// expected_fnum: pass the running world size; a checkpoint written at a
+// different world would silently route gids to fragments no rank serves
+// (reference semantics: one frag file per rank, frag_%d.s, config.h:67).
+// 0 skips the check (offline inspection tools).
 inline std::pair<std::shared_ptr<VertexMap>, std::unique_ptr<Fragment>>
-deserialize_graph(const std::string& prefix, fid_t fid) {
+deserialize_graph(const std::string& prefix, fid_t fid,
+                  uint32_t expected_fnum = 0) {
   ser::Reader r(ser::frag_path(prefix, fid));
   if (r.pod<uint64_t>() != ser::kMagic)
     throw std::runtime_error("deserialize: bad magic");
@@ -124,6 +129,10 @@ deserialize_graph(const std::string& prefix, fid_t fid) {
     throw std::runtime_error("deserialize: version mismatch");
   auto vm = std::make_shared<VertexMap>();
   uint32_t fnum = r.pod<uint32_t>();
+  if (expected_fnum && fnum != expected_fnum)
+    throw std::runtime_error(
+        "deserialize: checkpoint written at world " + std::to_string(fnum) +
+        " but engine runs world " + std::to_string(expected_fnum));
   IdxerKind idx = static_cast<IdxerKind>(r.pod<uint8_t>());
   PartitionerKind pk = static_cast<PartitionerKind>(r.pod<uint8_t>());
   uint64_t nv = r.pod<uint64_t>();
@@ -131,6 +140,11 @@ deserialize_graph(const std::string& prefix, fid_t fid) {
   r.vec(seg);
   if (idx == IdxerKind::kIdentity) {
     vm->init_identity(fnum, nv);
+    if (vm->segments() != seg)
+      throw std::runtime_error(
+          "deserialize: identity segments in checkpoint do not match the "
+          "uniform ceil-slices init_identity derives — non-uniform "
+          "segmented maps are not round-trippable");
   } else {
     std::vector<std::vector<oid_t>> oids(fnum);
     for (uint32_t f = 0; f < fnum; ++f) r.vec(oids[f]);
@@ -140,6 +154,10 @@ deserialize_graph(const std::string& prefix, fid_t fid) {
       vm->init_hashmap_local(fnum, pk, std::move(oids));
   }
   auto frag = Fragment::FromParts(vm, r, fnum);
+  if (frag->fid() != fid)
+    throw std::runtime_error(
+        "deserialize: file " + ser::frag_path(prefix, fid) +
+        " holds fragment " + std::to_string(frag->fid()));
   return {vm, std::move(frag)};
 }
 

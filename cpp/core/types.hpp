@@ -33,6 +33,10 @@ enum class LoadStrategy : uint8_t {
 enum class PartitionerKind : uint8_t {
   kHash = 0,      // owner = hash(oid) % fnum
   kSegmented = 1, // contiguous oid ranges (natural for synthetic/renumbered)
+  kMap = 2,       // explicit supplier-owned assignment (reference
+                  // MapPartitioner, partitioner.h:103): owner(oid) is not
+                  // computable from the oid alone — routing goes through
+                  // the replicated oid lists, never owner()
 };
 
 // Load-balancing strategy for GPU neighbor expansion (hip/gpu_engine.hip,

@@ -14,6 +14,7 @@
 #include <algorithm>
 #include <cstdint>
 #include <cstring>
+#include <stdexcept>
 #include <unordered_map>
 #include <vector>
 
@@ -112,6 +113,10 @@ class VertexMap {
       while (f + 1 < static_cast<fid_t>(fnum_) && o >= seg_[f + 1]) ++f;
       return f;
     }
+    if (pkind_ == PartitionerKind::kMap)
+      throw std::runtime_error(
+          "owner() undefined for map partitioning: ownership is whatever "
+          "rank supplied the oid — route through get_gid instead");
     return static_cast<fid_t>(hash_oid(oid) % fnum_);
   }
 

@@ -59,12 +59,14 @@ def read_ldbc_edges(efile: str, weighted: bool = False, rank: int = 0,
     ncols = _ncols(efile)
     if weighted and ncols < 3:
         raise ValueError("%s has no weight column" % efile)
-    fields = np.array(data.split(),
-                      dtype=np.float64 if ncols >= 3 else np.int64)
-    fields = fields.reshape(-1, ncols)
-    src = fields[:, 0].astype(np.int64)
-    dst = fields[:, 1].astype(np.int64)
-    w = fields[:, 2].astype(np.float32) if weighted else None
+    fields = data.split()
+    if len(fields) % ncols:
+        raise ValueError("%s: ragged row (column count varies)" % efile)
+    # parse id columns directly as int64 — going through float64 would
+    # silently corrupt oids above 2^53 (sparse LDBC id spaces)
+    src = np.array(fields[0::ncols], dtype=np.int64)
+    dst = np.array(fields[1::ncols], dtype=np.int64)
+    w = (np.array(fields[2::ncols], dtype=np.float32) if weighted else None)
     return src, dst, w
 
 
