@@ -1239,10 +1239,196 @@ __global__ void bfs_pull_large_kernel(const uint64_t* __restrict__ off,
 // GpuContext implementation
 // ===========================================================================
 
-struct GpuContext::Impl {
+// ===========================================================================
+// Device data plane. Production path: RCCL over xGMI, one rank per GPU
+// (reference gpu_message_manager.h NCCL ring, RCCL is the MI355X drop-in).
+// Validation path: TCP-staged collectives for when several ranks share one
+// device (RCCL refuses duplicate GPUs) — selected by GRAPEHIP_DATAPLANE=tcp
+// or auto-detected at init. Both paths drive the IDENTICAL call sites, so
+// the world-N halo/allgather/reduce-scatter logic is exercised end-to-end
+// on a one-GPU lease before an 8-GPU node ever sees it.
+// ===========================================================================
+struct DevComm {
   ncclComm_t nccl = nullptr;
+  TcpComm* tcp = nullptr;
+  int rank = 0, world = 1;
+  bool use_tcp = false;
+
+  bool active() const { return world > 1; }
+
+  // --- tcp staging helpers (host bounce; correctness path, not perf) ------
+  std::string stage_out(const void* dev, size_t n, hipStream_t s) const {
+    std::string h(n, '\0');
+    HIP_CHECK(hipStreamSynchronize(s));
+    if (n) HIP_CHECK(hipMemcpy(h.data(), dev, n, hipMemcpyDeviceToHost));
+    return h;
+  }
+  void stage_in(void* dev, const void* host, size_t n) const {
+    if (n) HIP_CHECK(hipMemcpy(dev, host, n, hipMemcpyHostToDevice));
+  }
+
+  // In-place allgather of 4-byte slices: buf[rank*slice .. ) is this rank's
+  // contribution; on return buf[0 .. world*slice) holds every rank's.
+  void allgather32(void* buf, uint64_t slice, bool fp, hipStream_t s) {
+    if (!active()) return;
+    if (!use_tcp) {
+      NCCL_CHECK(ncclAllGather(
+          static_cast<char*>(buf) + static_cast<uint64_t>(rank) * slice * 4,
+          buf, slice, fp ? ncclFloat : ncclUint32, nccl, s));
+      return;
+    }
+    std::string mine = stage_out(
+        static_cast<char*>(buf) + static_cast<uint64_t>(rank) * slice * 4,
+        slice * 4, s);
+    std::vector<char> all(static_cast<size_t>(world) * slice * 4);
+    tcp->allgather(mine.data(), slice * 4, all.data());
+    stage_in(buf, all.data(), all.size());
+  }
+
+  // In-place reduce-scatter (sum, f64): buf holds world*slice doubles; on
+  // return buf[rank*slice ..) = elementwise sum of every rank's slice.
+  void reduce_scatter_sum_f64(double* buf, uint64_t slice, hipStream_t s) {
+    if (!active()) return;
+    if (!use_tcp) {
+      NCCL_CHECK(ncclReduceScatter(
+          buf, buf + static_cast<uint64_t>(rank) * slice, slice, ncclDouble,
+          ncclSum, nccl, s));
+      return;
+    }
+    std::string full =
+        stage_out(buf, static_cast<size_t>(world) * slice * 8, s);
+    const double* h = reinterpret_cast<const double*>(full.data());
+    std::vector<std::string> send(world);
+    for (int p = 0; p < world; ++p)
+      send[p].assign(
+          reinterpret_cast<const char*>(h + static_cast<uint64_t>(p) * slice),
+          slice * 8);
+    auto recv = tcp->exchange_all(send);
+    std::vector<double> acc(slice, 0.0);
+    for (int p = 0; p < world; ++p) {
+      const double* rp = reinterpret_cast<const double*>(recv[p].data());
+      for (uint64_t i = 0; i < slice; ++i) acc[i] += rp[i];
+    }
+    stage_in(buf + static_cast<uint64_t>(rank) * slice, acc.data(),
+             slice * 8);
+  }
+
+  void allreduce_min_u32(uint32_t* buf, uint64_t n, hipStream_t s) {
+    if (!active()) return;
+    if (!use_tcp) {
+      NCCL_CHECK(
+          ncclAllReduce(buf, buf, n, ncclUint32, ncclMin, nccl, s));
+      return;
+    }
+    std::string mine = stage_out(buf, n * 4, s);
+    std::vector<uint32_t> all(static_cast<size_t>(world) * n);
+    tcp->allgather(mine.data(), n * 4, all.data());
+    uint32_t* acc = reinterpret_cast<uint32_t*>(mine.data());
+    for (int p = 0; p < world; ++p) {
+      const uint32_t* rp = all.data() + static_cast<uint64_t>(p) * n;
+      for (uint64_t i = 0; i < n; ++i)
+        if (rp[i] < acc[i]) acc[i] = rp[i];
+    }
+    stage_in(buf, acc, n * 4);
+  }
+
+  void allreduce_sum_u64(unsigned long long* buf, uint64_t n,
+                         hipStream_t s) {
+    if (!active()) return;
+    if (!use_tcp) {
+      NCCL_CHECK(ncclAllReduce(buf, buf, n, ncclUint64, ncclSum, nccl, s));
+      return;
+    }
+    std::string mine = stage_out(buf, n * 8, s);
+    std::vector<unsigned long long> all(static_cast<size_t>(world) * n);
+    tcp->allgather(mine.data(), n * 8, all.data());
+    unsigned long long* acc =
+        reinterpret_cast<unsigned long long*>(mine.data());
+    for (uint64_t i = 0; i < n; ++i) acc[i] = 0;
+    for (int p = 0; p < world; ++p) {
+      const unsigned long long* rp =
+          all.data() + static_cast<uint64_t>(p) * n;
+      for (uint64_t i = 0; i < n; ++i) acc[i] += rp[i];
+    }
+    stage_in(buf, acc, n * 8);
+  }
+
+  // Pairwise variable-size exchange: send_off/recv_off are [world+1] byte
+  // offsets into sendbuf/recvbuf; peer p receives this rank's region p.
+  // Self region is handled by the caller (halo appends skip owned ids).
+  void sendrecv(const uint8_t* sendbuf, const std::vector<uint64_t>& send_off,
+                uint8_t* recvbuf, const std::vector<uint64_t>& recv_off,
+                hipStream_t s) {
+    if (!active()) return;
+    if (!use_tcp) {
+      NCCL_CHECK(ncclGroupStart());
+      for (int p = 0; p < world; ++p) {
+        if (p == rank) continue;
+        uint64_t to_p = send_off[p + 1] - send_off[p];
+        uint64_t from_p = recv_off[p + 1] - recv_off[p];
+        if (to_p)
+          NCCL_CHECK(ncclSend(sendbuf + send_off[p], to_p, ncclChar, p,
+                              nccl, s));
+        if (from_p)
+          NCCL_CHECK(ncclRecv(recvbuf + recv_off[p], from_p, ncclChar, p,
+                              nccl, s));
+      }
+      NCCL_CHECK(ncclGroupEnd());
+      return;
+    }
+    HIP_CHECK(hipStreamSynchronize(s));
+    std::vector<std::string> send(world);
+    for (int p = 0; p < world; ++p) {
+      if (p == rank) continue;
+      uint64_t n = send_off[p + 1] - send_off[p];
+      send[p] = stage_out(sendbuf + send_off[p], n, s);
+    }
+    auto recv = tcp->exchange_all(send);
+    for (int p = 0; p < world; ++p) {
+      if (p == rank) continue;
+      uint64_t n = recv_off[p + 1] - recv_off[p];
+      if (n != recv[p].size())
+        throw std::runtime_error("sendrecv(tcp): size mismatch");
+      stage_in(recvbuf + recv_off[p], recv[p].data(), n);
+    }
+  }
+
+  // Ragged in-place allgather: rank f owns buf[region[f]*4 .. region[f+1]*4)
+  // (u32 elements); every rank ends with all regions populated.
+  void bcast_regions_u32(uint32_t* buf, const std::vector<uint64_t>& region,
+                         hipStream_t s) {
+    if (!active()) return;
+    if (!use_tcp) {
+      NCCL_CHECK(ncclGroupStart());
+      for (int f = 0; f < world; ++f) {
+        uint64_t cnt_f = region[f + 1] - region[f];
+        if (cnt_f)
+          NCCL_CHECK(ncclBroadcast(buf + region[f], buf + region[f], cnt_f,
+                                   ncclUint32, f, nccl, s));
+      }
+      NCCL_CHECK(ncclGroupEnd());
+      return;
+    }
+    uint64_t mine_n = region[rank + 1] - region[rank];
+    std::string mine = stage_out(buf + region[rank], mine_n * 4, s);
+    std::vector<std::string> send(world, mine);
+    auto recv = tcp->exchange_all(send);
+    for (int f = 0; f < world; ++f) {
+      if (f == rank) continue;
+      uint64_t n = region[f + 1] - region[f];
+      if (n * 4 != recv[f].size())
+        throw std::runtime_error("bcast_regions(tcp): size mismatch");
+      stage_in(buf + region[f], recv[f].data(), n * 4);
+    }
+  }
+};
+
+struct GpuContext::Impl {
+  DevComm dc;
   Stream compute;
   Stream comm_stream;
+  Event ev_pack;   // compute -> comm ordering for the halo exchange
+  Event ev_comm;   // comm -> compute ordering (recv payload ready)
   ScanTemp scan;
   // halo scratch (sized on first use)
   DeviceBuffer<uint32_t> halo_idx;
@@ -1307,21 +1493,45 @@ GpuContext::GpuContext(TcpComm* comm, int rank, int world)
   dev_ = lr ? (std::atoi(lr) % ndev) : (rank % ndev);
   HIP_CHECK(hipSetDevice(dev_));
   impl_ = std::make_unique<Impl>();
+  DevComm& dc = impl_->dc;
+  dc.tcp = comm;
+  dc.rank = rank;
+  dc.world = world;
   if (world > 1) {
-    ncclUniqueId id;
-    if (rank == 0) NCCL_CHECK(ncclGetUniqueId(&id));
-    comm->bcast(&id, sizeof(id), 0);
-    NCCL_CHECK(ncclCommInitRank(&impl_->nccl, world, id, rank));
-    // warm up the xGMI links (reference WarmupNccl, dev_utils.h:894)
-    DeviceBuffer<float> dummy(world * 256);
-    NCCL_CHECK(ncclAllReduce(dummy.data(), dummy.data(), 256, ncclFloat,
-                             ncclSum, impl_->nccl, impl_->compute));
-    impl_->compute.sync();
+    // data-plane pick: RCCL needs one distinct GPU per rank; ranks that
+    // share a device (world-N validation on a 1-GPU lease) stage through
+    // the TCP control plane instead. Same call sites either way.
+    const char* dp = std::getenv("GRAPEHIP_DATAPLANE");
+    if (dp && std::string(dp) == "tcp") {
+      dc.use_tcp = true;
+    } else if (dp && std::string(dp) == "rccl") {
+      dc.use_tcp = false;
+    } else {
+      std::vector<int> devs(world);
+      comm->allgather(&dev_, sizeof(int), devs.data());
+      std::sort(devs.begin(), devs.end());
+      dc.use_tcp = std::adjacent_find(devs.begin(), devs.end()) !=
+                   devs.end();
+    }
+    if (!dc.use_tcp) {
+      ncclUniqueId id;
+      if (rank == 0) NCCL_CHECK(ncclGetUniqueId(&id));
+      comm->bcast(&id, sizeof(id), 0);
+      NCCL_CHECK(ncclCommInitRank(&dc.nccl, world, id, rank));
+      // warm up the xGMI links (reference WarmupNccl, dev_utils.h:894)
+      DeviceBuffer<float> dummy(world * 256);
+      NCCL_CHECK(ncclAllReduce(dummy.data(), dummy.data(), 256, ncclFloat,
+                               ncclSum, dc.nccl, impl_->compute));
+      impl_->compute.sync();
+    } else if (getenv("GRAPEHIP_DEBUG")) {
+      fprintf(stderr, "[dc] rank %d: TCP data plane (shared device)\n",
+              rank);
+    }
   }
 }
 
 GpuContext::~GpuContext() {
-  if (impl_ && impl_->nccl) ncclCommDestroy(impl_->nccl);
+  if (impl_ && impl_->dc.nccl) ncclCommDestroy(impl_->dc.nccl);
 }
 
 void GpuContext::device_sync() { HIP_CHECK(hipDeviceSynchronize()); }
@@ -1675,22 +1885,24 @@ uint64_t halo_flush(GpuContext::Impl& I, TcpComm* comm, int rank, int world,
   std::vector<unsigned long long> cnts(world);
   HIP_CHECK(hipMemcpyAsync(cnts.data(), I.halo_cnt.data(), world * 8,
                            hipMemcpyDeviceToHost, s));
-  HIP_CHECK(hipStreamSynchronize(s));
+  HIP_CHECK(hipStreamSynchronize(s));  // the one inherent host sync: the
+                                       // pack launch shapes need the counts
   uint64_t send_total = 0;
   for (int p = 0; p < world; ++p) send_total += cnts[p];
   size_t pair_sz = sizeof(HaloPair<T>);
   if (I.sendbuf.size() < send_total * pair_sz)
     I.sendbuf.resize(send_total * pair_sz + (1 << 20));
-  // pack contiguously in peer order
+  // pack contiguously in peer order (async: runs while the TCP count
+  // matrix goes around on the host below)
   std::vector<uint64_t> send_off(world + 1, 0);
-  for (int p = 0; p < world; ++p) send_off[p + 1] = send_off[p] + cnts[p];
+  for (int p = 0; p < world; ++p)
+    send_off[p + 1] = send_off[p] + cnts[p] * pair_sz;
   for (int p = 0; p < world; ++p) {
     if (!cnts[p]) continue;
     halo_pack_kernel<T><<<grid_for(cnts[p]), kBlock, 0, s>>>(
         I.halo_idx.data() + static_cast<uint64_t>(p) * cap, cnts[p], state,
         DevBitmap{I.halo_bm.data()},
-        reinterpret_cast<HaloPair<T>*>(I.sendbuf.data() +
-                                       send_off[p] * pair_sz));
+        reinterpret_cast<HaloPair<T>*>(I.sendbuf.data() + send_off[p]));
   }
   I.halo_cnt.zero(s);
   // exchange the count matrix on the control plane
@@ -1701,25 +1913,20 @@ uint64_t halo_flush(GpuContext::Impl& I, TcpComm* comm, int rank, int world,
   std::vector<uint64_t> recv_off(world + 1, 0);
   for (int p = 0; p < world; ++p) {
     uint64_t from_p = matrix[static_cast<size_t>(p) * world + rank];
-    recv_off[p + 1] = recv_off[p] + from_p;
+    recv_off[p + 1] = recv_off[p] + from_p * pair_sz;
     recv_total += from_p;
   }
   if (I.recvbuf.size() < recv_total * pair_sz)
     I.recvbuf.resize(recv_total * pair_sz + (1 << 20));
-  HIP_CHECK(hipStreamSynchronize(s));  // packs done before nccl
-  NCCL_CHECK(ncclGroupStart());
-  for (int p = 0; p < world; ++p) {
-    if (p == rank) continue;
-    uint64_t to_p = cnts[p];
-    uint64_t from_p = recv_off[p + 1] - recv_off[p];
-    if (to_p)
-      NCCL_CHECK(ncclSend(I.sendbuf.data() + send_off[p] * pair_sz,
-                          to_p * pair_sz, ncclChar, p, I.nccl, s));
-    if (from_p)
-      NCCL_CHECK(ncclRecv(I.recvbuf.data() + recv_off[p] * pair_sz,
-                          from_p * pair_sz, ncclChar, p, I.nccl, s));
-  }
-  NCCL_CHECK(ncclGroupEnd());
+  // payloads over xGMI on the comm stream, event-ordered after the packs
+  // (reference gpu_message_manager.h:401-445 dual-stream design); no host
+  // sync — the caller's halo_process launch on `s` waits on ev_comm.
+  I.ev_pack.record(s);
+  I.ev_pack.wait_on(I.comm_stream);
+  I.dc.sendrecv(I.sendbuf.data(), send_off, I.recvbuf.data(), recv_off,
+                I.comm_stream);
+  I.ev_comm.record(I.comm_stream);
+  I.ev_comm.wait_on(s);
   // self pairs (should be none — owned handled locally)
   return recv_total;
 }
@@ -1822,10 +2029,7 @@ GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source,
       use_pull = gedges > pull_edge_threshold;
     }
     if (use_pull) {
-      if (multi)
-        NCCL_CHECK(ncclAllGather(
-            depth.data() + static_cast<uint64_t>(rank_) * slice, depth.data(),
-            slice, ncclUint32, I.nccl, s));
+      if (multi) I.dc.allgather32(depth.data(), slice, false, s);
       DevBitmap nb{next_bm.data()};
       if (g.n_small)
         bfs_pull_small_kernel<<<grid_for(g.n_small), kBlock, 0, s>>>(
@@ -2136,11 +2340,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
         rank_arr.data(), g.oe_off.data(), g.owned_real, g.v_begin,
         contrib.data() + g.v_begin);
     if (pull) {
-      if (multi)
-        NCCL_CHECK(ncclAllGather(contrib.data() + static_cast<uint64_t>(
-                                                      rank_) * slice,
-                                 contrib.data(), slice, ncclFloat, I.nccl,
-                                 s));
+      if (multi) I.dc.allgather32(contrib.data(), slice, true, s);
       if (g.n_small)
         pr_pull_small_kernel<<<grid_for(g.n_small), kBlock, 0, s>>>(
             pull_off, pull_dst, contrib.data(), g.rows_small.data(),
@@ -2161,10 +2361,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
           <<<std::min(nchunks, kMaxGrid), kBlock, 0, s>>>(
               view, PrPushOp{contrib.data() + g.v_begin, acc.data(),
                              g.v_begin});
-      if (multi)
-        NCCL_CHECK(ncclReduceScatter(
-            acc.data(), acc.data() + static_cast<uint64_t>(rank_) * slice,
-            slice, ncclDouble, ncclSum, I.nccl, s));
+      if (multi) I.dc.reduce_scatter_sum_f64(acc.data(), slice, s);
     }
     if (tol > 0) d_l1.zero(s);
     pr_apply_kernel<<<grid_for(owned), kBlock, 0, s>>>(
@@ -2330,8 +2527,7 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
     }
     if (!multi) break;
     bool any = comm_->allreduce_or(local_changed_any != 0);
-    NCCL_CHECK(ncclAllReduce(parent.data(), parent.data(), nv_pad,
-                             ncclUint32, ncclMin, I.nccl, s));
+    I.dc.allreduce_min_u32(parent.data(), nv_pad, s);
     wcc_compress_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(parent.data(),
                                                             nv_pad);
     HIP_CHECK(hipStreamSynchronize(s));
@@ -2993,10 +3189,7 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
                              s));
     HIP_CHECK(hipStreamSynchronize(s));
     uint64_t g_nch = multi ? comm_->allreduce_sum(nch) : nch;
-    if (multi)
-      NCCL_CHECK(ncclAllGather(
-          lab.data() + static_cast<uint64_t>(rank_) * slice, lab.data(),
-          slice, ncclUint32, I.nccl, s));
+    if (multi) I.dc.allgather32(lab.data(), slice, false, s);
     if (g_nch == 0) break;  // fixpoint: further rounds are no-ops
     if (it + 1 == iters) break;
     // adaptive: building the dirty set costs an edge expansion over the
@@ -4256,10 +4449,9 @@ struct GlobalDedupCsr {
   uint64_t total = 0;
 };
 
-void build_global_csr(GpuContext::Impl& I, ncclComm_t nccl, int rank,
-                      int world, uint32_t nv_pad, uint32_t slice,
-                      uint32_t owned, uint32_t v_begin,
-                      const DeviceBuffer<uint64_t>& ooff,
+void build_global_csr(GpuContext::Impl& I, int rank, int world,
+                      uint32_t nv_pad, uint32_t slice, uint32_t owned,
+                      uint32_t v_begin, const DeviceBuffer<uint64_t>& ooff,
                       const DeviceBuffer<uint32_t>& ocnt,
                       const DeviceBuffer<uint32_t>& oadj,
                       GlobalDedupCsr& out, hipStream_t s) {
@@ -4268,10 +4460,7 @@ void build_global_csr(GpuContext::Impl& I, ncclComm_t nccl, int rank,
   out.gcnt.zero(s);
   HIP_CHECK(hipMemcpyAsync(out.gcnt.data() + v_begin, ocnt.data(),
                            owned * 4, hipMemcpyDeviceToDevice, s));
-  if (multi)
-    NCCL_CHECK(ncclAllGather(
-        out.gcnt.data() + static_cast<uint64_t>(rank) * slice,
-        out.gcnt.data(), slice, ncclUint32, nccl, s));
+  if (multi) I.dc.allgather32(out.gcnt.data(), slice, false, s);
   out.goff.resize(static_cast<size_t>(nv_pad) + 1);
   out.total =
       exclusive_scan(out.gcnt.data(), out.goff.data(), nv_pad, s, I.scan);
@@ -4290,15 +4479,7 @@ void build_global_csr(GpuContext::Impl& I, ncclComm_t nccl, int rank,
                                hipMemcpyDeviceToHost, s));
     }
     HIP_CHECK(hipStreamSynchronize(s));
-    NCCL_CHECK(ncclGroupStart());
-    for (int f = 0; f < world; ++f) {
-      uint64_t cnt_f = region[f + 1] - region[f];
-      if (cnt_f)
-        NCCL_CHECK(ncclBroadcast(out.gdst.data() + region[f],
-                                 out.gdst.data() + region[f], cnt_f,
-                                 ncclUint32, f, nccl, s));
-    }
-    NCCL_CHECK(ncclGroupEnd());
+    I.dc.bcast_regions_u32(out.gdst.data(), region, s);
   }
 }
 
@@ -4399,10 +4580,7 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
                                           heavy_off.data(), tbl_lab.data(),
                                           g.v_begin, Dv.data());
     }
-    if (multi)
-      NCCL_CHECK(ncclAllGather(
-          Dv.data() + static_cast<uint64_t>(rank_) * slice, Dv.data(),
-          slice, ncclUint32, I.nccl, s));
+    if (multi) I.dc.allgather32(Dv.data(), slice, false, s);
 
     // family builder: run the dedup pass over (o1,d1,o2,d2) with given
     // capacities, then lift to a replicated global CSR
@@ -4441,8 +4619,8 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
                                           Dv.data(), ooff.data(),
                                           oadj.data(), ocnt.data(), false);
       }
-      build_global_csr(I, I.nccl, rank_, world_, nv_pad, slice, owned,
-                       g.v_begin, ooff, ocnt, oadj, out, s);
+      build_global_csr(I, rank_, world_, nv_pad, slice, owned, g.v_begin,
+                       ooff, ocnt, oadj, out, s);
     };
 
     // U family: in ∪ out, cap = D
@@ -4621,10 +4799,7 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
                                        n_large, heavy_off.data(),
                                        tbl_lab.data(), g.v_begin, Dv.data());
     }
-    if (multi)
-      NCCL_CHECK(ncclAllGather(
-          Dv.data() + static_cast<uint64_t>(rank_) * slice, Dv.data(), slice,
-          ncclUint32, I.nccl, s));
+    if (multi) I.dc.allgather32(Dv.data(), slice, false, s);
 
     // pass 2: oriented adjacency in capacity layout (cap = D per row)
     DeviceBuffer<uint32_t> dcap(owned ? owned : 1);
@@ -4664,10 +4839,7 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
     gcnt.zero(s);
     HIP_CHECK(hipMemcpyAsync(gcnt.data() + g.v_begin, ocnt.data(),
                              owned * 4, hipMemcpyDeviceToDevice, s));
-    if (multi)
-      NCCL_CHECK(ncclAllGather(
-          gcnt.data() + static_cast<uint64_t>(rank_) * slice, gcnt.data(),
-          slice, ncclUint32, I.nccl, s));
+    if (multi) I.dc.allgather32(gcnt.data(), slice, false, s);
     goff.resize(static_cast<size_t>(nv_pad) + 1);
     uint64_t g_total =
         exclusive_scan(gcnt.data(), goff.data(), nv_pad, s, I.scan);
@@ -4725,15 +4897,7 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
         HIP_CHECK(hipStreamSynchronize(s));
         region[f] = off_v;
       }
-      NCCL_CHECK(ncclGroupStart());
-      for (int f = 0; f < world_; ++f) {
-        uint64_t cnt_f = region[f + 1] - region[f];
-        if (cnt_f)
-          NCCL_CHECK(ncclBroadcast(gdst.data() + region[f],
-                                   gdst.data() + region[f], cnt_f,
-                                   ncclUint32, f, I.nccl, s));
-      }
-      NCCL_CHECK(ncclGroupEnd());
+      I.dc.bcast_regions_u32(gdst.data(), region, s);
     }
   }
 
@@ -4809,9 +4973,7 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
           goff.data(), gdst.data(), hoff.data(), htab.data(),
           heavy_q.data(), hn, Tcnt.data());
   }
-  if (multi)
-    NCCL_CHECK(ncclAllReduce(Tcnt.data(), Tcnt.data(), nv_pad, ncclUint64,
-                             ncclSum, I.nccl, s));
+  if (multi) I.dc.allreduce_sum_u64(Tcnt.data(), nv_pad, s);
   DeviceBuffer<double> lcc_out(owned ? owned : 1);
   if (owned)
     lcc_finalize_kernel<<<grid_for(owned), kBlock, 0, s>>>(

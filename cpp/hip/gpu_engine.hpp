@@ -40,6 +40,17 @@ struct DeviceGraph {
   DeviceBuffer<uint32_t> rows_small, rows_mid, rows_large;
   uint64_t n_small = 0, n_mid = 0, n_large = 0;
   bool buckets_built = false;
+  // mirror topology (multi-GPU, built lazily): the reference's mirror-info /
+  // BatchShuffle design (edgecut_fragment_base.h:569+, cuda batch_shuffle
+  // :81-103) recast for xGMI — per-peer sorted lists of the REMOTE vertices
+  // this rank's edges actually reference (recv side) and of the OWNED
+  // vertices each peer references (send side). Dense per-round refreshes
+  // then ship referenced values point-to-point on distinct xGMI links
+  // instead of allgathering whole slices around the ring.
+  DeviceBuffer<uint32_t> mr_recv_idx, mr_send_idx;   // concatenated, global ids
+  std::vector<uint64_t> mr_recv_off, mr_send_off;    // [world+1] element offs
+  DeviceBuffer<uint8_t> mr_sendbuf, mr_recvbuf;      // 4B-element staging
+  bool mirrors_built = false;
   uint32_t owned() const { return v_end - v_begin; }
   // cached hipGraph of one PageRank iteration (single-GPU fixed-iter
   // path) + the working set it references: the capture bakes device
