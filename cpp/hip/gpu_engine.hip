@@ -1075,31 +1075,73 @@ __global__ void bucket_rows_kernel(const uint64_t* __restrict__ off,
   }
 }
 
+// Phase-split pulls (multi-GPU compute/comm overlap): the per-row
+// adjacency is sorted ascending, so the owned dst range [lo,hi) is one
+// contiguous span found by two binary searches. PHASE 0 = whole row
+// (single-GPU / no overlap); PHASE 1 = owned span only, overwrite acc
+// (runs WHILE the mirror exchange moves remote contribs on the comm
+// stream); PHASE 2 = remaining spans, accumulate into acc.
+__device__ __forceinline__ uint64_t lb_u32(const uint32_t* __restrict__ a,
+                                           uint64_t lo, uint64_t hi,
+                                           uint32_t key) {
+  while (lo < hi) {
+    uint64_t m = (lo + hi) >> 1;
+    if (a[m] < key) lo = m + 1;
+    else hi = m;
+  }
+  return lo;
+}
+
+// maps a phase to an iteration domain over row [b,e): phase 1 iterates
+// [L,R); phase 2 iterates [b,L) ++ [R,e) via a folded index.
+struct PrSpan {
+  uint64_t base1, n1, base2, n2;
+  __device__ __forceinline__ uint64_t map(uint64_t k) const {
+    return k < n1 ? base1 + k : base2 + (k - n1);
+  }
+  __device__ __forceinline__ uint64_t n() const { return n1 + n2; }
+};
+
+template <int PHASE>
+__device__ __forceinline__ PrSpan pr_span(const uint32_t* __restrict__ dst,
+                                          uint64_t b, uint64_t e,
+                                          uint32_t lo, uint32_t hi) {
+  if (PHASE == 0) return {b, e - b, 0, 0};
+  uint64_t L = lb_u32(dst, b, e, lo), R = lb_u32(dst, b, e, hi);
+  if (PHASE == 1) return {L, R - L, 0, 0};
+  return {b, L - b, R, e - R};
+}
+
 // thread per row
+template <int PHASE>
 __global__ void pr_pull_small_kernel(const uint64_t* __restrict__ off,
                                      const uint32_t* __restrict__ dst,
                                      const float* __restrict__ contrib,
                                      const uint32_t* __restrict__ rows,
                                      uint64_t nrows, uint32_t v_begin,
+                                     uint32_t lo, uint32_t hi,
                                      double* __restrict__ acc) {
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
   for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * blockDim.x +
                     threadIdx.x;
        i < nrows; i += stride) {
     uint32_t r = rows[i];
-    uint64_t b = off[r], e = off[r + 1];
+    PrSpan sp = pr_span<PHASE>(dst, off[r], off[r + 1], lo, hi);
     double sum = 0;
-    for (uint64_t k = b; k < e; ++k) sum += contrib[dst[k]];
-    acc[v_begin + r] = sum;
+    for (uint64_t k = 0; k < sp.n(); ++k) sum += contrib[dst[sp.map(k)]];
+    if (PHASE == 2) acc[v_begin + r] += sum;
+    else acc[v_begin + r] = sum;
   }
 }
 
 // wave per row (4 waves per 256-block)
+template <int PHASE>
 __global__ void pr_pull_mid_kernel(const uint64_t* __restrict__ off,
                                    const uint32_t* __restrict__ dst,
                                    const float* __restrict__ contrib,
                                    const uint32_t* __restrict__ rows,
                                    uint64_t nrows, uint32_t v_begin,
+                                   uint32_t lo, uint32_t hi,
                                    double* __restrict__ acc) {
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -1108,29 +1150,35 @@ __global__ void pr_pull_mid_kernel(const uint64_t* __restrict__ off,
   for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * waves_per_block + wid;
        i < nrows; i += wstride) {
     uint32_t r = rows[i];
-    uint64_t b = off[r], e = off[r + 1];
+    PrSpan sp = pr_span<PHASE>(dst, off[r], off[r + 1], lo, hi);
     double sum = 0;
-    for (uint64_t k = b + lane; k < e; k += kWave) sum += contrib[dst[k]];
+    for (uint64_t k = lane; k < sp.n(); k += kWave)
+      sum += contrib[dst[sp.map(k)]];
 #pragma unroll
     for (int d = 32; d > 0; d >>= 1) sum += __shfl_down(sum, d, 64);
-    if (lane == 0) acc[v_begin + r] = sum;
+    if (lane == 0) {
+      if (PHASE == 2) acc[v_begin + r] += sum;
+      else acc[v_begin + r] = sum;
+    }
   }
 }
 
 // block per row
+template <int PHASE>
 __global__ void pr_pull_large_kernel(const uint64_t* __restrict__ off,
                                      const uint32_t* __restrict__ dst,
                                      const float* __restrict__ contrib,
                                      const uint32_t* __restrict__ rows,
                                      uint64_t nrows, uint32_t v_begin,
+                                     uint32_t lo, uint32_t hi,
                                      double* __restrict__ acc) {
   __shared__ double s_wave[kBlock / kWave];
   for (uint64_t i = blockIdx.x; i < nrows; i += gridDim.x) {
     uint32_t r = rows[i];
-    uint64_t b = off[r], e = off[r + 1];
+    PrSpan sp = pr_span<PHASE>(dst, off[r], off[r + 1], lo, hi);
     double sum = 0;
-    for (uint64_t k = b + threadIdx.x; k < e; k += blockDim.x)
-      sum += contrib[dst[k]];
+    for (uint64_t k = threadIdx.x; k < sp.n(); k += blockDim.x)
+      sum += contrib[dst[sp.map(k)]];
 #pragma unroll
     for (int d = 32; d > 0; d >>= 1) sum += __shfl_down(sum, d, 64);
     if ((threadIdx.x & 63) == 0) s_wave[threadIdx.x >> 6] = sum;
@@ -1139,7 +1187,8 @@ __global__ void pr_pull_large_kernel(const uint64_t* __restrict__ off,
       double t = 0;
 #pragma unroll
       for (int w = 0; w < kBlock / kWave; ++w) t += s_wave[w];
-      acc[v_begin + r] = t;
+      if (PHASE == 2) acc[v_begin + r] += t;
+      else acc[v_begin + r] = t;
     }
     __syncthreads();
   }
@@ -1429,6 +1478,12 @@ struct GpuContext::Impl {
   Stream comm_stream;
   Event ev_pack;   // compute -> comm ordering for the halo exchange
   Event ev_comm;   // comm -> compute ordering (recv payload ready)
+  Event ev_scal;   // pinned-scalar D2H landed (host folds while GPU works)
+  double* h_scal = nullptr;  // 2 pinned doubles: [0] local out, [1] global in
+  Impl() { HIP_CHECK(hipHostMalloc(&h_scal, 2 * sizeof(double))); }
+  ~Impl() {
+    if (h_scal) (void)hipHostFree(h_scal);
+  }
   ScanTemp scan;
   // halo scratch (sized on first use)
   DeviceBuffer<uint32_t> halo_idx;
@@ -1931,6 +1986,164 @@ uint64_t halo_flush(GpuContext::Impl& I, TcpComm* comm, int rank, int world,
   return recv_total;
 }
 
+// ===========================================================================
+// Mirror topology: per-peer lists of the remote vertices this rank's edges
+// reference (reference mirror-info + BatchShuffle, edgecut_fragment_base.h
+// :569+ / cuda batch_shuffle_message_manager.h:81-103). Built once per
+// graph; per-round dense refreshes then ship ONLY referenced values,
+// point-to-point — each peer pair on its own xGMI link — instead of
+// allgathering whole slices around the ring (O(V·world) volume and
+// per-link serialization). 4-byte element specialization (depth u32,
+// PR contrib f32, CDLP label u32).
+// ===========================================================================
+
+__global__ void mark_dsts_kernel(const uint32_t* __restrict__ dst,
+                                 uint64_t n, uint32_t v_begin,
+                                 uint32_t v_end, DevBitmap bm) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * blockDim.x +
+                    threadIdx.x;
+       i < n; i += stride) {
+    uint32_t v = dst[i];
+    if (v < v_begin || v >= v_end) (void)bm.set_once(v);
+  }
+}
+
+// out[t] = lower_bound(idx[0..n), t*slice) for t in [0, world]
+__global__ void slice_bounds_kernel(const uint32_t* __restrict__ idx,
+                                    uint64_t n, uint32_t slice, int world,
+                                    uint64_t* __restrict__ out) {
+  int t = threadIdx.x;
+  if (t > world) return;
+  uint64_t key = static_cast<uint64_t>(t) * slice;
+  uint64_t lo = 0, hi = n;
+  while (lo < hi) {
+    uint64_t m = (lo + hi) >> 1;
+    if (idx[m] < key) lo = m + 1;
+    else hi = m;
+  }
+  out[t] = lo;
+}
+
+__global__ void gather4_kernel(const uint32_t* __restrict__ idx, uint64_t n,
+                               const uint32_t* __restrict__ state,
+                               uint32_t* __restrict__ out) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * blockDim.x +
+                    threadIdx.x;
+       i < n; i += stride)
+    out[i] = state[idx[i]];
+}
+
+__global__ void scatter4_kernel(const uint32_t* __restrict__ idx, uint64_t n,
+                                const uint32_t* __restrict__ in,
+                                uint32_t* __restrict__ state) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * blockDim.x +
+                    threadIdx.x;
+       i < n; i += stride)
+    state[idx[i]] = in[i];
+}
+
+static void ensure_mirrors(GpuContext::Impl& I, TcpComm* comm,
+                           DeviceGraph& g, int rank, int world,
+                           hipStream_t s) {
+  if (g.mirrors_built || world <= 1) return;
+  uint32_t nv_pad = padded_nv(g, world);
+  uint32_t slice = g.seg_host.size() > 1 ? g.seg_host[1] - g.seg_host[0]
+                                         : g.nv_global;
+  size_t nwords = (static_cast<size_t>(nv_pad) + 31) / 32;
+  DeviceBuffer<uint32_t> bm(nwords);
+  bm.zero(s);
+  if (g.local_edges)
+    mark_dsts_kernel<<<grid_for(g.local_edges), kBlock, 0, s>>>(
+        g.oe_dst.data(), g.local_edges, g.v_begin, g.v_end, DevBitmap{bm.data()});
+  if (g.has_in && g.ie_dst.size())
+    mark_dsts_kernel<<<grid_for(g.ie_dst.size()), kBlock, 0, s>>>(
+        g.ie_dst.data(), g.ie_dst.size(), g.v_begin, g.v_end,
+        DevBitmap{bm.data()});
+  // compact ascending (popc + scan + fill, cleared as it fills)
+  if (I.frontier_deg.size() < nwords)
+    I.frontier_deg.resize(nwords + (nwords >> 2) + 64);
+  if (I.frontier_off.size() < nwords + 1)
+    I.frontier_off.resize(nwords + (nwords >> 2) + 65);
+  popc_words_kernel<<<grid_for(nwords), kBlock, 0, s>>>(bm.data(), nwords,
+                                                        I.frontier_deg.data());
+  uint64_t total = exclusive_scan(I.frontier_deg.data(),
+                                  I.frontier_off.data(), nwords, s, I.scan);
+  g.mr_recv_idx.resize(total ? total : 1);
+  if (total)
+    fill_frontier_kernel<<<grid_for(nwords), kBlock, 0, s>>>(
+        bm.data(), nwords, I.frontier_off.data(), 0, g.mr_recv_idx.data());
+  // per-peer boundaries (vid-ascending list, slice-contiguous regions)
+  DeviceBuffer<uint64_t> d_bounds(world + 1);
+  slice_bounds_kernel<<<1, world + 1, 0, s>>>(g.mr_recv_idx.data(), total,
+                                              slice, world, d_bounds.data());
+  g.mr_recv_off = d_bounds.download(s);
+  // counts matrix on the control plane -> send-side region sizes
+  std::vector<uint64_t> mine(world);
+  for (int p = 0; p < world; ++p)
+    mine[p] = g.mr_recv_off[p + 1] - g.mr_recv_off[p];
+  std::vector<uint64_t> matrix(static_cast<size_t>(world) * world);
+  comm->allgather(mine.data(), world * 8, matrix.data());
+  g.mr_send_off.assign(world + 1, 0);
+  for (int q = 0; q < world; ++q)
+    g.mr_send_off[q + 1] =
+        g.mr_send_off[q] +
+        (q == rank ? 0 : matrix[static_cast<size_t>(q) * world + rank]);
+  uint64_t total_send = g.mr_send_off[world];
+  g.mr_send_idx.resize(total_send ? total_send : 1);
+  // ship each peer its request list; receive who wants what from us
+  std::vector<uint64_t> sb(world + 1), rb(world + 1);
+  for (int p = 0; p <= world; ++p) {
+    sb[p] = g.mr_recv_off[p] * 4;    // my requests, region p -> peer p
+    rb[p] = g.mr_send_off[p] * 4;    // peer q's requests for my slice
+  }
+  I.dc.sendrecv(reinterpret_cast<const uint8_t*>(g.mr_recv_idx.data()), sb,
+                reinterpret_cast<uint8_t*>(g.mr_send_idx.data()), rb, s);
+  g.mr_sendbuf.resize(total_send * 4 + 4);
+  g.mr_recvbuf.resize(total * 4 + 4);
+  HIP_CHECK(hipStreamSynchronize(s));
+  g.mirrors_built = true;
+  if (getenv("GRAPEHIP_DEBUG"))
+    fprintf(stderr,
+            "[mirror] rank %d: recv %lu refs, send %lu (slice %u)\n", rank,
+            (unsigned long)total, (unsigned long)total_send, slice);
+}
+
+// Refresh referenced remote entries of a 4-byte-element state array.
+// begin() packs on the compute stream and launches the exchange on the
+// comm stream; independent local work may run between begin and end —
+// end() orders the scatter after the payloads land.
+static void mirror_sync_begin(GpuContext::Impl& I, DeviceGraph& g,
+                              const void* state4, hipStream_t s) {
+  uint64_t ns = g.mr_send_off.back();
+  if (ns)
+    gather4_kernel<<<grid_for(ns), kBlock, 0, s>>>(
+        g.mr_send_idx.data(), ns,
+        static_cast<const uint32_t*>(state4),
+        reinterpret_cast<uint32_t*>(g.mr_sendbuf.data()));
+  I.ev_pack.record(s);
+  I.ev_pack.wait_on(I.comm_stream);
+  std::vector<uint64_t> sb(g.mr_send_off.size()), rb(g.mr_recv_off.size());
+  for (size_t p = 0; p < sb.size(); ++p) sb[p] = g.mr_send_off[p] * 4;
+  for (size_t p = 0; p < rb.size(); ++p) rb[p] = g.mr_recv_off[p] * 4;
+  I.dc.sendrecv(g.mr_sendbuf.data(), sb, g.mr_recvbuf.data(), rb,
+                I.comm_stream);
+  I.ev_comm.record(I.comm_stream);
+}
+
+static void mirror_sync_end(GpuContext::Impl& I, DeviceGraph& g,
+                            void* state4, hipStream_t s) {
+  I.ev_comm.wait_on(s);
+  uint64_t nr = g.mr_recv_off.back();
+  if (nr)
+    scatter4_kernel<<<grid_for(nr), kBlock, 0, s>>>(
+        g.mr_recv_idx.data(), nr,
+        reinterpret_cast<const uint32_t*>(g.mr_recvbuf.data()),
+        static_cast<uint32_t*>(state4));
+}
+
 // generic frontier expansion helper: scan degrees then CM-expand
 template <typename Op>
 void expand_frontier(GpuContext::Impl& I, const DevGraphView& view,
@@ -1970,12 +2183,12 @@ GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source,
   DevGraphView view = make_view(g, rank_, world_);
   uint32_t nv_pad = padded_nv(g, world_);
   uint32_t owned = g.owned();
-  uint32_t slice = view.slice;
   uint64_t cap = view.slice;
   bool multi = world_ > 1;
 
   const bool pull_capable = !g.directed || g.has_in;
   if (pull_capable) ensure_buckets(g, s);
+  if (multi && pull_capable) ensure_mirrors(I, comm_, g, rank_, world_, s);
   const uint64_t* pull_off = !g.directed ? g.oe_off.data() : g.ie_off.data();
   const uint32_t* pull_dst = !g.directed ? g.oe_dst.data() : g.ie_dst.data();
 
@@ -2029,7 +2242,12 @@ GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source,
       use_pull = gedges > pull_edge_threshold;
     }
     if (use_pull) {
-      if (multi) I.dc.allgather32(depth.data(), slice, false, s);
+      if (multi) {
+        // refresh referenced remote depths only (vs allgathering every
+        // slice): the pull scan reads depth[dst] for local dsts alone
+        mirror_sync_begin(I, g, depth.data(), s);
+        mirror_sync_end(I, g, depth.data(), s);
+      }
       DevBitmap nb{next_bm.data()};
       if (g.n_small)
         bfs_pull_small_kernel<<<grid_for(g.n_small), kBlock, 0, s>>>(
@@ -2289,6 +2507,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
   const uint32_t* pull_dst =
       (!g.directed ? g.oe_dst.data() : g.ie_dst.data());
   if (pull) ensure_buckets(g, s);
+  if (pull && multi) ensure_mirrors(I, comm_, g, rank_, world_, s);
 
   // working set lives on the DeviceGraph: the cached hipGraph bakes these
   // pointers (locals would dangle across calls — GPU fault under realloc)
@@ -2313,48 +2532,75 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
   int rounds = 0;
   const double inv_n = 1.0 / N;
 
+  // phase-dispatched pull sweep (PHASE 0 whole row / 1 owned span / 2 rest)
+  auto pulls = [&](int phase) {
+    auto launch = [&](auto tag) {
+      constexpr int P = decltype(tag)::value;
+      if (g.n_small)
+        pr_pull_small_kernel<P><<<grid_for(g.n_small), kBlock, 0, s>>>(
+            pull_off, pull_dst, contrib.data(), g.rows_small.data(),
+            g.n_small, g.v_begin, g.v_begin, g.v_end, acc.data());
+      if (g.n_mid)
+        pr_pull_mid_kernel<P><<<grid_for(g.n_mid * kWave), kBlock, 0, s>>>(
+            pull_off, pull_dst, contrib.data(), g.rows_mid.data(), g.n_mid,
+            g.v_begin, g.v_begin, g.v_end, acc.data());
+      if (g.n_large)
+        pr_pull_large_kernel<P><<<std::min<uint64_t>(g.n_large, kMaxGrid),
+                                  kBlock, 0, s>>>(
+            pull_off, pull_dst, contrib.data(), g.rows_large.data(),
+            g.n_large, g.v_begin, g.v_begin, g.v_end, acc.data());
+    };
+    if (phase == 0) launch(std::integral_constant<int, 0>{});
+    else if (phase == 1) launch(std::integral_constant<int, 1>{});
+    else launch(std::integral_constant<int, 2>{});
+  };
+  // fold the global dangling sum on the host via the pinned scalar
+  // (the D2H was enqueued earlier; ev_scal gates only that copy, so the
+  // GPU keeps crunching whatever was launched after it)
+  auto fold_dangling = [&]() {
+    I.ev_scal.sync();
+    std::vector<double> all(world_);
+    comm_->allgather(I.h_scal, 8, all.data());
+    double dang = 0;
+    for (double x : all) dang += x;
+    I.h_scal[1] = dang;
+    HIP_CHECK(hipMemcpyAsync(d_dangling.data(), I.h_scal + 1, 8,
+                             hipMemcpyHostToDevice, s));
+  };
+
   // one iteration, recorded as a stream of kernels. Single-GPU fixed-iter
   // runs capture it into a hipGraph once and replay (the iteration is
   // fully device-side: the dangling sum feeds pr_apply through a device
-  // scalar); multi-GPU and tol-convergence runs stay on the plain path
-  // (RCCL ops + per-iteration host reads).
+  // scalar). Multi-GPU pull runs overlap compute with communication: the
+  // owned-span pull (phase 1) executes on the compute stream WHILE the
+  // mirror exchange of remote contribs rides the comm stream over xGMI
+  // and the host folds the dangling scalar — then phase 2 accumulates
+  // the remote spans (north-star overlap; ROADMAP r01 design, landed).
   auto record_iteration = [&]() {
     d_dangling.zero(s);
     pr_dangling_kernel<<<grid_for(g.owned_real), kBlock, 0, s>>>(
         rank_arr.data(), g.oe_off.data(), g.owned_real, g.v_begin,
         d_dangling.data());
     if (multi) {
-      // fold the global dangling sum on the host (outside captures)
-      double local_dangling = 0;
-      HIP_CHECK(hipMemcpyAsync(&local_dangling, d_dangling.data(), 8,
+      HIP_CHECK(hipMemcpyAsync(I.h_scal, d_dangling.data(), 8,
                                hipMemcpyDeviceToHost, s));
-      HIP_CHECK(hipStreamSynchronize(s));
-      std::vector<double> all(world_);
-      comm_->allgather(&local_dangling, 8, all.data());
-      double dangling = 0;
-      for (double x : all) dangling += x;
-      HIP_CHECK(hipMemcpyAsync(d_dangling.data(), &dangling, 8,
-                               hipMemcpyHostToDevice, s));
+      I.ev_scal.record(s);
     }
     pr_contrib_kernel<<<grid_for(g.owned_real), kBlock, 0, s>>>(
         rank_arr.data(), g.oe_off.data(), g.owned_real, g.v_begin,
         contrib.data() + g.v_begin);
     if (pull) {
-      if (multi) I.dc.allgather32(contrib.data(), slice, true, s);
-      if (g.n_small)
-        pr_pull_small_kernel<<<grid_for(g.n_small), kBlock, 0, s>>>(
-            pull_off, pull_dst, contrib.data(), g.rows_small.data(),
-            g.n_small, g.v_begin, acc.data());
-      if (g.n_mid)
-        pr_pull_mid_kernel<<<grid_for(g.n_mid * kWave), kBlock, 0, s>>>(
-            pull_off, pull_dst, contrib.data(), g.rows_mid.data(), g.n_mid,
-            g.v_begin, acc.data());
-      if (g.n_large)
-        pr_pull_large_kernel<<<std::min<uint64_t>(g.n_large, kMaxGrid),
-                               kBlock, 0, s>>>(
-            pull_off, pull_dst, contrib.data(), g.rows_large.data(),
-            g.n_large, g.v_begin, acc.data());
+      if (multi) {
+        mirror_sync_begin(I, g, contrib.data(), s);
+        pulls(1);          // owned spans, overlapped with the exchange
+        fold_dangling();   // host work, overlapped too
+        mirror_sync_end(I, g, contrib.data(), s);
+        pulls(2);          // remote spans once payloads landed
+      } else {
+        pulls(0);
+      }
     } else {
+      if (multi) fold_dangling();
       acc.zero(s);
       int nchunks = static_cast<int>((owned + kBlock - 1) / kBlock);
       expand_cm_range<false, PrPushOp>
@@ -3074,7 +3320,6 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
   hipStream_t s = I.compute;
   uint32_t nv_pad = padded_nv(g, world_);
   uint32_t owned = g.owned();
-  uint32_t slice = nv_pad / (world_ ? world_ : 1);
   bool multi = world_ > 1;
   if (g.directed && !g.has_in)
     throw std::runtime_error(
@@ -3142,6 +3387,7 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
     view_in.oe_dst = g.ie_dst.data();
     view_in.oe_w = nullptr;
   }
+  if (multi) ensure_mirrors(I, comm_, g, rank_, world_, s);
 
   if (comm_) comm_->barrier();
   HIP_CHECK(hipDeviceSynchronize());
@@ -3189,7 +3435,12 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
                              s));
     HIP_CHECK(hipStreamSynchronize(s));
     uint64_t g_nch = multi ? comm_->allreduce_sum(nch) : nch;
-    if (multi) I.dc.allgather32(lab.data(), slice, false, s);
+    if (multi) {
+      // refresh referenced remote labels point-to-point (per-link volume
+      // scales with boundary size, not V·world)
+      mirror_sync_begin(I, g, lab.data(), s);
+      mirror_sync_end(I, g, lab.data(), s);
+    }
     if (g_nch == 0) break;  // fixpoint: further rounds are no-ops
     if (it + 1 == iters) break;
     // adaptive: building the dirty set costs an edge expansion over the

@@ -191,6 +191,7 @@ class Event {
   Event& operator=(const Event&) = delete;
   void record(hipStream_t s) { HIP_CHECK(hipEventRecord(e_, s)); }
   void wait_on(hipStream_t s) { HIP_CHECK(hipStreamWaitEvent(s, e_, 0)); }
+  void sync() { HIP_CHECK(hipEventSynchronize(e_)); }
 
  private:
   hipEvent_t e_ = nullptr;
