@@ -2085,6 +2085,60 @@ __global__ void scatter4_kernel(const uint32_t* __restrict__ idx, uint64_t n,
     state[idx[i]] = in[i];
 }
 
+// Build per-peer request lists from a referenced bitmap (consumed/cleared).
+// recv_idx: ascending global ids this rank wants values for, regioned per
+// owner; send_idx: the ids each peer wants from this rank's slice.
+// Returns the recv total.
+static uint64_t build_ref_lists(GpuContext::Impl& I, TcpComm* comm,
+                                DeviceBuffer<uint32_t>& bm, size_t nwords,
+                                uint32_t slice, int rank, int world,
+                                DeviceBuffer<uint32_t>& recv_idx,
+                                std::vector<uint64_t>& recv_off,
+                                DeviceBuffer<uint32_t>& send_idx,
+                                std::vector<uint64_t>& send_off,
+                                hipStream_t s) {
+  // compact ascending (popc + scan + fill, cleared as it fills)
+  if (I.frontier_deg.size() < nwords)
+    I.frontier_deg.resize(nwords + (nwords >> 2) + 64);
+  if (I.frontier_off.size() < nwords + 1)
+    I.frontier_off.resize(nwords + (nwords >> 2) + 65);
+  popc_words_kernel<<<grid_for(nwords), kBlock, 0, s>>>(bm.data(), nwords,
+                                                        I.frontier_deg.data());
+  uint64_t total = exclusive_scan(I.frontier_deg.data(),
+                                  I.frontier_off.data(), nwords, s, I.scan);
+  recv_idx.resize(total ? total : 1);
+  if (total)
+    fill_frontier_kernel<<<grid_for(nwords), kBlock, 0, s>>>(
+        bm.data(), nwords, I.frontier_off.data(), 0, recv_idx.data());
+  // per-peer boundaries (vid-ascending list, slice-contiguous regions)
+  DeviceBuffer<uint64_t> d_bounds(world + 1);
+  slice_bounds_kernel<<<1, world + 1, 0, s>>>(recv_idx.data(), total, slice,
+                                              world, d_bounds.data());
+  recv_off = d_bounds.download(s);
+  // counts matrix on the control plane -> send-side region sizes
+  std::vector<uint64_t> mine(world);
+  for (int p = 0; p < world; ++p) mine[p] = recv_off[p + 1] - recv_off[p];
+  std::vector<uint64_t> matrix(static_cast<size_t>(world) * world);
+  comm->allgather(mine.data(), world * 8, matrix.data());
+  send_off.assign(world + 1, 0);
+  for (int q = 0; q < world; ++q)
+    send_off[q + 1] =
+        send_off[q] +
+        (q == rank ? 0 : matrix[static_cast<size_t>(q) * world + rank]);
+  uint64_t total_send = send_off[world];
+  send_idx.resize(total_send ? total_send : 1);
+  // ship each peer its request list; receive who wants what from us
+  std::vector<uint64_t> sb(world + 1), rb(world + 1);
+  for (int p = 0; p <= world; ++p) {
+    sb[p] = recv_off[p] * 4;    // my requests, region p -> peer p
+    rb[p] = send_off[p] * 4;    // peer q's requests for my slice
+  }
+  I.dc.sendrecv(reinterpret_cast<const uint8_t*>(recv_idx.data()), sb,
+                reinterpret_cast<uint8_t*>(send_idx.data()), rb, s);
+  HIP_CHECK(hipStreamSynchronize(s));
+  return total;
+}
+
 static void ensure_mirrors(GpuContext::Impl& I, TcpComm* comm,
                            DeviceGraph& g, int rank, int world,
                            hipStream_t s) {
@@ -2102,48 +2156,13 @@ static void ensure_mirrors(GpuContext::Impl& I, TcpComm* comm,
     mark_dsts_kernel<<<grid_for(g.ie_dst.size()), kBlock, 0, s>>>(
         g.ie_dst.data(), g.ie_dst.size(), g.v_begin, g.v_end,
         DevBitmap{bm.data()});
-  // compact ascending (popc + scan + fill, cleared as it fills)
-  if (I.frontier_deg.size() < nwords)
-    I.frontier_deg.resize(nwords + (nwords >> 2) + 64);
-  if (I.frontier_off.size() < nwords + 1)
-    I.frontier_off.resize(nwords + (nwords >> 2) + 65);
-  popc_words_kernel<<<grid_for(nwords), kBlock, 0, s>>>(bm.data(), nwords,
-                                                        I.frontier_deg.data());
-  uint64_t total = exclusive_scan(I.frontier_deg.data(),
-                                  I.frontier_off.data(), nwords, s, I.scan);
-  g.mr_recv_idx.resize(total ? total : 1);
-  if (total)
-    fill_frontier_kernel<<<grid_for(nwords), kBlock, 0, s>>>(
-        bm.data(), nwords, I.frontier_off.data(), 0, g.mr_recv_idx.data());
-  // per-peer boundaries (vid-ascending list, slice-contiguous regions)
-  DeviceBuffer<uint64_t> d_bounds(world + 1);
-  slice_bounds_kernel<<<1, world + 1, 0, s>>>(g.mr_recv_idx.data(), total,
-                                              slice, world, d_bounds.data());
-  g.mr_recv_off = d_bounds.download(s);
-  // counts matrix on the control plane -> send-side region sizes
-  std::vector<uint64_t> mine(world);
-  for (int p = 0; p < world; ++p)
-    mine[p] = g.mr_recv_off[p + 1] - g.mr_recv_off[p];
-  std::vector<uint64_t> matrix(static_cast<size_t>(world) * world);
-  comm->allgather(mine.data(), world * 8, matrix.data());
-  g.mr_send_off.assign(world + 1, 0);
-  for (int q = 0; q < world; ++q)
-    g.mr_send_off[q + 1] =
-        g.mr_send_off[q] +
-        (q == rank ? 0 : matrix[static_cast<size_t>(q) * world + rank]);
+  uint64_t total =
+      build_ref_lists(I, comm, bm, nwords, slice, rank, world,
+                      g.mr_recv_idx, g.mr_recv_off, g.mr_send_idx,
+                      g.mr_send_off, s);
   uint64_t total_send = g.mr_send_off[world];
-  g.mr_send_idx.resize(total_send ? total_send : 1);
-  // ship each peer its request list; receive who wants what from us
-  std::vector<uint64_t> sb(world + 1), rb(world + 1);
-  for (int p = 0; p <= world; ++p) {
-    sb[p] = g.mr_recv_off[p] * 4;    // my requests, region p -> peer p
-    rb[p] = g.mr_send_off[p] * 4;    // peer q's requests for my slice
-  }
-  I.dc.sendrecv(reinterpret_cast<const uint8_t*>(g.mr_recv_idx.data()), sb,
-                reinterpret_cast<uint8_t*>(g.mr_send_idx.data()), rb, s);
   g.mr_sendbuf.resize(total_send * 4 + 4);
   g.mr_recvbuf.resize(total * 4 + 4);
-  HIP_CHECK(hipStreamSynchronize(s));
   g.mirrors_built = true;
   if (getenv("GRAPEHIP_DEBUG"))
     fprintf(stderr,
@@ -2151,37 +2170,55 @@ static void ensure_mirrors(GpuContext::Impl& I, TcpComm* comm,
             (unsigned long)total, (unsigned long)total_send, slice);
 }
 
-// Refresh referenced remote entries of a 4-byte-element state array.
-// begin() packs on the compute stream and launches the exchange on the
-// comm stream; independent local work may run between begin and end —
-// end() orders the scatter after the payloads land.
-static void mirror_sync_begin(GpuContext::Impl& I, DeviceGraph& g,
-                              const void* state4, hipStream_t s) {
-  uint64_t ns = g.mr_send_off.back();
+// Refresh referenced remote entries of a 4-byte-element state array over
+// arbitrary request lists. begin() packs on the compute stream and
+// launches the exchange on the comm stream; independent local work may
+// run between begin and end — end() orders the scatter after payloads
+// land.
+static void ref_sync_begin(GpuContext::Impl& I,
+                           const DeviceBuffer<uint32_t>& send_idx,
+                           const std::vector<uint64_t>& send_off,
+                           const std::vector<uint64_t>& recv_off,
+                           DeviceBuffer<uint8_t>& sendbuf,
+                           DeviceBuffer<uint8_t>& recvbuf,
+                           const void* state4, hipStream_t s) {
+  uint64_t ns = send_off.back();
   if (ns)
     gather4_kernel<<<grid_for(ns), kBlock, 0, s>>>(
-        g.mr_send_idx.data(), ns,
-        static_cast<const uint32_t*>(state4),
-        reinterpret_cast<uint32_t*>(g.mr_sendbuf.data()));
+        send_idx.data(), ns, static_cast<const uint32_t*>(state4),
+        reinterpret_cast<uint32_t*>(sendbuf.data()));
   I.ev_pack.record(s);
   I.ev_pack.wait_on(I.comm_stream);
-  std::vector<uint64_t> sb(g.mr_send_off.size()), rb(g.mr_recv_off.size());
-  for (size_t p = 0; p < sb.size(); ++p) sb[p] = g.mr_send_off[p] * 4;
-  for (size_t p = 0; p < rb.size(); ++p) rb[p] = g.mr_recv_off[p] * 4;
-  I.dc.sendrecv(g.mr_sendbuf.data(), sb, g.mr_recvbuf.data(), rb,
-                I.comm_stream);
+  std::vector<uint64_t> sb(send_off.size()), rb(recv_off.size());
+  for (size_t p = 0; p < sb.size(); ++p) sb[p] = send_off[p] * 4;
+  for (size_t p = 0; p < rb.size(); ++p) rb[p] = recv_off[p] * 4;
+  I.dc.sendrecv(sendbuf.data(), sb, recvbuf.data(), rb, I.comm_stream);
   I.ev_comm.record(I.comm_stream);
+}
+
+static void ref_sync_end(GpuContext::Impl& I,
+                         const DeviceBuffer<uint32_t>& recv_idx,
+                         const std::vector<uint64_t>& recv_off,
+                         const DeviceBuffer<uint8_t>& recvbuf, void* state4,
+                         hipStream_t s) {
+  I.ev_comm.wait_on(s);
+  uint64_t nr = recv_off.back();
+  if (nr)
+    scatter4_kernel<<<grid_for(nr), kBlock, 0, s>>>(
+        recv_idx.data(), nr,
+        reinterpret_cast<const uint32_t*>(recvbuf.data()),
+        static_cast<uint32_t*>(state4));
+}
+
+static void mirror_sync_begin(GpuContext::Impl& I, DeviceGraph& g,
+                              const void* state4, hipStream_t s) {
+  ref_sync_begin(I, g.mr_send_idx, g.mr_send_off, g.mr_recv_off,
+                 g.mr_sendbuf, g.mr_recvbuf, state4, s);
 }
 
 static void mirror_sync_end(GpuContext::Impl& I, DeviceGraph& g,
                             void* state4, hipStream_t s) {
-  I.ev_comm.wait_on(s);
-  uint64_t nr = g.mr_recv_off.back();
-  if (nr)
-    scatter4_kernel<<<grid_for(nr), kBlock, 0, s>>>(
-        g.mr_recv_idx.data(), nr,
-        reinterpret_cast<const uint32_t*>(g.mr_recvbuf.data()),
-        static_cast<uint32_t*>(state4));
+  ref_sync_end(I, g.mr_recv_idx, g.mr_recv_off, g.mr_recvbuf, state4, s);
 }
 
 // generic frontier expansion helper: scan degrees then CM-expand
@@ -4802,6 +4839,45 @@ namespace {
 
 // Shared piece: turn capacity-layout per-owned-row lists into a replicated
 // global CSR (allgather counts -> scan -> compact -> per-rank broadcast).
+// mark remote dsts of the capacity-layout oriented adjacency (wave/row)
+__global__ void lcc_mark_oriented_kernel(
+    const uint64_t* __restrict__ ooff, const uint32_t* __restrict__ ocnt,
+    const uint32_t* __restrict__ oadj, uint32_t owned, uint32_t v_begin,
+    uint32_t v_end, DevBitmap bm) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wpb = kBlock / kWave;
+  size_t wstride = static_cast<size_t>(gridDim.x) * wpb;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid;
+       i < owned; i += wstride) {
+    uint64_t b = ooff[i];
+    uint32_t n = ocnt[i];
+    for (uint32_t k = lane; k < n; k += kWave) {
+      uint32_t v = oadj[b + k];
+      if (v < v_begin || v >= v_end) (void)bm.set_once(v);
+    }
+  }
+}
+
+// pack requested rows (sorted, variable length) into a contiguous payload
+__global__ void lcc_pack_rows_kernel(const uint32_t* __restrict__ idx,
+                                     uint64_t n,
+                                     const uint64_t* __restrict__ goff,
+                                     const uint32_t* __restrict__ gdst,
+                                     const uint64_t* __restrict__ pack_off,
+                                     uint32_t* __restrict__ out) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wpb = kBlock / kWave;
+  size_t wstride = static_cast<size_t>(gridDim.x) * wpb;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid; i < n;
+       i += wstride) {
+    uint64_t b = goff[idx[i]];
+    uint64_t o = pack_off[i], len = pack_off[i + 1] - o;
+    for (uint64_t k = lane; k < len; k += kWave) out[o + k] = gdst[b + k];
+  }
+}
+
 struct GlobalDedupCsr {
   DeviceBuffer<uint32_t> gcnt;
   DeviceBuffer<uint64_t> goff;
@@ -4809,18 +4885,77 @@ struct GlobalDedupCsr {
   uint64_t total = 0;
 };
 
+// Fetch the referenced remote rows of a sparse-count CSR: each owner
+// packs exactly the rows each peer requested (sidx/soff lists); payloads
+// land directly in the requester's contiguous per-owner gdst regions
+// (unfetched rows have zero count, so regions are hole-free). Counts
+// must already be synced over the same lists.
+void lcc_fetch_referenced_rows(GpuContext::Impl& I,
+                               const DeviceBuffer<uint32_t>& sidx,
+                               const std::vector<uint64_t>& soff,
+                               const DeviceBuffer<uint32_t>& gcnt,
+                               const DeviceBuffer<uint64_t>& goff,
+                               DeviceBuffer<uint32_t>& gdst, uint32_t slice,
+                               uint32_t nv_pad, int world, hipStream_t s) {
+  uint64_t ns = soff[world];
+  DeviceBuffer<uint32_t> slen(ns ? ns : 1);
+  if (ns)
+    gather4_kernel<<<grid_for(ns), kBlock, 0, s>>>(sidx.data(), ns,
+                                                   gcnt.data(), slen.data());
+  DeviceBuffer<uint64_t> spack(ns + 1);
+  uint64_t stotal = exclusive_scan(slen.data(), spack.data(), ns, s, I.scan);
+  DeviceBuffer<uint32_t> srows(stotal ? stotal : 1);
+  if (ns)
+    lcc_pack_rows_kernel<<<grid_for(ns * kWave), kBlock, 0, s>>>(
+        sidx.data(), ns, goff.data(), gdst.data(), spack.data(),
+        srows.data());
+  // per-peer byte offsets: send = packed scan at list boundaries,
+  // recv = goff at slice boundaries (both derive from the same counts)
+  std::vector<uint64_t> sb(world + 1), rb(world + 1);
+  for (int f = 0; f <= world; ++f) {
+    uint64_t v = 0;
+    HIP_CHECK(hipMemcpyAsync(&v, spack.data() + soff[f], 8,
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    sb[f] = v * 4;
+    uint64_t idx =
+        std::min<uint64_t>(static_cast<uint64_t>(f) * slice, nv_pad);
+    HIP_CHECK(hipMemcpyAsync(&v, goff.data() + idx, 8,
+                             hipMemcpyDeviceToHost, s));
+    HIP_CHECK(hipStreamSynchronize(s));
+    rb[f] = v * 4;
+  }
+  I.dc.sendrecv(reinterpret_cast<const uint8_t*>(srows.data()), sb,
+                reinterpret_cast<uint8_t*>(gdst.data()), rb, s);
+  HIP_CHECK(hipStreamSynchronize(s));
+}
+
+// Compact a capacity-layout family into a sparse global-index CSR. With
+// lists (multi): counts are synced over the request lists and referenced
+// remote rows fetched (volume scales with the boundary). Without lists:
+// own rows only — for families that are never probed at remote indices
+// (directed U, read only at owned v).
 void build_global_csr(GpuContext::Impl& I, int rank, int world,
                       uint32_t nv_pad, uint32_t slice, uint32_t owned,
                       uint32_t v_begin, const DeviceBuffer<uint64_t>& ooff,
                       const DeviceBuffer<uint32_t>& ocnt,
                       const DeviceBuffer<uint32_t>& oadj,
-                      GlobalDedupCsr& out, hipStream_t s) {
-  bool multi = world > 1;
+                      GlobalDedupCsr& out, hipStream_t s,
+                      const DeviceBuffer<uint32_t>* ridx = nullptr,
+                      const std::vector<uint64_t>* roff = nullptr,
+                      const DeviceBuffer<uint32_t>* sidx = nullptr,
+                      const std::vector<uint64_t>* soff = nullptr) {
+  bool fetch = world > 1 && ridx;
   out.gcnt.resize(nv_pad);
   out.gcnt.zero(s);
   HIP_CHECK(hipMemcpyAsync(out.gcnt.data() + v_begin, ocnt.data(),
                            owned * 4, hipMemcpyDeviceToDevice, s));
-  if (multi) I.dc.allgather32(out.gcnt.data(), slice, false, s);
+  if (fetch) {
+    DeviceBuffer<uint8_t> ss((*soff)[world] * 4 + 4),
+        sr(roff->back() * 4 + 4);
+    ref_sync_begin(I, *sidx, *soff, *roff, ss, sr, out.gcnt.data(), s);
+    ref_sync_end(I, *ridx, *roff, sr, out.gcnt.data(), s);
+  }
   out.goff.resize(static_cast<size_t>(nv_pad) + 1);
   out.total =
       exclusive_scan(out.gcnt.data(), out.goff.data(), nv_pad, s, I.scan);
@@ -4830,17 +4965,10 @@ void build_global_csr(GpuContext::Impl& I, int rank, int world,
                          kBlock, 0, s>>>(ooff.data(), ocnt.data(),
                                          oadj.data(), out.goff.data(),
                                          owned, v_begin, out.gdst.data());
-  if (multi) {
-    std::vector<uint64_t> region(world + 1);
-    for (int f = 0; f <= world; ++f) {
-      uint64_t idx =
-          std::min<uint64_t>(static_cast<uint64_t>(f) * slice, nv_pad);
-      HIP_CHECK(hipMemcpyAsync(&region[f], out.goff.data() + idx, 8,
-                               hipMemcpyDeviceToHost, s));
-    }
-    HIP_CHECK(hipStreamSynchronize(s));
-    I.dc.bcast_regions_u32(out.gdst.data(), region, s);
-  }
+  if (fetch)
+    lcc_fetch_referenced_rows(I, *sidx, *soff, out.gcnt, out.goff,
+                              out.gdst, slice, nv_pad, world, s);
+  (void)rank;
 }
 
 void build_hash_sets(GpuContext::Impl& I, uint32_t nv_pad,
@@ -5093,13 +5221,22 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
   const uint64_t* off2 = nullptr;
   const uint32_t* dst2 = nullptr;
 
-  // graph-sized working set (freed on return)
+  // graph-sized working set (freed on return). Multi-GPU holds a PARTIAL
+  // oriented CSR: own rows + the remote rows local oriented edges probe
+  // (reference TransferAdjList staging, cuda/lcc/lcc.h:490-496, recast as
+  // a one-shot referenced-row fetch) — memory and exchange volume scale
+  // with the boundary, not with V.
   uint64_t oriented_total = 0;
   DeviceBuffer<uint32_t> Dv;                 // distinct degree, global idx
-  DeviceBuffer<uint32_t> gcnt;               // oriented count, global idx
-  DeviceBuffer<uint64_t> goff;               // global oriented CSR offsets
-  DeviceBuffer<uint32_t> gdst;               // global oriented CSR dsts
+  DeviceBuffer<uint32_t> gcnt;               // oriented count (own+referenced)
+  DeviceBuffer<uint64_t> goff;               // oriented CSR offsets (sparse)
+  DeviceBuffer<uint32_t> gdst;               // oriented CSR dsts (partial)
   DeviceBuffer<unsigned long long> Tcnt;     // triangle credits
+  // referenced-row request lists + staging for the count sync
+  DeviceBuffer<uint32_t> ridx, sidx;
+  std::vector<uint64_t> roff, soff;
+  DeviceBuffer<uint8_t> rstage_s, rstage_r;
+  if (multi) ensure_mirrors(I, comm_, g, rank_, world_, s);
 
   if (comm_) comm_->barrier();
   HIP_CHECK(hipDeviceSynchronize());
@@ -5159,7 +5296,12 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
                                        n_large, heavy_off.data(),
                                        tbl_lab.data(), g.v_begin, Dv.data());
     }
-    if (multi) I.dc.allgather32(Dv.data(), slice, false, s);
+    if (multi) {
+      // the orient passes read Dv[dst] for local dsts only — refresh
+      // referenced entries point-to-point instead of allgathering slices
+      mirror_sync_begin(I, g, Dv.data(), s);
+      mirror_sync_end(I, g, Dv.data(), s);
+    }
 
     // pass 2: oriented adjacency in capacity layout (cap = D per row)
     DeviceBuffer<uint32_t> dcap(owned ? owned : 1);
@@ -5194,12 +5336,32 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
                                      true);
     }
 
-    // global oriented CSR: allgather counts, scan, compact, sort, exchange
+    // partial oriented CSR: own counts + counts of the rows local
+    // oriented edges reference (synced over explicit request lists so
+    // unfetched rows scan to zero length and the fetch regions stay
+    // contiguous per owner)
     gcnt.resize(nv_pad);
     gcnt.zero(s);
     HIP_CHECK(hipMemcpyAsync(gcnt.data() + g.v_begin, ocnt.data(),
                              owned * 4, hipMemcpyDeviceToDevice, s));
-    if (multi) I.dc.allgather32(gcnt.data(), slice, false, s);
+    if (multi) {
+      size_t nwords = (static_cast<size_t>(nv_pad) + 31) / 32;
+      DeviceBuffer<uint32_t> bm(nwords);
+      bm.zero(s);
+      if (owned)
+        lcc_mark_oriented_kernel<<<grid_for(static_cast<size_t>(owned) *
+                                            kWave),
+                                   kBlock, 0, s>>>(
+            ooff.data(), ocnt.data(), oadj.data(), owned, g.v_begin,
+            g.v_end, DevBitmap{bm.data()});
+      uint64_t nref = build_ref_lists(I, comm_, bm, nwords, slice, rank_,
+                                      world_, ridx, roff, sidx, soff, s);
+      rstage_s.resize(soff[world_] * 4 + 4);
+      rstage_r.resize(nref * 4 + 4);
+      ref_sync_begin(I, sidx, soff, roff, rstage_s, rstage_r, gcnt.data(),
+                     s);
+      ref_sync_end(I, ridx, roff, rstage_r, gcnt.data(), s);
+    }
     goff.resize(static_cast<size_t>(nv_pad) + 1);
     uint64_t g_total =
         exclusive_scan(gcnt.data(), goff.data(), nv_pad, s, I.scan);
@@ -5244,20 +5406,43 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
           scratch.data(), g.v_begin, gdst.data());
     }
 
-    // ragged allgather of gdst: each rank broadcasts its region
+    // fetch the referenced remote rows (sorted by their owners above):
+    // each owner packs exactly the rows each peer requested; payloads
+    // land directly in the requester's contiguous per-owner gdst region
+    // (unfetched rows have zero length, so regions are hole-free)
     if (multi) {
-      std::vector<uint64_t> region(world_ + 1);
-      // region boundaries = goff at slice edges (host copy of w+1 values)
+      uint64_t ns = soff[world_];
+      DeviceBuffer<uint32_t> slen(ns ? ns : 1);
+      if (ns)
+        gather4_kernel<<<grid_for(ns), kBlock, 0, s>>>(
+            sidx.data(), ns, gcnt.data(), slen.data());
+      DeviceBuffer<uint64_t> spack(ns + 1);
+      uint64_t stotal =
+          exclusive_scan(slen.data(), spack.data(), ns, s, I.scan);
+      DeviceBuffer<uint32_t> srows(stotal ? stotal : 1);
+      if (ns)
+        lcc_pack_rows_kernel<<<grid_for(ns * kWave), kBlock, 0, s>>>(
+            sidx.data(), ns, goff.data(), gdst.data(), spack.data(),
+            srows.data());
+      // per-peer byte offsets: send = packed scan at list boundaries,
+      // recv = goff at slice boundaries (both derive from the same gcnt)
+      std::vector<uint64_t> sb(world_ + 1), rb(world_ + 1);
       for (int f = 0; f <= world_; ++f) {
-        uint64_t off_v;
-        uint64_t idx = std::min<uint64_t>(
-            static_cast<uint64_t>(f) * slice, nv_pad);
-        HIP_CHECK(hipMemcpyAsync(&off_v, goff.data() + idx, 8,
+        uint64_t v = 0;
+        HIP_CHECK(hipMemcpyAsync(&v, spack.data() + soff[f], 8,
                                  hipMemcpyDeviceToHost, s));
         HIP_CHECK(hipStreamSynchronize(s));
-        region[f] = off_v;
+        sb[f] = v * 4;
+        uint64_t idx = std::min<uint64_t>(
+            static_cast<uint64_t>(f) * slice, nv_pad);
+        HIP_CHECK(hipMemcpyAsync(&v, goff.data() + idx, 8,
+                                 hipMemcpyDeviceToHost, s));
+        HIP_CHECK(hipStreamSynchronize(s));
+        rb[f] = v * 4;
       }
-      I.dc.bcast_regions_u32(gdst.data(), region, s);
+      I.dc.sendrecv(reinterpret_cast<const uint8_t*>(srows.data()), sb,
+                    reinterpret_cast<uint8_t*>(gdst.data()), rb, s);
+      HIP_CHECK(hipStreamSynchronize(s));
     }
   }
 
