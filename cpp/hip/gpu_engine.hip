@@ -2762,9 +2762,17 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
   uint32_t nv_pad = padded_nv(g, world_);
   uint32_t owned = g.owned();
   bool multi = world_ > 1;
-  if (g.directed && !g.has_in && !multi) {
-    // weak connectivity on a directed graph needs both directions; the
-    // out-CSR alone still merges u->v (hook is symmetric), so this is fine.
+  // Directed graphs: the Afforest giant-skip is only sound when every
+  // vertex can see all edges that touch it. An edge u->v stored solely in
+  // a skipped giant row u may be the ONLY link to v — with an in-CSR the
+  // rest pass co-expands v's in-edges to cover it; without one the skip
+  // is disabled (full fixpoint passes).
+  const bool rest_use_in = g.directed && g.has_in;
+  DevGraphView view_in = view;
+  if (rest_use_in) {
+    view_in.oe_off = g.ie_off.data();
+    view_in.oe_dst = g.ie_dst.data();
+    view_in.oe_w = nullptr;
   }
 
   DeviceBuffer<uint32_t> parent(nv_pad);
@@ -2831,20 +2839,30 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
         }
       }
     }
+    if (g.directed && !g.has_in) giant = 0xFFFFFFFFu;  // skip unsound
     // 3) full pass restricted to rows outside the giant component,
     //    repeated to the local fixpoint
     for (;;) {
       rest_bm.zero(s);
-      if (owned)
+      if (owned) {
         wcc_mark_rest_kernel<<<grid_for(owned), kBlock, 0, s>>>(
             parent.data(), view.oe_off, owned, g.v_begin, giant,
             DevBitmap{rest_bm.data()});
+        if (rest_use_in)
+          wcc_mark_rest_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+              parent.data(), view_in.oe_off, owned, g.v_begin, giant,
+              DevBitmap{rest_bm.data()});
+      }
       uint64_t nrest = compact_frontier(I, rest_bm.data(), owned, g.v_begin,
                                         rest_q.data(), s);
       if (nrest == 0) break;
       d_changed.zero(s);
       expand_frontier(I, view, rest_q.data(), static_cast<uint32_t>(nrest),
                       WccOp{parent.data(), d_changed.data(), marks}, s);
+      if (rest_use_in)
+        expand_frontier(I, view_in, rest_q.data(),
+                        static_cast<uint32_t>(nrest),
+                        WccOp{parent.data(), d_changed.data(), marks}, s);
       int ch = 0;
       HIP_CHECK(hipMemcpyAsync(&ch, d_changed.data(), 4,
                                hipMemcpyDeviceToHost, s));
@@ -5008,6 +5026,13 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
   const uint64_t* off2 = g.ie_off.data();
   const uint32_t* dst2 = g.ie_dst.data();
 
+  // referenced-row request lists (multi): only the O family is probed at
+  // remote indices (kernels read U solely at owned v), so U stays local
+  // and O rows are fetched for the u's own U rows actually name
+  DeviceBuffer<uint32_t> ridx, sidx;
+  std::vector<uint64_t> roff, soff;
+  if (multi) ensure_mirrors(I, comm_, g, rank_, world_, s);
+
   if (comm_) comm_->barrier();
   HIP_CHECK(hipDeviceSynchronize());
   double t0 = wall_s();
@@ -5068,14 +5093,18 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
                                           heavy_off.data(), tbl_lab.data(),
                                           g.v_begin, Dv.data());
     }
-    if (multi) I.dc.allgather32(Dv.data(), slice, false, s);
+    if (multi) {
+      mirror_sync_begin(I, g, Dv.data(), s);
+      mirror_sync_end(I, g, Dv.data(), s);
+    }
 
     // family builder: run the dedup pass over (o1,d1,o2,d2) with given
-    // capacities, then lift to a replicated global CSR
+    // capacities, then lift to a sparse global-index CSR (own rows, plus
+    // fetched referenced rows when lists are supplied)
     auto build_family = [&](const uint64_t* o1, const uint32_t* d1,
                             const uint64_t* o2, const uint32_t* d2,
                             const DeviceBuffer<uint32_t>& caps,
-                            GlobalDedupCsr& out) {
+                            GlobalDedupCsr& out, bool fetch_refs) {
       DeviceBuffer<uint64_t> ooff(owned + 1);
       uint64_t cap_total =
           exclusive_scan(caps.data(), ooff.data(), owned, s, I.scan);
@@ -5108,15 +5137,40 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
                                           oadj.data(), ocnt.data(), false);
       }
       build_global_csr(I, rank_, world_, nv_pad, slice, owned, g.v_begin,
-                       ooff, ocnt, oadj, out, s);
+                       ooff, ocnt, oadj, out, s,
+                       fetch_refs ? &ridx : nullptr,
+                       fetch_refs ? &roff : nullptr,
+                       fetch_refs ? &sidx : nullptr,
+                       fetch_refs ? &soff : nullptr);
     };
 
-    // U family: in ∪ out, cap = D
+    // U family: in ∪ out, cap = D — never probed remotely, stays local
     {
       DeviceBuffer<uint32_t> caps(owned ? owned : 1);
       HIP_CHECK(hipMemcpyAsync(caps.data(), Dv.data() + g.v_begin,
                                owned * 4, hipMemcpyDeviceToDevice, s));
-      build_family(off1, dst1, off2, dst2, caps, U);
+      build_family(off1, dst1, off2, dst2, caps, U, false);
+    }
+    // referenced u's = dsts of own U rows (the only rows whose O family
+    // the count kernels probe)
+    if (multi) {
+      uint64_t ub = 0, ue = 0;
+      HIP_CHECK(hipMemcpyAsync(&ub, U.goff.data() + g.v_begin, 8,
+                               hipMemcpyDeviceToHost, s));
+      HIP_CHECK(hipMemcpyAsync(&ue,
+                               U.goff.data() + std::min<uint64_t>(
+                                   g.v_end, nv_pad),
+                               8, hipMemcpyDeviceToHost, s));
+      HIP_CHECK(hipStreamSynchronize(s));
+      size_t nwords = (static_cast<size_t>(nv_pad) + 31) / 32;
+      DeviceBuffer<uint32_t> bm(nwords);
+      bm.zero(s);
+      if (ue > ub)
+        mark_dsts_kernel<<<grid_for(ue - ub), kBlock, 0, s>>>(
+            U.gdst.data() + ub, ue - ub, g.v_begin, g.v_end,
+            DevBitmap{bm.data()});
+      build_ref_lists(I, comm_, bm, nwords, slice, rank_, world_, ridx,
+                      roff, sidx, soff, s);
     }
     // O family: distinct out, cap = min(out_deg, D)
     {
@@ -5124,7 +5178,7 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
       if (owned)
         lcc_outcap_kernel<<<grid_for(owned), kBlock, 0, s>>>(
             off1, Dv.data(), owned, g.v_begin, caps.data());
-      build_family(off1, dst1, nullptr, nullptr, caps, O);
+      build_family(off1, dst1, nullptr, nullptr, caps, O, multi);
     }
   }
 

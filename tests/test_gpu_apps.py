@@ -82,6 +82,25 @@ def test_gpu_wcc(eng):
     assert np.array_equal(vals, wcc_oracle(8000, src, dst))
 
 
+def test_gpu_wcc_directed_sparse(eng):
+    # regression: on a directed graph the Afforest giant-skip must not
+    # drop edges stored only in skipped giant rows — an in-degree-only
+    # vertex reachable solely via such an edge was left unmerged
+    src, dst, _ = random_graph(num_v=60000, num_e=200000, seed=7)
+    g = eng.load_edges(src, dst, directed=True, num_vertices=60000,
+                       build_in_csr=True)
+    _, vals = by_oid(eng.wcc(g))
+    assert np.array_equal(vals, wcc_oracle(60000, src, dst))
+
+
+def test_gpu_wcc_directed_no_incsr(eng):
+    # without an in-CSR the skip is disabled entirely; result must agree
+    src, dst, _ = random_graph(num_v=30000, num_e=90000, seed=23)
+    g = eng.load_edges(src, dst, directed=True, num_vertices=30000)
+    _, vals = by_oid(eng.wcc(g))
+    assert np.array_equal(vals, wcc_oracle(30000, src, dst))
+
+
 def test_gpu_synthetic(eng):
     g = eng.load_synthetic(num_vertices=100000, num_edges=1600000, seed=1,
                            weighted=True)

@@ -37,8 +37,10 @@ sys.path.insert(0, os.environ["GRAPEHIP_REPO"])
 import grapehip
 import numpy as np
 eng = grapehip.engine_from_env(gpu=True)
+directed = os.environ.get("GRAPEHIP_TEST_DIRECTED") == "1"
 g = eng.load_synthetic(num_vertices=200000, num_edges=1600000, seed=7,
-                       weighted=True)
+                       weighted=True, directed=directed,
+                       build_in_csr=directed)
 out = {}
 for app in ("bfs", "sssp", "pagerank", "wcc", "cdlp", "lcc"):
     if app == "bfs":
@@ -60,12 +62,13 @@ json.dump(out, open(path, "w"))
 '''
 
 
-def run_world(world, outdir, port):
+def run_world(world, outdir, port, directed=False):
     procs = []
     for rank in range(world):
         env = dict(os.environ, GRAPEHIP_REPO=str(REPO),
                    GRAPEHIP_OUT=str(outdir), RANK=str(rank),
                    LOCAL_RANK=str(rank), WORLD_SIZE=str(world),
+                   GRAPEHIP_TEST_DIRECTED="1" if directed else "0",
                    MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
         procs.append(subprocess.Popen([sys.executable, "-c", WORKER],
                                       env=env, stdout=subprocess.PIPE,
@@ -112,4 +115,17 @@ def test_multirank_matches_single(tmp_path, world):
     multi_dir = tmp_path / ("w%d" % world)
     multi_dir.mkdir()
     got = run_world(world, multi_dir, 29740 + world)
+    check_against_single(single, got)
+
+
+@pytest.mark.skipif(device_count() < 1, reason="needs a GPU")
+def test_multirank_directed_matches_single(tmp_path):
+    # directed path: in-CSR pulls, directed CDLP multiset, directed LCC
+    # (local U family + fetched O rows)
+    single_dir = tmp_path / "single"
+    single_dir.mkdir()
+    single = run_world(1, single_dir, 29760, directed=True)
+    multi_dir = tmp_path / "w2d"
+    multi_dir.mkdir()
+    got = run_world(2, multi_dir, 29770, directed=True)
     check_against_single(single, got)
