@@ -962,7 +962,11 @@ __device__ __forceinline__ uint32_t wcc_find(uint32_t* parent, uint32_t v) {
     if (p == v) return v;
     uint32_t gp = parent[p];
     if (gp == p) return p;
-    parent[v] = gp;  // benign race: path halving
+    // path halving MUST be monotone: a plain store can overwrite a
+    // concurrently-lowered entry with a stale-derived higher value — or
+    // even re-self-root it (gp read stale == v), severing a hook link
+    // and splitting the component (seen at 1M-vertex scale)
+    atomicMin(&parent[v], gp);
     v = gp;
   }
 }
@@ -979,13 +983,16 @@ struct WccOp {
       uint32_t rv = wcc_find(parent, v);
       if (ru == rv) return;
       uint32_t hi = ru > rv ? ru : rv, lo = ru > rv ? rv : ru;
-      if (atomicCAS(&parent[hi], hi, lo) == hi) {
+      uint32_t old = atomicCAS(&parent[hi], hi, lo);
+      if (old == hi) {
         *changed = 1;
         if (mark_words)
           atomicOr(&mark_words[hi >> 5], 1u << (hi & 31));
         return;
       }
-      u = hi;
+      // CAS returned the CURRENT entry — chase from it rather than
+      // re-reading parent[hi] (a stale cached line would spin forever)
+      u = old;
       v = lo;
     }
   }
@@ -998,8 +1005,29 @@ __global__ void wcc_compress_kernel(uint32_t* parent, uint32_t n,
        v += stride) {
     uint32_t r = wcc_find(parent, v);
     if (parent[v] != r) {
-      parent[v] = r;
+      atomicMin(&parent[v], r);  // monotone (see wcc_find)
       if (changed) *changed = 1;
+    }
+  }
+}
+
+// Output canonicalization. wcc_find's path-halving store races
+// non-monotonically (a stale-derived higher value can overwrite a
+// concurrently lowered entry), so a "no-change" wcc_compress pass can
+// still leave entries pointing at non-roots. This variant chases
+// READ-ONLY and writes only true static-forest roots — two clean passes
+// guarantee parent[v] == root(v) exactly.
+__global__ void wcc_finalize_kernel(uint32_t* __restrict__ parent,
+                                    uint32_t n, int* __restrict__ changed) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint32_t v = blockIdx.x * blockDim.x + threadIdx.x; v < n;
+       v += stride) {
+    uint32_t r = v;
+    uint32_t p;
+    while ((p = parent[r]) != r) r = p;
+    if (parent[v] != r) {
+      parent[v] = r;
+      *changed = 1;
     }
   }
 }
@@ -2927,21 +2955,27 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
     HIP_CHECK(hipStreamSynchronize(s));
     if (!any) break;
   }
-  if (multi) {
-    // exact agreement for output: one dense merge, then compress to the
-    // fixpoint (deterministic on identical input, so ranks converge to
-    // identical root labels)
-    I.dc.allreduce_min_u32(parent.data(), nv_pad, s);
+  // exact labels for output. Multi: one dense merge first so ranks agree.
+  // All paths: compress to the FIXPOINT — the loop can exit via
+  // nrest==0 with entries still multiple hops from their root (path
+  // halving races decide which), and the fetch below reads parent raw.
+  if (multi) I.dc.allreduce_min_u32(parent.data(), nv_pad, s);
+  {
+    int passes = 0;
     for (;;) {
       d_changed.zero(s);
-      wcc_compress_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(
+      wcc_finalize_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(
           parent.data(), nv_pad, d_changed.data());
+      ++passes;
       int ch = 0;
       HIP_CHECK(hipMemcpyAsync(&ch, d_changed.data(), 4,
                                hipMemcpyDeviceToHost, s));
       HIP_CHECK(hipStreamSynchronize(s));
       if (!ch) break;
     }
+    if (getenv("GRAPEHIP_DEBUG"))
+      fprintf(stderr, "[wcc] output fixpoint passes=%d rounds=%d\n", passes,
+              rounds);
   }
   HIP_CHECK(hipDeviceSynchronize());
   if (comm_) comm_->barrier();
