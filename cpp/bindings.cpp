@@ -430,14 +430,16 @@ PYBIND11_MODULE(_core, m) {
            py::arg("partitioner") = "segmented",
            py::arg("idxer") = "hashmap")
       .def("mutate_graph",
-           [](PyEngine& eng, PyGraph& g, arr_i64 add_src, arr_i64 add_dst,
-              std::optional<arr_f32> add_w, arr_i64 rm_src, arr_i64 rm_dst,
-              arr_i64 rm_vertices) {
+           [](PyEngine& eng, std::shared_ptr<PyGraph> gp, arr_i64 add_src,
+              arr_i64 add_dst, std::optional<arr_f32> add_w, arr_i64 rm_src,
+              arr_i64 rm_dst, arr_i64 rm_vertices) {
              // Reference parity: LoadGraphAndMutate / Mutation{add/remove
-             // edges,vertices} (loader.h:55-68, basic_fragment_mutator.h).
-             // grapehip mutates by functional rebuild: owned triples are
-             // filtered against the (replicated) removal sets, additions
-             // are merged, and the distributed builder reruns.
+             // edges,vertices} (loader.h:55-68, basic_fragment_mutator.h,
+             // mutable_edgecut_fragment.h:289-399). Edge-only deltas apply
+             // IN PLACE with cost proportional to the delta (slack-tracked
+             // rows, end-arena relocation); vertex removals and
+             // weight-introducing deltas fall back to a functional rebuild.
+             PyGraph& g = *gp;
              if (!g.frag)
                throw std::runtime_error("mutate_graph needs a host fragment");
              size_t na = add_src.size();
@@ -446,6 +448,35 @@ PYBIND11_MODULE(_core, m) {
              size_t nr = rm_src.size();
              if (static_cast<size_t>(rm_dst.size()) != nr)
                throw std::runtime_error("remove src/dst size mismatch");
+             const bool delta_ok =
+                 rm_vertices.size() == 0 &&
+                 (!add_w.has_value() || g.frag->has_weights());
+             if (delta_ok) {
+               std::vector<EdgeTriple> adds(na);
+               {
+                 auto sp = add_src.unchecked<1>();
+                 auto dp = add_dst.unchecked<1>();
+                 const float* wp = add_w ? add_w->data() : nullptr;
+                 for (size_t i = 0; i < na; ++i)
+                   adds[i] = {sp(i), dp(i), wp ? wp[i] : 1.0f};
+               }
+               std::vector<std::pair<oid_t, oid_t>> rms(nr);
+               {
+                 auto sp = rm_src.unchecked<1>();
+                 auto dp = rm_dst.unchecked<1>();
+                 for (size_t i = 0; i < nr; ++i)
+                   rms[i] = {sp(i), dp(i)};
+               }
+               py::gil_scoped_release rel;
+               g.frag->MutateDelta(eng.c(), adds, rms);
+#ifdef GRAPEHIP_WITH_HIP
+               if (eng.use_gpu) {
+                 g.frag->compact();
+                 g.dev = eng.gpu->upload(*g.frag);
+               }
+#endif
+               return gp;
+             }
              auto out = std::make_shared<PyGraph>();
              out->vm = g.vm;
              bool weighted =

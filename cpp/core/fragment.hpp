@@ -19,7 +19,9 @@
 #include <cassert>
 #include <cstring>
 #include <memory>
+#include <type_traits>
 #include <unordered_map>
+#include <utility>
 #include <vector>
 
 #include "archive.hpp"
@@ -79,23 +81,36 @@ class Fragment {
   }
 
   // --- adjacency --------------------------------------------------------
+  // Mutable mode (after an in-place Mutate): rows live at explicit
+  // (start, len) with capacity slack, relocated to an end arena when they
+  // outgrow it (reference DeMutableCSR semantics, de_mutable_csr.h).
+  // compact() squeezes back to canonical monotonic-offset form.
   struct AdjRange {
     const vid_t* dst;
     const weight_t* w;  // nullptr if unweighted
     size_t n;
   };
   AdjRange out_edges(vid_t u) const {
-    eid_t b = oe_off_[u], e = oe_off_[u + 1];
+    eid_t b = mutable_mode_ ? oe_start_[u] : oe_off_[u];
+    size_t n = mutable_mode_ ? oe_len_[u]
+                             : static_cast<size_t>(oe_off_[u + 1] - b);
     return {oe_dst_.data() + b, oe_w_.empty() ? nullptr : oe_w_.data() + b,
-            static_cast<size_t>(e - b)};
+            n};
   }
   AdjRange in_edges(vid_t u) const {
-    eid_t b = ie_off_[u], e = ie_off_[u + 1];
+    eid_t b = mutable_mode_ ? ie_start_[u] : ie_off_[u];
+    size_t n = mutable_mode_ ? ie_len_[u]
+                             : static_cast<size_t>(ie_off_[u + 1] - b);
     return {ie_dst_.data() + b, ie_w_.empty() ? nullptr : ie_w_.data() + b,
-            static_cast<size_t>(e - b)};
+            n};
   }
-  size_t out_degree(vid_t u) const { return oe_off_[u + 1] - oe_off_[u]; }
-  size_t in_degree(vid_t u) const { return ie_off_[u + 1] - ie_off_[u]; }
+  size_t out_degree(vid_t u) const {
+    return mutable_mode_ ? oe_len_[u] : oe_off_[u + 1] - oe_off_[u];
+  }
+  size_t in_degree(vid_t u) const {
+    return mutable_mode_ ? ie_len_[u] : ie_off_[u + 1] - ie_off_[u];
+  }
+  bool mutable_mode() const { return mutable_mode_; }
 
   // Raw arrays (used by the GPU uploader and serializer).
   const std::vector<eid_t>& oe_offsets() const { return oe_off_; }
@@ -355,7 +370,268 @@ class Fragment {
     return frag;
   }
 
+  // --- in-place delta mutation -------------------------------------------
+  // Reference parity: MutableEdgecutFragment::Mutate
+  // (mutable_edgecut_fragment.h:289-399) — cost proportional to the delta
+  // (plus, only when NEW outer vertices appear, one local remap sweep to
+  // keep the sorted-outer invariant the dense paths rely on). Collective.
+  // adds/removes are this rank's arbitrary slice of the delta, in oids.
+  // Unknown oids are skipped like the builder does.
+  void MutateDelta(TcpComm* comm,
+                   const std::vector<EdgeTriple>& adds,
+                   const std::vector<std::pair<oid_t, oid_t>>& removes) {
+    const VertexMap& V = *vm_;
+    const IdParser& P = parser_;
+    const bool weighted = has_weights();
+    const bool want_in = has_in_csr();
+    struct GidEdge {
+      vid_t src, dst;
+      weight_t w;
+    };
+    // 1) route deltas to owners (same shape as Build): tag removals with
+    //    a NaN-free sentinel channel — two separate streams.
+    std::vector<std::vector<GidEdge>> add_out(fnum_), add_in(fnum_);
+    std::vector<std::vector<std::pair<vid_t, vid_t>>> rm_out(fnum_),
+        rm_in(fnum_);
+    for (const EdgeTriple& e : adds) {
+      vid_t sg, dg;
+      if (!V.get_gid(e.src, &sg) || !V.get_gid(e.dst, &dg)) continue;
+      add_out[P.fid(sg)].push_back({sg, dg, e.w});
+      if (!directed_) add_out[P.fid(dg)].push_back({dg, sg, e.w});
+      else if (want_in) add_in[P.fid(dg)].push_back({dg, sg, e.w});
+    }
+    for (const auto& r : removes) {
+      vid_t sg, dg;
+      if (!V.get_gid(r.first, &sg) || !V.get_gid(r.second, &dg)) continue;
+      rm_out[P.fid(sg)].push_back({sg, dg});
+      if (!directed_) rm_out[P.fid(dg)].push_back({dg, sg});
+      else if (want_in) rm_in[P.fid(dg)].push_back({dg, sg});
+    }
+    auto exchange_vec = [&](auto& bins) {
+      using ElemT =
+          typename std::decay_t<decltype(bins[0])>::value_type;
+      std::vector<std::string> send(fnum_);
+      for (int f = 0; f < fnum_; ++f)
+        send[f].assign(reinterpret_cast<const char*>(bins[f].data()),
+                       bins[f].size() * sizeof(ElemT));
+      std::vector<std::string> recv =
+          (comm && fnum_ > 1) ? comm->exchange_all(send) : std::move(send);
+      std::vector<ElemT> all;
+      for (auto& blob : recv) {
+        size_t n = blob.size() / sizeof(ElemT);
+        const ElemT* p = reinterpret_cast<const ElemT*>(blob.data());
+        all.insert(all.end(), p, p + n);
+      }
+      return all;
+    };
+    std::vector<GidEdge> my_add_out = exchange_vec(add_out);
+    std::vector<GidEdge> my_add_in = exchange_vec(add_in);
+    std::vector<std::pair<vid_t, vid_t>> my_rm_out = exchange_vec(rm_out);
+    std::vector<std::pair<vid_t, vid_t>> my_rm_in = exchange_vec(rm_in);
+
+    // 2) new outer gids referenced by the adds
+    std::vector<vid_t> fresh;
+    for (auto* lst : {&my_add_out, &my_add_in})
+      for (const GidEdge& e : *lst)
+        if (P.fid(e.dst) != fid_ && !ovg2l_.count(e.dst))
+          fresh.push_back(e.dst);
+    std::sort(fresh.begin(), fresh.end());
+    fresh.erase(std::unique(fresh.begin(), fresh.end()), fresh.end());
+    uint64_t fresh_any = fresh.size();
+    if (comm && fnum_ > 1) fresh_any = comm->allreduce_sum(fresh_any);
+    if (fresh_any) grow_outer(comm, fresh);
+
+    // 3) enter mutable mode and apply
+    enter_mutable_mode();
+    auto apply = [&](std::vector<eid_t>& start, std::vector<vid_t>& len,
+                     std::vector<vid_t>& cap, std::vector<vid_t>& dst,
+                     std::vector<weight_t>& wts,
+                     const std::vector<std::pair<vid_t, vid_t>>& rms,
+                     const std::vector<GidEdge>& ads) {
+      for (const auto& r : rms) {
+        vid_t u = P.lid(r.first);
+        vid_t dl = P.fid(r.second) == fid_ ? P.lid(r.second)
+                                           : lookup_outer(r.second);
+        if (dl == kInvalidVid) continue;
+        eid_t b = start[u];
+        for (vid_t k = 0; k < len[u];) {
+          if (dst[b + k] == dl) {
+            dst[b + k] = dst[b + len[u] - 1];
+            if (weighted) wts[b + k] = wts[b + len[u] - 1];
+            --len[u];
+          } else {
+            ++k;
+          }
+        }
+      }
+      for (const GidEdge& e : ads) {
+        vid_t u = P.lid(e.src);
+        vid_t dl = P.fid(e.dst) == fid_ ? P.lid(e.dst)
+                                        : lookup_outer(e.dst);
+        if (dl == kInvalidVid) continue;
+        if (len[u] == cap[u]) {
+          // relocate to the end arena with doubled capacity
+          vid_t ncap = cap[u] ? cap[u] * 2 : 4;
+          eid_t nb = static_cast<eid_t>(dst.size());
+          dst.resize(nb + ncap);
+          if (weighted) wts.resize(nb + ncap);
+          std::memcpy(dst.data() + nb, dst.data() + start[u],
+                      len[u] * sizeof(vid_t));
+          if (weighted)
+            std::memcpy(wts.data() + nb, wts.data() + start[u],
+                        len[u] * sizeof(weight_t));
+          start[u] = nb;
+          cap[u] = ncap;
+        }
+        dst[start[u] + len[u]] = dl;
+        if (weighted) wts[start[u] + len[u]] = e.w;
+        ++len[u];
+      }
+    };
+    apply(oe_start_, oe_len_, oe_cap_, oe_dst_, oe_w_, my_rm_out,
+          my_add_out);
+    if (want_in)
+      apply(ie_start_, ie_len_, ie_cap_, ie_dst_, ie_w_, my_rm_in,
+            my_add_in);
+
+    // 4) refresh global counts
+    uint64_t local = 0;
+    for (vid_t v = 0; v < ivnum_; ++v) local += oe_len_[v];
+    total_edges_ = (comm && fnum_ > 1) ? comm->allreduce_sum(local) : local;
+    input_edges_ = directed_ ? total_edges_ : total_edges_ / 2;
+  }
+
+  // Squeeze mutable-mode slack back into canonical monotonic offsets
+  // (called before checkpointing or a GPU upload).
+  void compact() {
+    if (!mutable_mode_) return;
+    auto squeeze = [&](std::vector<eid_t>& off, std::vector<vid_t>& dst,
+                       std::vector<weight_t>& wts, std::vector<eid_t>& start,
+                       std::vector<vid_t>& len) {
+      std::vector<eid_t> noff(ivnum_ + 1);
+      eid_t run = 0;
+      for (vid_t v = 0; v < ivnum_; ++v) {
+        noff[v] = run;
+        run += len[v];
+      }
+      noff[ivnum_] = run;
+      std::vector<vid_t> ndst(run);
+      std::vector<weight_t> nw(wts.empty() ? 0 : run);
+      parallel_for(0, static_cast<size_t>(ivnum_), [&](size_t v) {
+        std::memcpy(ndst.data() + noff[v], dst.data() + start[v],
+                    len[v] * sizeof(vid_t));
+        if (!wts.empty())
+          std::memcpy(nw.data() + noff[v], wts.data() + start[v],
+                      len[v] * sizeof(weight_t));
+      }, 256);
+      off = std::move(noff);
+      dst = std::move(ndst);
+      wts = std::move(nw);
+    };
+    squeeze(oe_off_, oe_dst_, oe_w_, oe_start_, oe_len_);
+    if (has_in_csr()) squeeze(ie_off_, ie_dst_, ie_w_, ie_start_, ie_len_);
+    oe_start_.clear();
+    oe_len_.clear();
+    oe_cap_.clear();
+    ie_start_.clear();
+    ie_len_.clear();
+    ie_cap_.clear();
+    mutable_mode_ = false;
+  }
+
  private:
+  vid_t lookup_outer(vid_t gid) const {
+    auto it = ovg2l_.find(gid);
+    return it == ovg2l_.end() ? kInvalidVid : it->second;
+  }
+
+  void enter_mutable_mode() {
+    if (mutable_mode_) return;
+    auto init = [&](const std::vector<eid_t>& off, std::vector<eid_t>& start,
+                    std::vector<vid_t>& len, std::vector<vid_t>& cap) {
+      start.resize(ivnum_);
+      len.resize(ivnum_);
+      cap.resize(ivnum_);
+      for (vid_t v = 0; v < ivnum_; ++v) {
+        start[v] = off[v];
+        len[v] = static_cast<vid_t>(off[v + 1] - off[v]);
+        cap[v] = len[v];
+      }
+    };
+    init(oe_off_, oe_start_, oe_len_, oe_cap_);
+    if (has_in_csr()) init(ie_off_, ie_start_, ie_len_, ie_cap_);
+    mutable_mode_ = true;
+  }
+
+  // Insert new outer gids while keeping ovgid_ sorted (the contiguity
+  // invariant every dense path relies on): one local remap sweep of the
+  // stored dst lids plus a mirror-info refresh — no global rebuild.
+  void grow_outer(TcpComm* comm, const std::vector<vid_t>& fresh) {
+    std::vector<vid_t> merged(ovgid_.size() + fresh.size());
+    std::merge(ovgid_.begin(), ovgid_.end(), fresh.begin(), fresh.end(),
+               merged.begin());
+    // old outer lid -> new outer lid
+    std::vector<vid_t> remap(ovnum_);
+    {
+      size_t j = 0;
+      for (size_t i = 0; i < ovgid_.size(); ++i) {
+        while (merged[j] != ovgid_[i]) ++j;
+        remap[i] = ivnum_ + static_cast<vid_t>(j);
+      }
+    }
+    auto sweep = [&](std::vector<vid_t>& dst, const std::vector<eid_t>& start,
+                     const std::vector<vid_t>& len) {
+      if (mutable_mode_) {
+        parallel_for(0, static_cast<size_t>(ivnum_), [&](size_t v) {
+          eid_t b = start[v];
+          for (vid_t k = 0; k < len[v]; ++k)
+            if (dst[b + k] >= ivnum_) dst[b + k] = remap[dst[b + k] - ivnum_];
+        }, 256);
+      } else {
+        parallel_for(0, dst.size(), [&](size_t i) {
+          if (dst[i] >= ivnum_) dst[i] = remap[dst[i] - ivnum_];
+        }, 8192);
+      }
+    };
+    sweep(oe_dst_, oe_start_, oe_len_);
+    if (has_in_csr()) sweep(ie_dst_, ie_start_, ie_len_);
+    ovgid_ = std::move(merged);
+    ovnum_ = static_cast<vid_t>(ovgid_.size());
+    ovg2l_.clear();
+    ovg2l_.reserve(ovnum_ * 2);
+    for (vid_t i = 0; i < ovnum_; ++i)
+      ovg2l_.emplace(ovgid_[i], ivnum_ + i);
+    outer_range_.assign(fnum_, {0, 0});
+    {
+      vid_t i = 0;
+      for (int f = 0; f < fnum_; ++f) {
+        vid_t b = i;
+        while (i < ovnum_ &&
+               parser_.fid(ovgid_[i]) == static_cast<fid_t>(f))
+          ++i;
+        outer_range_[f] = {ivnum_ + b, ivnum_ + i};
+      }
+    }
+    // mirror info depends on every peer's outer order — refresh it
+    mirrors_.assign(fnum_, {});
+    if (comm && fnum_ > 1) {
+      std::vector<std::string> send(fnum_);
+      for (int f = 0; f < fnum_; ++f) {
+        auto [b, e] = outer_range_[f];
+        send[f].assign(
+            reinterpret_cast<const char*>(ovgid_.data() + (b - ivnum_)),
+            (e - b) * sizeof(vid_t));
+      }
+      auto recv = comm->exchange_all(send);
+      for (int f = 0; f < fnum_; ++f) {
+        size_t n = recv[f].size() / sizeof(vid_t);
+        mirrors_[f].resize(n);
+        const vid_t* g = reinterpret_cast<const vid_t*>(recv[f].data());
+        for (size_t i = 0; i < n; ++i) mirrors_[f][i] = parser_.lid(g[i]);
+      }
+    }
+  }
+
   fid_t fid_ = 0;
   int fnum_ = 1;
   IdParser parser_;
@@ -370,6 +646,10 @@ class Fragment {
   std::unordered_map<vid_t, vid_t> ovg2l_;
   std::vector<std::pair<vid_t, vid_t>> outer_range_;
   std::vector<std::vector<vid_t>> mirrors_;
+  // mutable-mode row bookkeeping (empty until the first MutateDelta)
+  bool mutable_mode_ = false;
+  std::vector<eid_t> oe_start_, ie_start_;
+  std::vector<vid_t> oe_len_, oe_cap_, ie_len_, ie_cap_;
 };
 
 }  // namespace grapehip

@@ -79,8 +79,8 @@ inline std::string frag_path(const std::string& prefix, fid_t fid) {
 
 }  // namespace ser
 
-inline void serialize_graph(const Fragment& frag,
-                            const std::string& prefix) {
+inline void serialize_graph(Fragment& frag, const std::string& prefix) {
+  frag.compact();  // mutable-mode slack is not checkpointed
   ser::Writer w(ser::frag_path(prefix, frag.fid()));
   w.pod(ser::kMagic);
   w.pod(ser::kVersion);
