@@ -5223,7 +5223,11 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
 
   DeviceBuffer<unsigned long long> Tcnt(nv_pad);
   Tcnt.zero(s);
-  uint32_t kHeavyThresh = 96;  // min-side size above which an edge defers
+  // min-side size above which an edge defers to the wave-per-edge pass;
+  // 96 -> 48 after the two-box sweep + clean full-scale re-measurement
+  // (profiles/r01_lcc_heavy_thresh_sweep.md, r02_lcc_heavy48_datagen90:
+  // 2.62 s -> 2.50 s)
+  uint32_t kHeavyThresh = 48;
   if (const char* ht = getenv("GRAPEHIP_LCC_HEAVY"))
     kHeavyThresh = static_cast<uint32_t>(atoi(ht));
   DeviceBuffer<unsigned long long> heavy_q;
@@ -5552,7 +5556,11 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
   // wave-per-edge pass (power-law tail otherwise stalls single lanes)
   Tcnt.resize(nv_pad);
   Tcnt.zero(s);
-  uint32_t kHeavyThresh = 96;  // min-side size above which an edge defers
+  // min-side size above which an edge defers to the wave-per-edge pass;
+  // 96 -> 48 after the two-box sweep + clean full-scale re-measurement
+  // (profiles/r01_lcc_heavy_thresh_sweep.md, r02_lcc_heavy48_datagen90:
+  // 2.62 s -> 2.50 s)
+  uint32_t kHeavyThresh = 48;
   if (const char* ht = getenv("GRAPEHIP_LCC_HEAVY"))
     kHeavyThresh = static_cast<uint32_t>(atoi(ht));
   const bool nowit = getenv("GRAPEHIP_LCC_NOWIT") != nullptr;

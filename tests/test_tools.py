@@ -44,3 +44,14 @@ def test_verify_output_wcc_isomorphism(tmp_path):
                         str(out), str(broken)], cwd=REPO,
                        capture_output=True, text=True)
     assert r.returncode != 0
+
+
+def test_app_matrix_smoke():
+    # app_tests.sh-equivalent matrix (reference misc/app_tests.sh): a
+    # reduced CPU matrix — full sweep via `python tools/app_matrix.py`
+    r = subprocess.run([sys.executable, "tools/app_matrix.py",
+                        "--apps", "bfs,wcc", "--worlds", "1,2",
+                        "--skip-serialize", "--port", "29890"],
+                       cwd=REPO, capture_output=True, text=True,
+                       timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
