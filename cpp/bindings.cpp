@@ -252,7 +252,8 @@ DevIdMap dev_id_map(const PyGraph& g) {
 }
 
 py::dict gpu_dict(const GpuRunResult& r, const PyGraph& pg, bool is_i64,
-                  bool with_values, bool labels_are_ids = false) {
+                  bool with_values, bool labels_are_ids = false,
+                  const std::vector<int64_t>* label_lut = nullptr) {
   const DeviceGraph& g = *pg.dev;
   py::dict out;
   out["rounds"] = r.rounds;
@@ -264,18 +265,34 @@ py::dict gpu_dict(const GpuRunResult& r, const PyGraph& pg, bool is_i64,
     uint32_t n = m.frag ? m.frag->ivnum() : g.owned();
     py::array_t<int64_t> oids(n);
     auto* p = oids.mutable_data();
-    for (uint32_t i = 0; i < n; ++i)
-      p[i] = m.frag ? m.frag->lid2oid(i)
-                    : static_cast<int64_t>(g.v_begin) + i;
+    if (!m.frag && g.permuted) {
+      // hub renumbering: row i holds old id inv[v_begin+i] (cached)
+      DeviceGraph& gm = const_cast<DeviceGraph&>(g);
+      if (gm.inv_host.size() != n) {
+        gm.inv_host.resize(n);
+        if (hipMemcpy(gm.inv_host.data(), gm.inv.data() + gm.v_begin,
+                      static_cast<size_t>(n) * 4,
+                      hipMemcpyDeviceToHost) != hipSuccess)
+          throw std::runtime_error("inv fetch failed");
+      }
+      for (uint32_t i = 0; i < n; ++i) p[i] = gm.inv_host[i];
+    } else {
+      for (uint32_t i = 0; i < n; ++i)
+        p[i] = m.frag ? m.frag->lid2oid(i)
+                      : static_cast<int64_t>(g.v_begin) + i;
+    }
     out["oids"] = oids;
     if (is_i64) {
       py::array_t<int64_t> vals(n);
       auto* q = vals.mutable_data();
       for (uint32_t i = 0; i < n; ++i)
-        q[i] = (labels_are_ids && m.frag &&
-                r.i64[i] != std::numeric_limits<int64_t>::max())
-                   ? m.to_oid(r.i64[i])
-                   : r.i64[i];
+        q[i] = (label_lut && r.i64[i] >= 0 &&
+                static_cast<size_t>(r.i64[i]) < label_lut->size())
+                   ? (*label_lut)[r.i64[i]]
+                   : ((labels_are_ids && m.frag &&
+                       r.i64[i] != std::numeric_limits<int64_t>::max())
+                          ? m.to_oid(r.i64[i])
+                          : r.i64[i]);
       out["values"] = vals;
     } else {
       py::array_t<double> vals(n);
@@ -716,7 +733,24 @@ PYBIND11_MODULE(_core, m) {
                  py::gil_scoped_release rel;
                  r = eng.gpu->cdlp(*g.dev, iters, values);
                }
-               return gpu_dict(r, g, true, values, /*labels_are_ids=*/true);
+               // non-identity maps run CDLP in global sorted-OID order
+               // space (reference tie-break semantics); translate back
+               std::vector<int64_t> lut;
+               if (g.frag &&
+                   g.frag->vm().idxer() != IdxerKind::kIdentity) {
+                 const VertexMap& vm = g.frag->vm();
+                 const IdParser& P = g.frag->parser();
+                 lut.reserve(vm.total_vertices());
+                 for (int f = 0; f < vm.fnum(); ++f) {
+                   vid_t nn = vm.frag_vnum(static_cast<fid_t>(f));
+                   for (vid_t l = 0; l < nn; ++l)
+                     lut.push_back(
+                         vm.get_oid(P.gid(static_cast<fid_t>(f), l)));
+                 }
+                 std::sort(lut.begin(), lut.end());
+               }
+               return gpu_dict(r, g, true, values, /*labels_are_ids=*/true,
+                               lut.empty() ? nullptr : &lut);
              }
 #endif
              CDLPApp app;

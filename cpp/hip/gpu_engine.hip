@@ -316,6 +316,86 @@ __global__ void gen_edges_kernel(uint64_t ne, uint64_t seed, int scale,
   }
 }
 
+// ===========================================================================
+// Hub-clustering renumber. RMAT/power-law hot vertices are the ids with
+// few set bits — scattered across the id range, so the pull kernels'
+// contrib/label gathers fetch a cold 64B line per 4B value. Renumbering
+// each slice by descending (capped) degree packs the hot ~1M vertices
+// into a contiguous L2-resident prefix (measured: top-1M ids carry ~53%
+// of endpoint references at datagen-9_0 shape, but only 8% sit in any
+// contiguous 1M window). Owner-preserving: new id = v_begin + position
+// within the slice, so slicing, halos and collectives are untouched.
+// ===========================================================================
+
+constexpr int kRenumBuckets = 4096;  // degree-capped counting sort
+
+__global__ void renum_hist_kernel(const uint32_t* __restrict__ deg,
+                                  uint32_t n,
+                                  unsigned long long* __restrict__ hist) {
+  __shared__ uint32_t h[kRenumBuckets];
+  for (int b = threadIdx.x; b < kRenumBuckets; b += blockDim.x) h[b] = 0;
+  __syncthreads();
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint32_t k = deg[i];
+    atomicAdd(&h[k < kRenumBuckets ? k : kRenumBuckets - 1], 1u);
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < kRenumBuckets; b += blockDim.x)
+    if (h[b]) atomicAdd(&hist[b], static_cast<unsigned long long>(h[b]));
+}
+
+// block-synchronous 3-phase scatter: LDS-aggregated cursor claims keep the
+// hot low-degree buckets off the global atomic units
+__global__ void renum_scatter_kernel(const uint32_t* __restrict__ deg,
+                                     uint32_t n, uint32_t v_begin,
+                                     unsigned long long* __restrict__ cur,
+                                     uint32_t* __restrict__ perm) {
+  __shared__ uint32_t h[kRenumBuckets];
+  __shared__ unsigned long long base[kRenumBuckets];
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint32_t chunk = blockIdx.x * blockDim.x; chunk < n;
+       chunk += stride) {
+    uint32_t i = chunk + threadIdx.x;
+    uint32_t k = kRenumBuckets;
+    if (i < n) {
+      uint32_t d = deg[i];
+      k = d < kRenumBuckets ? d : kRenumBuckets - 1;
+    }
+    for (int b = threadIdx.x; b < kRenumBuckets; b += blockDim.x) h[b] = 0;
+    __syncthreads();
+    uint32_t mypos = 0;
+    if (k < kRenumBuckets) mypos = atomicAdd(&h[k], 1u);
+    __syncthreads();
+    for (int b = threadIdx.x; b < kRenumBuckets; b += blockDim.x)
+      if (h[b])
+        base[b] = atomicAdd(&cur[b], static_cast<unsigned long long>(h[b]));
+    __syncthreads();
+    if (k < kRenumBuckets)
+      perm[v_begin + i] =
+          v_begin + static_cast<uint32_t>(base[k] + mypos);
+    __syncthreads();
+  }
+}
+
+__global__ void remap_ids_kernel(uint32_t* __restrict__ ids, uint64_t n,
+                                 const uint32_t* __restrict__ perm) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * blockDim.x +
+                    threadIdx.x;
+       i < n; i += stride)
+    ids[i] = perm[ids[i]];
+}
+
+__global__ void inv_scatter_kernel(const uint32_t* __restrict__ perm,
+                                   uint32_t n, uint32_t* __restrict__ inv) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    inv[perm[i]] = i;
+}
+
 __global__ void count_deg_kernel(const uint32_t* src, uint64_t n,
                                  uint32_t v_begin, uint32_t* deg) {
   size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
@@ -1262,6 +1342,140 @@ __global__ void pr_pull_large_kernel(const uint64_t* __restrict__ off,
   }
 }
 
+// ===========================================================================
+// PageRank tiled pull. The row-major pull gathers contrib[dst] across the
+// whole vertex range: at datagen-9_0 scale every 4-byte gather fetches a
+// distinct 64-byte line (~6% line use, r01 PMC: 2.3 TB/s fetch, >80%
+// latency stall). Regrouping edges by DST TILE (4096 vertices = 16 KB of
+// contribs) makes consecutive gathers hit the same L2-resident window, so
+// per-iteration traffic drops to ~(8 B/edge edge stream + one pass over
+// contrib + L2-resident acc atomics). Rows' dst lists are sorted, so each
+// row contributes contiguous per-tile runs; runs are placed contiguously
+// inside their tile bucket (order across rows is irrelevant — each run
+// has constant src, and a wave-segmented sum emits one fp64 atomic per
+// run per 64-edge window). Built once per graph, cached in HBM3E
+// (8 B/stored edge).
+// ===========================================================================
+
+constexpr int kPrTileBits = 12;  // 4096 vertices/tile: 16 KB fp32 window
+
+// count (out==nullptr) or place (out!=nullptr) per-row tile runs.
+// thread-per-row variant for short rows.
+__global__ void pr_tile_runs_small_kernel(
+    const uint64_t* __restrict__ off, const uint32_t* __restrict__ dst,
+    const uint32_t* __restrict__ rows, uint64_t nrows, uint32_t v_begin,
+    unsigned long long* __restrict__ cnt, unsigned long long* __restrict__ out) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * blockDim.x +
+                    threadIdx.x;
+       i < nrows; i += stride) {
+    uint32_t r = rows[i];
+    uint64_t b = off[r], e = off[r + 1];
+    uint64_t src64 = static_cast<uint64_t>(v_begin + r) << 32;
+    uint64_t k = b;
+    while (k < e) {
+      uint32_t t = dst[k] >> kPrTileBits;
+      uint64_t j = k + 1;
+      while (j < e && (dst[j] >> kPrTileBits) == t) ++j;
+      unsigned long long pos = atomicAdd(&cnt[t], j - k);
+      if (out)
+        for (uint64_t q = k; q < j; ++q)
+          out[pos + (q - k)] = src64 | dst[q];
+      k = j;
+    }
+  }
+}
+
+// wave-per-row variant: lanes stride the row; run boundaries via ballot.
+__global__ void pr_tile_runs_wave_kernel(
+    const uint64_t* __restrict__ off, const uint32_t* __restrict__ dst,
+    const uint32_t* __restrict__ rows, uint64_t nrows, uint32_t v_begin,
+    unsigned long long* __restrict__ cnt, unsigned long long* __restrict__ out) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wpb = kBlock / kWave;
+  size_t wstride = static_cast<size_t>(gridDim.x) * wpb;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * wpb + wid; i < nrows;
+       i += wstride) {
+    uint32_t r = rows[i];
+    uint64_t b = off[r], e = off[r + 1];
+    uint64_t src64 = static_cast<uint64_t>(v_begin + r) << 32;
+    for (uint64_t base = b; base < e; base += kWave) {
+      uint64_t k = base + lane;
+      bool act = k < e;
+      uint32_t d = act ? dst[k] : 0xFFFFFFFFu;
+      uint32_t t = d >> kPrTileBits;
+      // head: first lane of window, or tile differs from previous lane's
+      uint32_t tprev = __shfl_up(t, 1, 64);
+      bool head = act && (lane == 0 || tprev != t ||
+                          (base + lane - 1) < b);
+      // window runs may continue the previous window's run: lane0 head
+      // still claims fresh space — runs split at window boundaries,
+      // which only adds an extra atomic, not an error
+      unsigned long long hb = __ballot(head);
+      unsigned long long pos = 0;
+      if (head && act) {
+        // run length = next head (or window end / row end) - k
+        unsigned long long later = hb >> 1 >> lane;  // heads after me
+        int next = later ? lane + 1 + __ffsll(later) - 1 : 64;
+        uint64_t rend = base + next;
+        if (rend > e) rend = e;
+        pos = atomicAdd(&cnt[t], rend - k);
+      }
+      if (out && act) {
+        // my position: broadcast pos from my run's head lane
+        unsigned long long mine =
+            hb & ((lane == 63) ? ~0ull : ((1ull << (lane + 1)) - 1));
+        int myhead = 63 - __clzll(mine);
+        unsigned long long hpos =
+            __shfl(pos, myhead, 64);
+        out[hpos + (lane - myhead)] = src64 | d;
+      }
+    }
+  }
+}
+
+// one pass over the tiled edge stream: gather contrib from the L2-hot
+// window, wave-segmented sum per constant-src run, fp64 atomic per run.
+__global__ void pr_tiled_pull_kernel(
+    const unsigned long long* __restrict__ tiles, uint64_t lo, uint64_t hi,
+    const float* __restrict__ contrib, double* __restrict__ acc) {
+  const int lane = threadIdx.x & 63;
+  uint64_t n = hi - lo;
+  if (!n) return;
+  // contiguous chunk per wave (keeps each wave inside one tile window)
+  uint64_t waves = (static_cast<uint64_t>(gridDim.x) * blockDim.x) / kWave;
+  uint64_t wid = (static_cast<uint64_t>(blockIdx.x) * blockDim.x +
+                  threadIdx.x) / kWave;
+  uint64_t chunk = (n + waves - 1) / waves;
+  uint64_t wb = lo + wid * chunk;
+  uint64_t we = wb + chunk < hi ? wb + chunk : hi;
+  for (uint64_t base = wb; base < we; base += kWave) {
+    uint64_t k = base + lane;
+    bool act = k < we;
+    unsigned long long rec = act ? tiles[k] : ~0ull;
+    uint32_t srcv = static_cast<uint32_t>(rec >> 32);
+    double c = act ? static_cast<double>(contrib[static_cast<uint32_t>(rec)])
+                   : 0.0;
+    uint32_t sprev = __shfl_up(srcv, 1, 64);
+    bool head = (lane == 0) || sprev != srcv;
+    unsigned long long hb = __ballot(head);
+    // segment start = highest head lane <= me
+    unsigned long long mine =
+        hb & ((lane == 63) ? ~0ull : ((1ull << (lane + 1)) - 1));
+    int seg = 63 - __clzll(mine);
+    double sum = c;
+#pragma unroll
+    for (int d = 1; d < 64; d <<= 1) {
+      double up = __shfl_up(sum, d, 64);
+      if (lane - d >= seg) sum += up;
+    }
+    bool tail = (lane == 63) || ((hb >> (lane + 1)) & 1ull);
+    if (tail && act && srcv != 0xFFFFFFFFu) atomicAdd(&acc[srcv], sum);
+  }
+}
+
+
 // ---------------------------------------------------------------------------
 // Direction-optimizing BFS pull kernels (reference cuda/bfs/bfs.h :206-260):
 // sweep unvisited owned rows looking for a parent at the current level.
@@ -1807,6 +2021,55 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
   if (getenv("GRAPEHIP_DEBUG"))
     fprintf(stderr, "[gen] n_local=%llu\n", n_local);
 
+  // hub-clustering renumber: order each slice by descending appearance
+  // count (out-stream src count = how often an id is REFERENCED by pull
+  // gathers: full degree when undirected, out-degree when directed).
+  // Counting sort over capped degrees; within-bucket order is arbitrary.
+  const bool renumber =
+      !(getenv("GRAPEHIP_RENUMBER") &&
+        atoi(getenv("GRAPEHIP_RENUMBER")) == 0);
+  if (renumber) {
+    uint32_t nv_pad = padded_nv(*g, world_);
+    DeviceBuffer<uint32_t> deg(owned ? owned : 1);
+    deg.zero(s);
+    if (n_local)
+      count_deg_kernel<<<grid_for(n_local), kBlock, 0, s>>>(
+          e_src.data(), n_local, g->v_begin, deg.data());
+    DeviceBuffer<unsigned long long> hist(kRenumBuckets);
+    hist.zero(s);
+    if (owned)
+      renum_hist_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+          deg.data(), owned, hist.data());
+    auto h = hist.download(s);
+    // descending bucket offsets (bucket 4095 = hottest, first)
+    std::vector<unsigned long long> cur_h(kRenumBuckets);
+    unsigned long long run = 0;
+    for (int b = kRenumBuckets - 1; b >= 0; --b) {
+      cur_h[b] = run;
+      run += h[b];
+    }
+    DeviceBuffer<unsigned long long> cur;
+    cur.upload(cur_h, s);
+    g->perm.resize(nv_pad);
+    iota_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(g->perm.data(), 0,
+                                                    nv_pad);
+    if (owned)
+      renum_scatter_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+          deg.data(), owned, g->v_begin, cur.data(), g->perm.data());
+    if (world_ > 1)
+      impl_->dc.allgather32(g->perm.data(), slice, false, s);
+    if (n_local) {
+      remap_ids_kernel<<<grid_for(n_local), kBlock, 0, s>>>(
+          e_src.data(), n_local, g->perm.data());
+      remap_ids_kernel<<<grid_for(n_local), kBlock, 0, s>>>(
+          e_dst.data(), n_local, g->perm.data());
+    }
+    g->inv.resize(nv_pad);
+    inv_scatter_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(
+        g->perm.data(), nv_pad, g->inv.data());
+    g->permuted = true;
+  }
+
   build_csr_from_coo(e_src, e_dst, e_w, n_local, g->v_begin, owned, weighted,
                      g->oe_off, g->oe_dst, g->oe_w, s, impl_->scan);
   sort_csr_rows(g->oe_off, g->oe_dst, weighted ? &g->oe_w : nullptr, owned,
@@ -1845,6 +2108,12 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
                              s));
     HIP_CHECK(hipStreamSynchronize(s));
     if (n_in > est) throw std::runtime_error("gen_synthetic: in overflow");
+    if (g->permuted && n_in) {
+      remap_ids_kernel<<<grid_for(n_in), kBlock, 0, s>>>(
+          e_src.data(), n_in, g->perm.data());
+      remap_ids_kernel<<<grid_for(n_in), kBlock, 0, s>>>(
+          e_dst.data(), n_in, g->perm.data());
+    }
     build_csr_from_coo(e_src, e_dst, e_w, n_in, g->v_begin, owned, weighted,
                        g->ie_off, g->ie_dst, g->ie_w, s, impl_->scan);
     sort_csr_rows(g->ie_off, g->ie_dst, weighted ? &g->ie_w : nullptr,
@@ -1943,6 +2212,28 @@ std::unique_ptr<DeviceGraph> GpuContext::upload(const Fragment& frag) {
   if (g->has_in)
     sort_csr_rows(g->ie_off, g->ie_dst, g->weighted ? &g->ie_w : nullptr,
                   owned_rows, s, impl_->scan);
+  if (!identity) {
+    // CDLP tie-breaks compare label VALUES; the reference's labels are
+    // oids, so under dense renumbering the comparison space must be the
+    // global sorted-OID order (vertex map is replicated — every rank
+    // derives the identical table)
+    const VertexMap& vm = frag.vm();
+    const IdParser& P = frag.parser();
+    std::vector<std::pair<oid_t, uint32_t>> pairs;
+    pairs.reserve(vm.total_vertices());
+    for (int f = 0; f < vm.fnum(); ++f) {
+      vid_t n = vm.frag_vnum(static_cast<fid_t>(f));
+      for (vid_t l = 0; l < n; ++l)
+        pairs.emplace_back(
+            vm.get_oid(P.gid(static_cast<fid_t>(f), l)),
+            static_cast<uint32_t>(static_cast<uint64_t>(f) * slice + l));
+    }
+    std::sort(pairs.begin(), pairs.end());
+    std::vector<uint32_t> order(padded_nv(*g, world_), 0xFFFFFFFFu);
+    for (uint32_t i = 0; i < pairs.size(); ++i)
+      order[pairs[i].second] = i;
+    g->oid_order.upload(order, s);
+  }
   HIP_CHECK(hipStreamSynchronize(s));
   return g;
 }
@@ -1973,6 +2264,69 @@ static void ensure_buckets(DeviceGraph& g, hipStream_t s) {
   g.n_mid = h[1];
   g.n_large = h[2];
   g.buckets_built = true;
+}
+
+static void ensure_pr_tiles(DeviceGraph& g, GpuContext::Impl& I,
+                            hipStream_t s) {
+  if (g.pr_tiles_built) return;
+  const uint64_t* pull_off = !g.directed ? g.oe_off.data() : g.ie_off.data();
+  const uint32_t* pull_dst = !g.directed ? g.oe_dst.data() : g.ie_dst.data();
+  uint64_t ne = !g.directed ? g.oe_dst.size() : g.ie_dst.size();
+  uint32_t ntiles =
+      (g.nv_global + (1u << kPrTileBits) - 1) >> kPrTileBits;
+  // memory gate: 8 B/edge for the stream + counters; fall back to the
+  // row-major pull when HBM is tight (huge 1B+-vertex graphs)
+  size_t free_b = 0, total_b = 0;
+  HIP_CHECK(hipMemGetInfo(&free_b, &total_b));
+  if (free_b < ne * 8 + (static_cast<size_t>(ntiles) + 2) * 16 + (512u << 20)) {
+    g.pr_tiles_built = true;  // decided: stay on fallback
+    g.pr_ntiles = 0;
+    return;
+  }
+  DeviceBuffer<unsigned long long> cnt(ntiles + 1);
+  cnt.zero(s);
+  auto runs = [&](unsigned long long* c, unsigned long long* out) {
+    if (g.n_small)
+      pr_tile_runs_small_kernel<<<grid_for(g.n_small), kBlock, 0, s>>>(
+          pull_off, pull_dst, g.rows_small.data(), g.n_small, g.v_begin, c,
+          out);
+    if (g.n_mid)
+      pr_tile_runs_wave_kernel<<<grid_for(g.n_mid * kWave), kBlock, 0, s>>>(
+          pull_off, pull_dst, g.rows_mid.data(), g.n_mid, g.v_begin, c, out);
+    if (g.n_large)
+      pr_tile_runs_wave_kernel<<<std::min<uint64_t>(g.n_large, kMaxGrid),
+                                 kBlock, 0, s>>>(
+          pull_off, pull_dst, g.rows_large.data(), g.n_large, g.v_begin, c,
+          out);
+  };
+  runs(cnt.data(), nullptr);
+  g.pr_tile_off.resize(static_cast<size_t>(ntiles) + 1);
+  exclusive_scan(cnt.data(), g.pr_tile_off.data(), ntiles, s, I.scan);
+  g.pr_tiles.resize(ne ? ne : 1);
+  // cursors = copy of offsets
+  DeviceBuffer<unsigned long long> cur(ntiles + 1);
+  HIP_CHECK(hipMemcpyAsync(cur.data(), g.pr_tile_off.data(),
+                           (static_cast<size_t>(ntiles) + 1) * 8,
+                           hipMemcpyDeviceToDevice, s));
+  runs(cur.data(), g.pr_tiles.data());
+  // host copies of the slice-boundary tile offsets for the phase split
+  uint32_t t0 = (g.v_begin + (1u << kPrTileBits) - 1) >> kPrTileBits;
+  uint32_t t1 = g.v_end >> kPrTileBits;
+  if (t1 < t0) t1 = t0;
+  if (t1 > ntiles) t1 = ntiles;
+  uint64_t v[3] = {0, 0, 0};
+  HIP_CHECK(hipMemcpyAsync(&v[0], g.pr_tile_off.data() + t0, 8,
+                           hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipMemcpyAsync(&v[1], g.pr_tile_off.data() + t1, 8,
+                           hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipMemcpyAsync(&v[2], g.pr_tile_off.data() + ntiles, 8,
+                           hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipStreamSynchronize(s));
+  g.pr_own_lo = v[0];
+  g.pr_own_hi = v[1];
+  g.pr_total = v[2];
+  g.pr_ntiles = ntiles;
+  g.pr_tiles_built = true;
 }
 
 // ---------------------------------------------------------------------------
@@ -2249,6 +2603,37 @@ static void mirror_sync_end(GpuContext::Impl& I, DeviceGraph& g,
   ref_sync_end(I, g.mr_recv_idx, g.mr_recv_off, g.mr_recvbuf, state4, s);
 }
 
+// old-id -> renumbered-id for API source arguments
+static uint32_t map_source(DeviceGraph& g, int64_t source, hipStream_t s) {
+  uint32_t v = static_cast<uint32_t>(source);
+  if (!g.permuted || v >= g.perm.size()) return v;
+  uint32_t out = 0;
+  HIP_CHECK(hipMemcpyAsync(&out, g.perm.data() + v, 4,
+                           hipMemcpyDeviceToHost, s));
+  HIP_CHECK(hipStreamSynchronize(s));
+  return out;
+}
+
+// canonical WCC labels under renumbering: component label = min OLD id
+__global__ void wcc_minold_kernel(const uint32_t* __restrict__ parent,
+                                  const uint32_t* __restrict__ inv,
+                                  uint32_t n, uint32_t* __restrict__ m) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint32_t v = blockIdx.x * blockDim.x + threadIdx.x; v < n;
+       v += stride)
+    atomicMin(&m[parent[v]], inv[v]);
+}
+
+__global__ void wcc_relabel_kernel(const uint32_t* __restrict__ parent,
+                                   const uint32_t* __restrict__ m,
+                                   uint32_t v_begin, uint32_t n,
+                                   uint32_t* __restrict__ out) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = m[parent[v_begin + i]];
+}
+
 // generic frontier expansion helper: scan degrees then CM-expand
 template <typename Op>
 void expand_frontier(GpuContext::Impl& I, const DevGraphView& view,
@@ -2319,7 +2704,7 @@ GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source,
     I.halo_cnt.zero(s);
     I.halo_bm.zero(s);
   }
-  uint32_t src = static_cast<uint32_t>(source);
+  uint32_t src = map_source(g, source, s);
   if (src >= g.v_begin && src < g.v_end)
     bfs_seed_kernel<<<1, 1, 0, s>>>(depth.data(), src, g.v_begin,
                                     next_bm.data());
@@ -2483,7 +2868,7 @@ GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
     I.halo_cnt.zero(s);
     I.halo_bm.zero(s);
   }
-  uint32_t src = static_cast<uint32_t>(source);
+  uint32_t src = map_source(g, source, s);
   if (src >= g.v_begin && src < g.v_end)
     sssp_seed_kernel<<<1, 1, 0, s>>>(dist.data(), src, g.v_begin,
                                      near_bm.data());
@@ -2613,6 +2998,17 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
       (!g.directed ? g.oe_dst.data() : g.ie_dst.data());
   if (pull) ensure_buckets(g, s);
   if (pull && multi) ensure_mirrors(I, comm_, g, rank_, world_, s);
+  // dst-tiled pull stream: opt-in (GRAPEHIP_PR_TILED=1). Measured at
+  // datagen-9_0 shape it loses to the row pull (~2B short per-src runs
+  // mean one fp64 L2 atomic per edge, which caps at ~40 G/s — slower
+  // than the gather traffic it saves); kept for dense-graph shapes where
+  // runs are long. The hub-clustering renumber is the default fix for
+  // gather locality instead.
+  const bool want_tiled =
+      pull && getenv("GRAPEHIP_PR_TILED") &&
+      atoi(getenv("GRAPEHIP_PR_TILED")) != 0;
+  if (want_tiled) ensure_pr_tiles(g, *impl_, s);
+  const bool tiled = want_tiled && g.pr_ntiles > 0;
 
   // working set lives on the DeviceGraph: the cached hipGraph bakes these
   // pointers (locals would dangle across calls — GPU fault under realloc)
@@ -2694,7 +3090,38 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
     pr_contrib_kernel<<<grid_for(g.owned_real), kBlock, 0, s>>>(
         rank_arr.data(), g.oe_off.data(), g.owned_real, g.v_begin,
         contrib.data() + g.v_begin);
-    if (pull) {
+    if (pull && tiled) {
+      // tiled pull: edge stream grouped by dst tile (L2-hot gathers);
+      // accumulates, so clear acc first. Phase A sweeps the tiles whose
+      // dsts lie in the owned slice while the mirror exchange runs.
+      acc.zero(s);
+      static int tile_grid = [] {
+        const char* e = getenv("GRAPEHIP_PR_GRID");
+        return e ? atoi(e) : 8192;
+      }();
+      if (multi) {
+        mirror_sync_begin(I, g, contrib.data(), s);
+        if (g.pr_own_hi > g.pr_own_lo)
+          pr_tiled_pull_kernel<<<tile_grid, kBlock, 0, s>>>(
+              g.pr_tiles.data(), g.pr_own_lo, g.pr_own_hi, contrib.data(),
+              acc.data());
+        fold_dangling();
+        mirror_sync_end(I, g, contrib.data(), s);
+        if (g.pr_own_lo > 0)
+          pr_tiled_pull_kernel<<<tile_grid, kBlock, 0, s>>>(
+              g.pr_tiles.data(), 0, g.pr_own_lo, contrib.data(),
+              acc.data());
+        if (g.pr_total > g.pr_own_hi)
+          pr_tiled_pull_kernel<<<tile_grid, kBlock, 0, s>>>(
+              g.pr_tiles.data(), g.pr_own_hi, g.pr_total, contrib.data(),
+              acc.data());
+      } else {
+        if (g.pr_total)
+          pr_tiled_pull_kernel<<<tile_grid, kBlock, 0, s>>>(
+              g.pr_tiles.data(), 0, g.pr_total, contrib.data(),
+              acc.data());
+      }
+    } else if (pull) {
       if (multi) {
         mirror_sync_begin(I, g, contrib.data(), s);
         pulls(1);          // owned spans, overlapped with the exchange
@@ -2987,8 +3414,21 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
   res.traversed_edges = g.input_edges;
   if (fetch) {
     std::vector<uint32_t> lab(owned);
-    HIP_CHECK(hipMemcpyAsync(lab.data(), parent.data() + g.v_begin,
-                             owned * 4, hipMemcpyDeviceToHost, s));
+    const uint32_t* lab_src = parent.data() + g.v_begin;
+    DeviceBuffer<uint32_t> relab;
+    if (g.permuted) {
+      // canonical labels = component min OLD id (oracle/LDBC convention)
+      DeviceBuffer<uint32_t> mo(nv_pad);
+      mo.fill_bytes(0xFF, s);
+      wcc_minold_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(
+          parent.data(), g.inv.data(), nv_pad, mo.data());
+      relab.resize(owned);
+      wcc_relabel_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+          parent.data(), mo.data(), g.v_begin, owned, relab.data());
+      lab_src = relab.data();
+    }
+    HIP_CHECK(hipMemcpyAsync(lab.data(), lab_src, owned * 4,
+                             hipMemcpyDeviceToHost, s));
     HIP_CHECK(hipStreamSynchronize(s));
     res.i64.assign(lab.begin(), lab.end());
   }
@@ -3591,7 +4031,17 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
   HIP_CHECK(hipDeviceSynchronize());
   double t0 = wall_s();
 
-  iota_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(lab.data(), 0, nv_pad);
+  // label space must compare in OID order (reference tie-break): dense
+  // renumber uses the sorted-oid table, hub renumber the inverse perm
+  const uint32_t* lab0 = g.oid_order.size()
+                             ? g.oid_order.data()
+                             : (g.permuted ? g.inv.data() : nullptr);
+  if (lab0)
+    HIP_CHECK(hipMemcpyAsync(lab.data(), lab0,
+                             static_cast<size_t>(nv_pad) * 4,
+                             hipMemcpyDeviceToDevice, s));
+  else
+    iota_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(lab.data(), 0, nv_pad);
   changed_bm.zero(s);
   int rounds = 0;
   bool use_dirty = false;  // iteration 0 recomputes everything

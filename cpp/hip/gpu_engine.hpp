@@ -51,6 +51,24 @@ struct DeviceGraph {
   std::vector<uint64_t> mr_recv_off, mr_send_off;    // [world+1] element offs
   DeviceBuffer<uint8_t> mr_sendbuf, mr_recvbuf;      // 4B-element staging
   bool mirrors_built = false;
+  // CDLP label order (non-identity maps only): dev id -> rank in global
+  // sorted-OID order, so min-label tie-breaks match the reference's
+  // oid-space semantics under dense renumbering. Empty for identity maps.
+  DeviceBuffer<uint32_t> oid_order;
+  // hub-clustering renumber (synthetic graphs): new = perm[old],
+  // old = inv[new]; slices preserved. inv_host caches the owned slice
+  // for output oid translation (filled on first fetch).
+  DeviceBuffer<uint32_t> perm, inv;
+  bool permuted = false;
+  std::vector<uint32_t> inv_host;
+  // PageRank tiled-pull stream (built lazily; 8 B per stored edge):
+  // (src<<32|dst) records grouped by 4096-vertex dst tile so gathers stay
+  // L2-resident. pr_ntiles==0 after build => fell back (memory gate).
+  DeviceBuffer<unsigned long long> pr_tiles;
+  DeviceBuffer<uint64_t> pr_tile_off;
+  uint32_t pr_ntiles = 0;
+  uint64_t pr_own_lo = 0, pr_own_hi = 0, pr_total = 0;
+  bool pr_tiles_built = false;
   uint32_t owned() const { return v_end - v_begin; }
   // cached hipGraph of one PageRank iteration (single-GPU fixed-iter
   // path) + the working set it references: the capture bakes device
