@@ -392,6 +392,10 @@ PYBIND11_MODULE(_core, m) {
                  out["hip_total"] = total_b;
                  out["hip_used"] = total_b - free_b;
                }
+               // engine-tracked DeviceBuffer bytes (reference
+               // utils/memory_tracker parity): current + high-water
+               out["hip_alloc_current"] = hip_alloc_current().load();
+               out["hip_alloc_peak"] = hip_alloc_peak().load();
              }
 #endif
              return out;

@@ -5,6 +5,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <atomic>
 #include <cstdint>
 #include <stdexcept>
 #include <string>
@@ -72,6 +73,26 @@ inline int grid_for(size_t n, int block = kBlock, int cap = kMaxGrid) {
   return static_cast<int>(g < static_cast<size_t>(cap) ? g : cap);
 }
 
+// HBM allocation tracking (reference utils/memory_tracker.{h,cc},
+// TRACKING_MEMORY): every DeviceBuffer counts toward current + peak,
+// surfaced by Engine.memory_info().
+inline std::atomic<long long>& hip_alloc_current() {
+  static std::atomic<long long> v{0};
+  return v;
+}
+inline std::atomic<long long>& hip_alloc_peak() {
+  static std::atomic<long long> v{0};
+  return v;
+}
+inline void hip_alloc_note(long long delta) {
+  long long cur = hip_alloc_current().fetch_add(delta) + delta;
+  if (delta > 0) {
+    long long p = hip_alloc_peak().load();
+    while (cur > p && !hip_alloc_peak().compare_exchange_weak(p, cur)) {
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 template <typename T>
 class DeviceBuffer {
@@ -97,11 +118,17 @@ class DeviceBuffer {
   void resize(size_t n) {
     if (n == n_) return;
     free();
-    if (n) HIP_CHECK(hipMalloc(&p_, n * sizeof(T)));
+    if (n) {
+      HIP_CHECK(hipMalloc(&p_, n * sizeof(T)));
+      hip_alloc_note(static_cast<long long>(n * sizeof(T)));
+    }
     n_ = n;
   }
   void free() {
-    if (p_) (void)hipFree(p_);
+    if (p_) {
+      (void)hipFree(p_);
+      hip_alloc_note(-static_cast<long long>(n_ * sizeof(T)));
+    }
     p_ = nullptr;
     n_ = 0;
   }

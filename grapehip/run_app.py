@@ -156,6 +156,12 @@ def main(argv=None):
     if rank == 0:
         for name, secs in timers:
             print("[timer] %-14s %.6f s" % (name, secs))
+        # reference GetMemoryUsage line printed with the timer table
+        mi = eng.memory_info()
+        parts = ["%s=%.1fMB" % (k, mi[k] / 1048576.0)
+                 for k in ("VmHWM", "VmRSS", "hip_alloc_peak", "hip_used")
+                 if k in mi]
+        print("[memory] " + " ".join(parts))
         if app == "kclique":
             print("clique_num = %d" % res["clique_count"])
     return 0

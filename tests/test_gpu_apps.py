@@ -305,3 +305,14 @@ def test_gpu_exclusive_scan_sizes(eng):
         got = np.array(eng._debug_scan(a.tolist()), dtype=np.uint64)
         exp = np.concatenate([[0], np.cumsum(a, dtype=np.uint64)])
         assert np.array_equal(got, exp), n
+
+
+def test_memory_tracking(eng):
+    # reference memory_tracker parity: DeviceBuffer bytes are accounted
+    # with a high-water mark
+    g = eng.load_synthetic(num_vertices=100000, num_edges=800000, seed=3)
+    eng.bfs(g, 0)
+    mi = eng.memory_info()
+    assert mi["hip_alloc_current"] > 0
+    assert mi["hip_alloc_peak"] >= mi["hip_alloc_current"]
+    assert mi["VmHWM"] > 0
