@@ -259,6 +259,8 @@ py::dict gpu_dict(const GpuRunResult& r, const PyGraph& pg, bool is_i64,
   out["rounds"] = r.rounds;
   out["seconds"] = r.seconds;
   out["traversed_edges"] = r.traversed_edges;
+  out["bytes_p2p"] = r.bytes_p2p;
+  out["bytes_coll"] = r.bytes_coll;
   if (with_values) {
     DevIdMap m = dev_id_map(pg);
     // hashmap uploads pad the owned range; emit only real vertices

@@ -1584,6 +1584,10 @@ struct DevComm {
   TcpComm* tcp = nullptr;
   int rank = 0, world = 1;
   bool use_tcp = false;
+  // per-rank data-plane byte accounting (this rank's sends), split into
+  // point-to-point (halo/mirror/row-fetch) and collective traffic —
+  // read back per app run so comm volume vs boundary size is observable
+  uint64_t bytes_p2p = 0, bytes_coll = 0;
 
   bool active() const { return world > 1; }
 
@@ -1602,6 +1606,7 @@ struct DevComm {
   // contribution; on return buf[0 .. world*slice) holds every rank's.
   void allgather32(void* buf, uint64_t slice, bool fp, hipStream_t s) {
     if (!active()) return;
+    bytes_coll += slice * 4;
     if (!use_tcp) {
       NCCL_CHECK(ncclAllGather(
           static_cast<char*>(buf) + static_cast<uint64_t>(rank) * slice * 4,
@@ -1620,6 +1625,7 @@ struct DevComm {
   // return buf[rank*slice ..) = elementwise sum of every rank's slice.
   void reduce_scatter_sum_f64(double* buf, uint64_t slice, hipStream_t s) {
     if (!active()) return;
+    bytes_coll += static_cast<uint64_t>(world) * slice * 8;
     if (!use_tcp) {
       NCCL_CHECK(ncclReduceScatter(
           buf, buf + static_cast<uint64_t>(rank) * slice, slice, ncclDouble,
@@ -1646,6 +1652,7 @@ struct DevComm {
 
   void allreduce_min_u32(uint32_t* buf, uint64_t n, hipStream_t s) {
     if (!active()) return;
+    bytes_coll += n * 4;
     if (!use_tcp) {
       NCCL_CHECK(
           ncclAllReduce(buf, buf, n, ncclUint32, ncclMin, nccl, s));
@@ -1666,6 +1673,7 @@ struct DevComm {
   void allreduce_sum_u64(unsigned long long* buf, uint64_t n,
                          hipStream_t s) {
     if (!active()) return;
+    bytes_coll += n * 8;
     if (!use_tcp) {
       NCCL_CHECK(ncclAllReduce(buf, buf, n, ncclUint64, ncclSum, nccl, s));
       return;
@@ -1691,6 +1699,8 @@ struct DevComm {
                 uint8_t* recvbuf, const std::vector<uint64_t>& recv_off,
                 hipStream_t s) {
     if (!active()) return;
+    for (int p = 0; p < world; ++p)
+      if (p != rank) bytes_p2p += send_off[p + 1] - send_off[p];
     if (!use_tcp) {
       NCCL_CHECK(ncclGroupStart());
       for (int p = 0; p < world; ++p) {
@@ -1729,6 +1739,7 @@ struct DevComm {
   void bcast_regions_u32(uint32_t* buf, const std::vector<uint64_t>& region,
                          hipStream_t s) {
     if (!active()) return;
+    bytes_coll += (region[rank + 1] - region[rank]) * 4;
     if (!use_tcp) {
       NCCL_CHECK(ncclGroupStart());
       for (int f = 0; f < world; ++f) {
@@ -2697,6 +2708,7 @@ GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source,
   if (comm_) comm_->barrier();
   HIP_CHECK(hipDeviceSynchronize());
   double t0 = wall_s();
+  const uint64_t b_p2p0 = I.dc.bytes_p2p, b_coll0 = I.dc.bytes_coll;
 
   depth.fill_bytes(0xFF, s);
   next_bm.zero(s);
@@ -2791,6 +2803,8 @@ GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source,
   double t1 = wall_s();
 
   GpuRunResult res;
+  res.bytes_p2p = I.dc.bytes_p2p - b_p2p0;
+  res.bytes_coll = I.dc.bytes_coll - b_coll0;
   res.rounds = rounds;
   res.seconds = comm_ ? comm_->allreduce_max_double(t1 - t0) : (t1 - t0);
   res.traversed_edges = g.input_edges;
@@ -2859,6 +2873,7 @@ GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
   if (comm_) comm_->barrier();
   HIP_CHECK(hipDeviceSynchronize());
   double t0 = wall_s();
+  const uint64_t b_p2p0 = I.dc.bytes_p2p, b_coll0 = I.dc.bytes_coll;
 
   fill(dist.data(), std::numeric_limits<float>::max(), nv_pad, s);
   fcnt.zero(s);
@@ -2957,6 +2972,8 @@ GpuRunResult GpuContext::sssp(DeviceGraph& g, int64_t source,
   double t1 = wall_s();
 
   GpuRunResult res;
+  res.bytes_p2p = I.dc.bytes_p2p - b_p2p0;
+  res.bytes_coll = I.dc.bytes_coll - b_coll0;
   res.rounds = rounds;
   res.seconds = comm_ ? comm_->allreduce_max_double(t1 - t0) : (t1 - t0);
   res.traversed_edges = g.input_edges;
@@ -3028,6 +3045,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
   if (comm_) comm_->barrier();
   HIP_CHECK(hipDeviceSynchronize());
   double t0 = wall_s();
+  const uint64_t b_p2p0 = I.dc.bytes_p2p, b_coll0 = I.dc.bytes_coll;
 
   fill(rank_arr.data(), 1.0 / N, nv_pad, s);
   int rounds = 0;
@@ -3194,6 +3212,8 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
   double t1 = wall_s();
 
   GpuRunResult res;
+  res.bytes_p2p = I.dc.bytes_p2p - b_p2p0;
+  res.bytes_coll = I.dc.bytes_coll - b_coll0;
   res.rounds = rounds;
   res.seconds = comm_ ? comm_->allreduce_max_double(t1 - t0) : (t1 - t0);
   res.traversed_edges = static_cast<uint64_t>(iters) * g.total_edges;
@@ -3241,6 +3261,7 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
   if (comm_) comm_->barrier();
   HIP_CHECK(hipDeviceSynchronize());
   double t0 = wall_s();
+  const uint64_t b_p2p0 = I.dc.bytes_p2p, b_coll0 = I.dc.bytes_coll;
 
   iota_kernel<<<grid_for(nv_pad), kBlock, 0, s>>>(parent.data(), 0, nv_pad);
   if (multi) chg_bm.zero(s);
@@ -3409,6 +3430,8 @@ GpuRunResult GpuContext::wcc(DeviceGraph& g, bool fetch) {
   double t1 = wall_s();
 
   GpuRunResult res;
+  res.bytes_p2p = I.dc.bytes_p2p - b_p2p0;
+  res.bytes_coll = I.dc.bytes_coll - b_coll0;
   res.rounds = rounds;
   res.seconds = comm_ ? comm_->allreduce_max_double(t1 - t0) : (t1 - t0);
   res.traversed_edges = g.input_edges;
@@ -4030,6 +4053,7 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
   if (comm_) comm_->barrier();
   HIP_CHECK(hipDeviceSynchronize());
   double t0 = wall_s();
+  const uint64_t b_p2p0 = I.dc.bytes_p2p, b_coll0 = I.dc.bytes_coll;
 
   // label space must compare in OID order (reference tie-break): dense
   // renumber uses the sorted-oid table, hub renumber the inverse perm
@@ -4135,6 +4159,8 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
   double t1 = wall_s();
 
   GpuRunResult res;
+  res.bytes_p2p = I.dc.bytes_p2p - b_p2p0;
+  res.bytes_coll = I.dc.bytes_coll - b_coll0;
   res.rounds = rounds;
   res.seconds = comm_ ? comm_->allreduce_max_double(t1 - t0) : (t1 - t0);
   res.traversed_edges =
@@ -5520,6 +5546,7 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
   if (comm_) comm_->barrier();
   HIP_CHECK(hipDeviceSynchronize());
   double t0 = wall_s();
+  const uint64_t b_p2p0 = I.dc.bytes_p2p, b_coll0 = I.dc.bytes_coll;
 
   DeviceBuffer<uint32_t> Dv(nv_pad);
   Dv.zero(s);
@@ -5736,6 +5763,8 @@ GpuRunResult GpuContext::lcc_directed(DeviceGraph& g, bool fetch) {
   double t1 = wall_s();
 
   GpuRunResult res;
+  res.bytes_p2p = I.dc.bytes_p2p - b_p2p0;
+  res.bytes_coll = I.dc.bytes_coll - b_coll0;
   res.rounds = 1;
   res.seconds = comm_ ? comm_->allreduce_max_double(t1 - t0) : (t1 - t0);
   res.traversed_edges = g.input_edges;
@@ -5783,6 +5812,7 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
   if (comm_) comm_->barrier();
   HIP_CHECK(hipDeviceSynchronize());
   double t0 = wall_s();
+  const uint64_t b_p2p0 = I.dc.bytes_p2p, b_coll0 = I.dc.bytes_coll;
 
   // tier rows by combined degree (reuses the CDLP bucketer)
   DeviceBuffer<uint32_t> t_small(owned), t_mid(owned), t_large(owned);
@@ -6074,6 +6104,8 @@ GpuRunResult GpuContext::lcc(DeviceGraph& g, bool fetch) {
   double t1 = wall_s();
 
   GpuRunResult res;
+  res.bytes_p2p = I.dc.bytes_p2p - b_p2p0;
+  res.bytes_coll = I.dc.bytes_coll - b_coll0;
   res.rounds = 1;
   res.seconds = comm_ ? comm_->allreduce_max_double(t1 - t0) : (t1 - t0);
   res.traversed_edges = g.input_edges;

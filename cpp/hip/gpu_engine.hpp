@@ -88,6 +88,9 @@ struct GpuRunResult {
   int rounds = 0;
   double seconds = 0;         // max over ranks, kernel-side barrier-bracketed
   uint64_t traversed_edges = 0;  // for TEPS accounting (global)
+  // this rank's data-plane sends during the run (comm-volume evidence:
+  // p2p halo/mirror/row-fetch vs collective traffic)
+  uint64_t bytes_p2p = 0, bytes_coll = 0;
 };
 
 class GpuContext {
