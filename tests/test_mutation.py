@@ -95,3 +95,64 @@ def test_mutate_weighted(eng):
     expect = sssp_oracle(nv, e_src, e_dst, e_w, 1, directed=True)
     finite = expect < 1e300
     assert np.allclose(r["values"][order][finite], expect[finite], rtol=1e-6)
+
+
+def test_repeated_deltas_accumulate(eng):
+    # in-place path: successive deltas reuse slack, grow rows into the
+    # end arena, and add new outer vertices; final state must equal a
+    # fresh build of the accumulated edge list
+    src, dst = base_graph(seed=101, num_v=400, num_e=1200)
+    nv = 400
+    g = eng.load_edges(src, dst, directed=False, num_vertices=nv)
+    cur_s, cur_d = src.copy(), dst.copy()
+    rng = np.random.default_rng(103)
+    empty = np.array([], dtype=np.int64)
+    for step in range(5):
+        ad_s = rng.integers(0, nv, 150).astype(np.int64)
+        ad_d = rng.integers(0, nv, 150).astype(np.int64)
+        k = ad_s != ad_d
+        ad_s, ad_d = ad_s[k], ad_d[k]
+        rm_s, rm_d = cur_s[:20].copy(), cur_d[:20].copy()
+        g = eng.mutate_graph(g, add_src=ad_s, add_dst=ad_d,
+                             remove_src=rm_s, remove_dst=rm_d,
+                             remove_vertices=empty)
+        pairs = set()
+        for a, b in zip(rm_s, rm_d):
+            pairs.add((a, b))
+            pairs.add((b, a))
+        keep = np.array([(a, b) not in pairs
+                         for a, b in zip(cur_s, cur_d)])
+        cur_s = np.concatenate([cur_s[keep], ad_s])
+        cur_d = np.concatenate([cur_d[keep], ad_d])
+    r = eng.bfs(g, 5)
+    order = np.argsort(r["oids"])
+    assert np.array_equal(r["values"][order],
+                          bfs_oracle(nv, cur_s, cur_d, 5, directed=False))
+    r = eng.wcc(g)
+    assert np.array_equal(r["values"][np.argsort(r["oids"])],
+                          wcc_oracle(nv, cur_s, cur_d))
+
+
+def test_serialize_after_mutate(eng, tmp_path):
+    # mutable-mode slack must compact into a canonical checkpoint
+    src, dst = base_graph(seed=107)
+    nv = 600
+    g = eng.load_edges(src, dst, directed=False, num_vertices=nv)
+    rng = np.random.default_rng(109)
+    ad_s = rng.integers(0, nv, 200).astype(np.int64)
+    ad_d = rng.integers(0, nv, 200).astype(np.int64)
+    k = ad_s != ad_d
+    ad_s, ad_d = ad_s[k], ad_d[k]
+    empty = np.array([], dtype=np.int64)
+    g = eng.mutate_graph(g, add_src=ad_s, add_dst=ad_d, remove_src=empty,
+                         remove_dst=empty, remove_vertices=empty)
+    ck = tmp_path / "ck"
+    ck.mkdir()
+    eng.save_graph(g, str(ck))
+    g2 = eng.load_serialized(str(ck))
+    e_src = np.concatenate([src, ad_s])
+    e_dst = np.concatenate([dst, ad_d])
+    r = eng.bfs(g2, 5)
+    order = np.argsort(r["oids"])
+    assert np.array_equal(r["values"][order],
+                          bfs_oracle(nv, e_src, e_dst, 5, directed=False))
