@@ -14,6 +14,7 @@
 // cpp/hip/.
 #pragma once
 
+#include <atomic>
 #include <chrono>
 #include <cstdio>
 #include <cstdlib>
@@ -41,11 +42,20 @@ class MessageManager {
     force_continue_ = false;
     terminated_ = false;
     round_ = 0;
+    eager_bytes_.store(0, std::memory_order_relaxed);
+    // compute/comm overlap (reference parallel_message_manager.h send
+    // thread): channel blocks stream to peers over the data mesh WHILE
+    // PEval/IncEval runs; the round boundary only drains stragglers
+    if (comm_ && fnum_ > 1) comm_->enable_data_mesh();
   }
 
   void start_round() { force_continue_ = false; }
 
   // ---- send API (thread-safe via tid channel) --------------------------
+  // blocks above this size flush mid-round to the background sender
+  // (reference thread_local_message_buffer.h block flush)
+  static constexpr size_t kFlushBlock = 256 << 10;
+
   template <typename MSG>
   void sync_state_on_outer_vertex(int tid, vid_t outer_lid, const MSG& msg) {
     vid_t gid = frag_->lid2gid(outer_lid);
@@ -53,12 +63,14 @@ class MessageManager {
     auto& ar = channels_[tid][dst];
     ar.add(gid);
     ar.add(msg);
+    maybe_flush(ar, dst);
   }
   template <typename MSG>
   void send_to_fragment(int tid, fid_t dst, vid_t gid, const MSG& msg) {
     auto& ar = channels_[tid][dst];
     ar.add(gid);
     ar.add(msg);
+    maybe_flush(ar, dst);
   }
 
   void force_continue() { force_continue_ = true; }
@@ -74,7 +86,8 @@ class MessageManager {
 
   // ---- round boundary --------------------------------------------------
   void finish_round() {
-    // concatenate per-thread channels per destination
+    // concatenate the channel tails per destination (most volume already
+    // streamed mid-round by the data-mesh sender)
     std::vector<std::string> send(fnum_);
     for (int f = 0; f < fnum_; ++f) {
       size_t total = 0;
@@ -85,10 +98,11 @@ class MessageManager {
         channels_[t][f].clear();
       }
     }
-    uint64_t moved = 0;
+    uint64_t moved = eager_bytes_.exchange(0, std::memory_order_relaxed);
     for (auto& s : send) moved += s.size();
     std::vector<std::string> recv =
-        (comm_ && fnum_ > 1) ? comm_->exchange_all(send) : std::move(send);
+        (comm_ && fnum_ > 1) ? comm_->data_round_end(std::move(send))
+                             : std::move(send);
     for (int f = 0; f < fnum_; ++f) recv_[f].reset(std::move(recv[f]));
 
     uint64_t global_moved =
@@ -112,10 +126,10 @@ class MessageManager {
     ++round_;
   }
 
-  // bytes queued in channels this round (message-size accounting parity:
+  // bytes queued/streamed this round (message-size accounting parity:
   // reference GetMsgSize, default_message_manager.h:146)
   uint64_t bytes_sent() const {
-    uint64_t n = 0;
+    uint64_t n = eager_bytes_.load(std::memory_order_relaxed);
     for (auto& per_tid : channels_)
       for (auto& a : per_tid) n += a.size();
     return n;
@@ -204,10 +218,20 @@ class MessageManager {
   }
 
  private:
+  void maybe_flush(InArchive& ar, fid_t dst) {
+    if (ar.size() < kFlushBlock || !comm_ || fnum_ <= 1 ||
+        dst == frag_->fid())
+      return;
+    eager_bytes_.fetch_add(ar.size(), std::memory_order_relaxed);
+    comm_->post_block(static_cast<int>(dst), ar.release());
+    ar.clear();
+  }
+
   TcpComm* comm_ = nullptr;
   const Fragment* frag_ = nullptr;
   int fnum_ = 1;
   int n_threads_ = 1;
+  std::atomic<uint64_t> eager_bytes_{0};
   bool force_continue_ = false;
   bool force_terminate_ = false;
   std::string terminate_info_;

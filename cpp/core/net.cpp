@@ -105,6 +105,7 @@ void TcpComm::init(int rank, int world, const std::string& master_addr,
                    int master_port) {
   rank_ = rank;
   world_ = world;
+  master_addr_ = master_addr;
   fds_.assign(world, -1);
   if (world == 1) return;
 
@@ -170,6 +171,19 @@ void TcpComm::init(int rank, int world, const std::string& master_addr,
 }
 
 void TcpComm::finalize() {
+  if (sender_.joinable()) {
+    {
+      std::lock_guard<std::mutex> lk(q_mu_);
+      sender_quit_ = true;
+    }
+    q_cv_.notify_all();
+    sender_.join();
+  }
+  for (int& fd : data_fds_) {
+    if (fd >= 0) close(fd);
+    fd = -1;
+  }
+  data_fds_.clear();
   for (int& fd : fds_) {
     if (fd >= 0) close(fd);
     fd = -1;
@@ -282,6 +296,113 @@ void TcpComm::barrier() {
   char b = 0;
   std::vector<char> sink(world_);
   allgather(&b, 1, sink.data());
+}
+
+// ---------------------------------------------------------------------------
+// Async data plane (second mesh + background sender; see net.hpp)
+// ---------------------------------------------------------------------------
+
+namespace {
+constexpr uint32_t kRoundMarker = 0xFFFFFFFFu;
+}
+
+void TcpComm::enable_data_mesh() {
+  if (world_ == 1 || !data_fds_.empty()) return;
+  data_fds_.assign(world_, -1);
+  uint16_t my_port = 0;
+  int lfd = listen_on(0, &my_port);
+  std::vector<uint16_t> ports(world_);
+  allgather(&my_port, sizeof(uint16_t), ports.data());
+  int expect = world_ - 1 - rank_;
+  std::thread acceptor([&] {
+    for (int k = 0; k < expect; ++k) {
+      int fd = accept(lfd, nullptr, nullptr);
+      if (fd < 0) throw_errno("accept(data mesh)");
+      set_common_opts(fd);
+      int32_t peer_rank;
+      read_all(fd, &peer_rank, sizeof(peer_rank));
+      data_fds_[peer_rank] = fd;
+    }
+  });
+  for (int i = 0; i < rank_; ++i) {
+    int fd = connect_to(master_addr_, ports[i]);
+    int32_t r32 = rank_;
+    write_all(fd, &r32, sizeof(r32));
+    data_fds_[i] = fd;
+  }
+  acceptor.join();
+  close(lfd);
+  sender_quit_ = false;
+  sender_ = std::thread([this] { sender_loop(); });
+}
+
+void TcpComm::sender_loop() {
+  for (;;) {
+    std::pair<int, std::string> item;
+    {
+      std::unique_lock<std::mutex> lk(q_mu_);
+      q_cv_.wait(lk, [&] { return sender_quit_ || !q_.empty(); });
+      if (q_.empty()) return;  // quit with drained queue
+      item = std::move(q_.front());
+      q_.pop_front();
+    }
+    if (item.first >= world_) {
+      // round marker for peer (first - world_)
+      write_all(data_fds_[item.first - world_], &kRoundMarker, 4);
+    } else {
+      uint32_t len = static_cast<uint32_t>(item.second.size());
+      write_all(data_fds_[item.first], &len, 4);
+      if (len) write_all(data_fds_[item.first], item.second.data(), len);
+    }
+  }
+}
+
+void TcpComm::post_block(int peer, std::string blob) {
+  if (blob.empty()) return;
+  if (peer == rank_) {
+    std::lock_guard<std::mutex> lk(self_mu_);
+    self_accum_.append(blob);
+    return;
+  }
+  {
+    std::lock_guard<std::mutex> lk(q_mu_);
+    q_.emplace_back(peer, std::move(blob));
+  }
+  q_cv_.notify_one();
+}
+
+std::vector<std::string> TcpComm::data_round_end(
+    std::vector<std::string> tail) {
+  // flush tails + markers (the sender streams them after any pending
+  // mid-round blocks — FIFO per construction)
+  {
+    std::lock_guard<std::mutex> lk(q_mu_);
+    for (int p = 0; p < world_; ++p) {
+      if (p == rank_) continue;
+      if (!tail[p].empty()) q_.emplace_back(p, std::move(tail[p]));
+      q_.emplace_back(world_ + p, std::string());
+    }
+  }
+  q_cv_.notify_one();
+  std::vector<std::string> recv(world_);
+  {
+    std::lock_guard<std::mutex> lk(self_mu_);
+    recv[rank_] = std::move(self_accum_);
+    self_accum_.clear();
+  }
+  recv[rank_].append(tail[rank_]);
+  for (int p = 0; p < world_; ++p) {
+    if (p == rank_) continue;
+    for (;;) {
+      uint32_t len = 0;
+      read_all(data_fds_[p], &len, 4);
+      if (len == kRoundMarker) break;
+      size_t off = recv[p].size();
+      recv[p].resize(off + len);
+      if (len) read_all(data_fds_[p], recv[p].data() + off, len);
+    }
+  }
+  return recv;
 }
 
 }  // namespace grapehip

@@ -8,8 +8,12 @@
 // allreduce, RCCL uniqueId broadcast, and the (test-only) CPU data path.
 #pragma once
 
+#include <condition_variable>
 #include <cstdint>
+#include <deque>
+#include <mutex>
 #include <string>
+#include <thread>
 #include <vector>
 
 namespace grapehip {
@@ -53,11 +57,39 @@ class TcpComm {
   void bcast(void* buf, size_t n, int root);
   void barrier();
 
+  // ---- async data plane (reference parallel_message_manager.h:396-434
+  // send-thread parity) -------------------------------------------------
+  // A SECOND full mesh carries framed message blocks streamed by a
+  // background sender thread WHILE compute runs; control collectives stay
+  // on the primary mesh, so mid-round allreduces never interleave with
+  // payload frames. Collective to enable; rendezvous rides the existing
+  // mesh (no extra master port).
+  void enable_data_mesh();
+  bool data_mesh_enabled() const { return !data_fds_.empty(); }
+  // enqueue a block for peer (thread-safe; background thread streams it)
+  void post_block(int peer, std::string blob);
+  // flush tails + per-peer round markers, then drain every peer's frames
+  // until its marker; returns the concatenated payload per source rank
+  // (recv[rank] = self blocks + tail[rank]).
+  std::vector<std::string> data_round_end(std::vector<std::string> tail);
+
  private:
+  void sender_loop();
+
   int rank_ = 0;
   int world_ = 1;
   std::vector<int> fds_;  // fds_[peer]; -1 for self
   int listen_fd_ = -1;
+  std::string master_addr_;
+  // data mesh state
+  std::vector<int> data_fds_;
+  std::thread sender_;
+  std::mutex q_mu_;
+  std::condition_variable q_cv_;
+  std::deque<std::pair<int, std::string>> q_;  // peer, blob ("" + peer<0: quit)
+  bool sender_quit_ = false;
+  std::string self_accum_;
+  std::mutex self_mu_;
 };
 
 }  // namespace grapehip
