@@ -54,7 +54,7 @@ class MessageManager {
   // ---- send API (thread-safe via tid channel) --------------------------
   // blocks above this size flush mid-round to the background sender
   // (reference thread_local_message_buffer.h block flush)
-  static constexpr size_t kFlushBlock = 256 << 10;
+  static constexpr size_t kFlushBlock = 32 << 10;
 
   template <typename MSG>
   void sync_state_on_outer_vertex(int tid, vid_t outer_lid, const MSG& msg) {

@@ -276,3 +276,17 @@ def test_force_terminate_mp(world, free_port, tmp_path):
     oids, vals = run_world(world, cfg, free_port, tmp_path)
     assert len(vals) == world
     assert (vals == 1).all()  # 1 = raised with the right info on that rank
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_large_messages_eager_flush(world, free_port, tmp_path):
+    # boundary messages exceeding the flush block force the mid-round
+    # flush (background sender thread); results must still merge exactly
+    cfg = dict(BASE, app="sssp", num_v=60000, num_e=360000, weighted=True,
+               directed=False, source=11)
+    oids, vals = run_world(world, cfg, free_port, tmp_path, timeout=300)
+    src, dst, w = graph_arrays(cfg)
+    expect = sssp_oracle(cfg["num_v"], src, dst, w, 11, directed=False)
+    finite = expect < 1e300
+    assert np.array_equal(vals >= 1e300, ~finite)
+    assert np.allclose(vals[finite], expect[finite], rtol=1e-4)
