@@ -69,6 +69,7 @@ def main():
                "data": "synthetic RMAT stand-in", "n_gpus": eng.world,
                "load_seconds": load_s, "algorithms": {}}
     for algo in args.algorithms.split(","):
+        t_algo0 = time.time()
         if algo == "bfs":
             r = eng.bfs(g, args.source, values=False)
             edges = ne
@@ -89,14 +90,20 @@ def main():
             edges = ne
         else:
             raise SystemExit("unknown algorithm " + algo)
+        # Graphalytics splits makespan (wall around the job incl. driver
+        # overheads) from processing time (the algorithm phase proper)
+        makespan = time.time() - t_algo0
         results["algorithms"][algo] = {
-            "makespan_s": r["seconds"],
+            "makespan_s": makespan,
+            "processing_s": r["seconds"],
             "teps": edges / r["seconds"] if r["seconds"] > 0 else None,
             "rounds": r["rounds"],
+            "bytes_p2p": int(r.get("bytes_p2p", 0)),
+            "bytes_coll": int(r.get("bytes_coll", 0)),
         }
         if rank == 0:
-            print("%-10s %10.2f ms   %12.3g TEPS" %
-                  (algo, r["seconds"] * 1e3,
+            print("%-10s proc %10.2f ms  makespan %10.2f ms  %12.3g TEPS" %
+                  (algo, r["seconds"] * 1e3, makespan * 1e3,
                    edges / max(r["seconds"], 1e-12)))
 
     if rank == 0:
