@@ -247,9 +247,14 @@ __global__ void iota_kernel(uint32_t* p, uint32_t base, size_t n) {
 // ===========================================================================
 
 // Quadrant thresholds in 16-bit fixed point.
+// scr_a/scr_b: bijective LCG scramble (gcd(scr_a,nv)=1) decorrelating
+// RMAT's low-popcount hub ids from the contiguous ownership slices —
+// without it rank 0 of an 8-way run owns ~45% of all edges (measured),
+// which would sink the scaling curve. scr_a=0 disables.
 __global__ void gen_edges_kernel(uint64_t ne, uint64_t seed, int scale,
                                  uint32_t nv, uint32_t t_a, uint32_t t_ab,
-                                 uint32_t t_abc, uint32_t my_begin,
+                                 uint32_t t_abc, uint64_t scr_a,
+                                 uint64_t scr_b, uint32_t my_begin,
                                  uint32_t my_end, bool undirected,
                                  bool reverse, bool weighted,
                                  uint32_t* out_src, uint32_t* out_dst,
@@ -278,6 +283,15 @@ __global__ void gen_edges_kernel(uint64_t ne, uint64_t seed, int scale,
     }
     if (s >= nv) s -= nv;  // fold 2^scale domain onto [0, nv)
     if (d >= nv) d -= nv;
+    if (scr_a) {
+      // LCG then swap {0, scr_b}: raw vertex 0 (the deepest RMAT hub)
+      // keeps id 0, so the benchmark's named source stays the hub while
+      // every slice still gets a uniform share of the edge mass
+      uint64_t ss = (scr_a * s + scr_b) % nv;
+      uint64_t dd = (scr_a * d + scr_b) % nv;
+      s = static_cast<uint32_t>(ss == scr_b ? 0 : (ss == 0 ? scr_b : ss));
+      d = static_cast<uint32_t>(dd == scr_b ? 0 : (dd == 0 ? scr_b : dd));
+    }
     float w = 1.0f;
     if (weighted)
       w = static_cast<float>((h >> 16) & 0xFFFFFF) * (99.0f / 16777216.0f) +
@@ -1997,6 +2011,23 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
   uint32_t t_a = static_cast<uint32_t>(a * 65536.0);
   uint32_t t_ab = static_cast<uint32_t>((a + b) * 65536.0);
   uint32_t t_abc = static_cast<uint32_t>((a + b + c) * 65536.0);
+  // slice-balance scramble (see gen_edges_kernel); deterministic per nv
+  uint64_t scr_a = 0, scr_b = 0;
+  if (!(getenv("GRAPEHIP_SCRAMBLE") &&
+        atoi(getenv("GRAPEHIP_SCRAMBLE")) == 0) && nv > 2) {
+    scr_a = static_cast<uint64_t>(nv * 0.6180339887498949);
+    if (scr_a < 2) scr_a = 2;
+    auto gcd_u64 = [](uint64_t x, uint64_t y) {
+      while (y) {
+        uint64_t t = x % y;
+        x = y;
+        y = t;
+      }
+      return x;
+    };
+    while (gcd_u64(scr_a, nv) != 1) ++scr_a;
+    scr_b = (seed * 0x9E3779B97F4A7C15ULL) % nv;
+  }
 
   hipStream_t s = impl_->compute;
   // exact per-rank edge count via a counting pass (see gen_edges_kernel)
@@ -2007,7 +2038,7 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
     est = static_cast<uint64_t>((directed ? 1.0 : 2.0) * ne) + 16;
   } else {
     gen_edges_kernel<<<kMaxGrid, kBlock, 0, s>>>(
-        ne, seed, scale, g->nv_global, t_a, t_ab, t_abc,
+        ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, scr_a, scr_b,
         g->v_begin, g->v_end, !directed, false, false, nullptr, nullptr,
         nullptr, cnt.data());
     unsigned long long c = 0;
@@ -2022,7 +2053,7 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
     fprintf(stderr, "[gen] nv=%u ne=%lu est=%lu scale=%d\n", g->nv_global,
             (unsigned long)ne, (unsigned long)est, scale);
   gen_edges_kernel<<<kMaxGrid, kBlock, 0, s>>>(
-      ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, g->v_begin, g->v_end,
+      ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, scr_a, scr_b, g->v_begin, g->v_end,
       !directed, false, weighted, e_src.data(), e_dst.data(),
       weighted ? e_w.data() : nullptr, cnt.data());
   unsigned long long n_local = 0;
@@ -2092,7 +2123,7 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
       // in-edge ownership skews differently than out; recount exactly
       cnt.zero(s);
       gen_edges_kernel<<<kMaxGrid, kBlock, 0, s>>>(
-          ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, g->v_begin,
+          ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, scr_a, scr_b, g->v_begin,
           g->v_end, false, true, false, nullptr, nullptr, nullptr,
           cnt.data());
       unsigned long long c_in = 0;
@@ -2111,7 +2142,7 @@ std::unique_ptr<DeviceGraph> GpuContext::gen_synthetic(
     }
     cnt.zero(s);
     gen_edges_kernel<<<kMaxGrid, kBlock, 0, s>>>(
-        ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, g->v_begin,
+        ne, seed, scale, g->nv_global, t_a, t_ab, t_abc, scr_a, scr_b, g->v_begin,
         g->v_end, false, true, weighted, e_src.data(), e_dst.data(),
         weighted ? e_w.data() : nullptr, cnt.data());
     unsigned long long n_in = 0;
