@@ -2645,6 +2645,106 @@ static void mirror_sync_end(GpuContext::Impl& I, DeviceGraph& g,
   ref_sync_end(I, g.mr_recv_idx, g.mr_recv_off, g.mr_recvbuf, state4, s);
 }
 
+// Changed-gated sparse refresh: ship only (id, value) pairs whose OWNED
+// row is marked in `chg` (bit index = id - v_begin). Converging apps
+// (CDLP late iterations) change few labels — the dense mirror refresh
+// would re-ship every referenced value regardless. Pair order within a
+// peer is irrelevant (receiver scatters by id), so packing appends with
+// plain atomic cursors into capacity regions; per-round counts ride the
+// control plane.
+__global__ void mirror_pack_sparse_kernel(
+    const uint32_t* __restrict__ send_idx, uint64_t b, uint64_t e,
+    const uint32_t* __restrict__ chg_words, uint32_t v_begin,
+    const uint32_t* __restrict__ state, unsigned long long* __restrict__ cnt,
+    unsigned long long* __restrict__ out) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint64_t i = b + blockIdx.x * blockDim.x + threadIdx.x; i < e;
+       i += stride) {
+    uint32_t v = send_idx[i];
+    uint32_t bit = v - v_begin;
+    if ((chg_words[bit >> 5] >> (bit & 31)) & 1u) {
+      unsigned long long pos = atomicAdd(cnt, 1ull);
+      out[b + pos] = (static_cast<unsigned long long>(v) << 32) | state[v];
+    }
+  }
+}
+
+__global__ void mirror_scatter_pairs_kernel(
+    const unsigned long long* __restrict__ pairs, uint64_t n,
+    uint32_t* __restrict__ state) {
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint64_t i = static_cast<uint64_t>(blockIdx.x) * blockDim.x +
+                    threadIdx.x;
+       i < n; i += stride)
+    state[pairs[i] >> 32] = static_cast<uint32_t>(pairs[i]);
+}
+
+static void mirror_sync_changed(GpuContext::Impl& I, TcpComm* comm,
+                                DeviceGraph& g, void* state4,
+                                const uint32_t* chg_words, int rank,
+                                int world, hipStream_t s) {
+  uint64_t ns = g.mr_send_off.back();
+  uint64_t nr = g.mr_recv_off.back();
+  // pair staging reuses the mirror buffers when large enough (8 B/entry;
+  // recvbuf doubles as the compaction bounce, so size it for both sides)
+  uint64_t big = (ns > nr ? ns : nr) * 8 + 8;
+  if (g.mr_sendbuf.size() < ns * 8 + 8) g.mr_sendbuf.resize(ns * 8 + 8);
+  if (g.mr_recvbuf.size() < big) g.mr_recvbuf.resize(big);
+  DeviceBuffer<unsigned long long> cnts(world);
+  cnts.zero(s);
+  auto* pairs = reinterpret_cast<unsigned long long*>(g.mr_sendbuf.data());
+  for (int q = 0; q < world; ++q) {
+    uint64_t b = g.mr_send_off[q], e = g.mr_send_off[q + 1];
+    if (e > b)
+      mirror_pack_sparse_kernel<<<grid_for(e - b), kBlock, 0, s>>>(
+          g.mr_send_idx.data(), b, e, chg_words, g.v_begin,
+          static_cast<const uint32_t*>(state4), cnts.data() + q, pairs);
+  }
+  auto hc = cnts.download(s);
+  std::vector<uint64_t> mine(world);
+  for (int q = 0; q < world; ++q) mine[q] = hc[q];
+  std::vector<uint64_t> matrix(static_cast<size_t>(world) * world);
+  comm->allgather(mine.data(), world * 8, matrix.data());
+  std::vector<uint64_t> sb(world + 1), rb(world + 1);
+  sb[0] = 0;
+  rb[0] = 0;
+  for (int q = 0; q < world; ++q) sb[q + 1] = sb[q] + mine[q] * 8;
+  uint64_t recv_pairs = 0;
+  for (int p = 0; p < world; ++p) {
+    uint64_t from_p =
+        p == rank ? 0 : matrix[static_cast<size_t>(p) * world + rank];
+    rb[p + 1] = rb[p] + from_p * 8;
+    recv_pairs += from_p;
+  }
+  // compact the capacity regions into contiguous send order: copy each
+  // peer's [b*8, b*8 + cnt*8) down — regions are already at b*8; the
+  // contiguous layout needs offsets sb. D2D per peer (async, small).
+  for (int q = world - 1; q >= 0; --q) {
+    if (!mine[q] || g.mr_send_off[q] * 8 == sb[q]) continue;
+    HIP_CHECK(hipMemcpyAsync(
+        g.mr_recvbuf.data() + sb[q],
+        g.mr_sendbuf.data() + g.mr_send_off[q] * 8, mine[q] * 8,
+        hipMemcpyDeviceToDevice, s));
+  }
+  // regions that moved were staged into recvbuf — move back contiguously
+  for (int q = world - 1; q >= 0; --q) {
+    if (!mine[q] || g.mr_send_off[q] * 8 == sb[q]) continue;
+    HIP_CHECK(hipMemcpyAsync(g.mr_sendbuf.data() + sb[q],
+                             g.mr_recvbuf.data() + sb[q], mine[q] * 8,
+                             hipMemcpyDeviceToDevice, s));
+  }
+  I.ev_pack.record(s);
+  I.ev_pack.wait_on(I.comm_stream);
+  I.dc.sendrecv(g.mr_sendbuf.data(), sb, g.mr_recvbuf.data(), rb,
+                I.comm_stream);
+  I.ev_comm.record(I.comm_stream);
+  I.ev_comm.wait_on(s);
+  if (recv_pairs)
+    mirror_scatter_pairs_kernel<<<grid_for(recv_pairs), kBlock, 0, s>>>(
+        reinterpret_cast<const unsigned long long*>(g.mr_recvbuf.data()),
+        recv_pairs, static_cast<uint32_t*>(state4));
+}
+
 // old-id -> renumbered-id for API source arguments
 static uint32_t map_source(DeviceGraph& g, int64_t source, hipStream_t s) {
   uint32_t v = static_cast<uint32_t>(source);
@@ -4140,9 +4240,17 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
     uint64_t g_nch = multi ? comm_->allreduce_sum(nch) : nch;
     if (multi) {
       // refresh referenced remote labels point-to-point (per-link volume
-      // scales with boundary size, not V·world)
-      mirror_sync_begin(I, g, lab.data(), s);
-      mirror_sync_end(I, g, lab.data(), s);
+      // scales with boundary size, not V·world); once churn decays,
+      // ship only the CHANGED labels as (id, value) pairs
+      uint64_t g_owned_now =
+          comm_->allreduce_sum(static_cast<uint64_t>(owned));
+      if (g_nch * 8 < g_owned_now) {
+        mirror_sync_changed(I, comm_, g, lab.data(), changed_bm.data(),
+                            rank_, world_, s);
+      } else {
+        mirror_sync_begin(I, g, lab.data(), s);
+        mirror_sync_end(I, g, lab.data(), s);
+      }
     }
     if (g_nch == 0) break;  // fixpoint: further rounds are no-ops
     if (it + 1 == iters) break;
