@@ -59,6 +59,17 @@ def build_parser():
     ap.add_argument("--serialize", action="store_true")
     ap.add_argument("--deserialize", action="store_true")
     ap.add_argument("--serialization_prefix", default="")
+    # reference flags.cc parity
+    ap.add_argument("--lb", default="", choices=("", "cm", "strict", "none"),
+                    help="GPU edge-expansion scheduler (reference --lb)")
+    ap.add_argument("--rebalance", action="store_true",
+                    help="degree-balanced ownership (reference rebalancer)")
+    ap.add_argument("--rebalance_vertex_factor", type=float, default=1.0)
+    ap.add_argument("--vertex_num", type=int, default=0,
+                    help="declared |V| (extends/validates the id space)")
+    ap.add_argument("--edge_num", type=int, default=0,
+                    help="declared |E| (accepted for reference parity)")
+    ap.add_argument("--jobid", default="", help="echoed in the timer header")
     return ap
 
 
@@ -70,6 +81,8 @@ def fmt_value(v):
 
 def main(argv=None):
     args = build_parser().parse_args(argv)
+    if args.lb:
+        os.environ["GRAPEHIP_LB"] = args.lb
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     eng = grapehip.engine_from_env(n_threads=args.app_concurrency,
@@ -87,15 +100,27 @@ def main(argv=None):
         src, dst, w = read_ldbc_edges(args.efile, weighted=weighted,
                                       rank=rank, world=world)
         kw = {}
-        if args.vfile:
+        if args.rebalance:
+            from grapehip.io import rebalance_partition
+            nv_hint = args.vertex_num or int(
+                max(src.max(initial=0), dst.max(initial=0)) + 1)
+            owned = rebalance_partition(eng, src, dst, nv_hint,
+                                        factor=args.rebalance_vertex_factor)
+            kw["vertex_oids"] = owned
+            kw["partitioner"] = "map"
+            kw["idxer"] = args.idxer
+        elif args.vfile:
             oids = read_ldbc_vertices(args.vfile, rank=rank, world=world)
             kw["vertex_oids"] = oids
             kw["idxer"] = args.idxer
         else:
             hi = max(src.max(initial=0), dst.max(initial=0)) + 1
-            kw["num_vertices"] = int(hi)
+            kw["num_vertices"] = int(max(hi, args.vertex_num))
         g = eng.load_edges(src, dst, weights=w, directed=args.directed,
                            build_in_csr=args.directed, **kw)
+        if args.vertex_num and g.num_vertices != args.vertex_num and rank == 0:
+            print("[warn] --vertex_num %d but loaded %d" %
+                  (args.vertex_num, g.num_vertices))
         if args.serialize:
             os.makedirs(args.serialization_prefix, exist_ok=True)
             eng.save_graph(g, args.serialization_prefix)
@@ -154,6 +179,8 @@ def main(argv=None):
     timers.append(("output", time.time() - t))
 
     if rank == 0:
+        if args.jobid:
+            print("[job] %s" % args.jobid)
         for name, secs in timers:
             print("[timer] %-14s %.6f s" % (name, secs))
         # reference GetMemoryUsage line printed with the timer table
