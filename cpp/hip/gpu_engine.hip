@@ -1274,6 +1274,98 @@ __device__ __forceinline__ PrSpan pr_span(const uint32_t* __restrict__ dst,
   return {b, L - b, R, e - R};
 }
 
+// Edge-parallel small-row pull (CM idiom): thread-per-row streaming left
+// each lane walking its own row — scattered 64B lines per 4B edge read.
+// A block stages 256 short rows' spans in LDS, prefix-sums their
+// lengths, then sweeps the chunk's edges with CONSECUTIVE lanes on
+// consecutive edge slots (coalesced dst/contrib streams), wave-segmented
+// sums per row and LDS fp64 accumulation at run tails.
+template <int PHASE>
+__global__ void pr_pull_small_cm_kernel(
+    const uint64_t* __restrict__ off, const uint32_t* __restrict__ dst,
+    const float* __restrict__ contrib, const uint32_t* __restrict__ rows,
+    uint64_t nrows, uint32_t v_begin, uint32_t lo, uint32_t hi,
+    double* __restrict__ acc) {
+  __shared__ uint64_t s_b1[kBlock], s_b2[kBlock];
+  __shared__ uint32_t s_n1[kBlock], s_pref[kBlock + 1];
+  __shared__ double s_acc[kBlock];
+  const int lane = threadIdx.x & 63;
+  for (uint64_t chunk = static_cast<uint64_t>(blockIdx.x) * kBlock;
+       chunk < nrows;
+       chunk += static_cast<uint64_t>(gridDim.x) * kBlock) {
+    uint64_t i = chunk + threadIdx.x;
+    uint32_t n = 0;
+    if (i < nrows) {
+      uint32_t r = rows[i];
+      PrSpan sp = pr_span<PHASE>(dst, off[r], off[r + 1], lo, hi);
+      s_b1[threadIdx.x] = sp.base1;
+      s_n1[threadIdx.x] = static_cast<uint32_t>(sp.n1);
+      s_b2[threadIdx.x] = sp.base2;
+      n = static_cast<uint32_t>(sp.n());
+    }
+    s_acc[threadIdx.x] = 0.0;
+    s_pref[threadIdx.x] = n;
+    __syncthreads();
+    // block-exclusive scan of the 256 lengths (Hillis-Steele in LDS)
+    for (int d = 1; d < kBlock; d <<= 1) {
+      uint32_t v = s_pref[threadIdx.x];
+      uint32_t up = threadIdx.x >= d ? s_pref[threadIdx.x - d] : 0;
+      __syncthreads();
+      s_pref[threadIdx.x] = v + up;
+      __syncthreads();
+    }
+    uint32_t total = s_pref[kBlock - 1];
+    // shift to exclusive
+    uint32_t excl = threadIdx.x ? s_pref[threadIdx.x - 1] : 0;
+    __syncthreads();
+    s_pref[threadIdx.x] = excl;
+    if (threadIdx.x == 0) s_pref[kBlock] = total;
+    __syncthreads();
+    for (uint32_t base = 0; base < total; base += kBlock) {
+      uint32_t e = base + threadIdx.x;
+      bool act = e < total;
+      uint32_t owner = 0xFFFFFFFFu;
+      double c = 0.0;
+      if (act) {
+        uint32_t a = 0, b = kBlock;  // last owner with pref <= e
+        while (a + 1 < b) {
+          uint32_t m = (a + b) >> 1;
+          if (s_pref[m] <= e) a = m;
+          else b = m;
+        }
+        owner = a;
+        uint32_t k = e - s_pref[a];
+        uint64_t idx = k < s_n1[a]
+                           ? s_b1[a] + k
+                           : s_b2[a] + (k - s_n1[a]);
+        c = static_cast<double>(contrib[dst[idx]]);
+      }
+      // wave-segmented sum keyed by owner (runs are contiguous)
+      uint32_t oprev = __shfl_up(owner, 1, 64);
+      bool head = (lane == 0) || oprev != owner;
+      unsigned long long hb = __ballot(head);
+      unsigned long long mine =
+          hb & ((lane == 63) ? ~0ull : ((1ull << (lane + 1)) - 1));
+      int seg = 63 - __clzll(mine);
+      double sum = c;
+#pragma unroll
+      for (int d = 1; d < 64; d <<= 1) {
+        double up = __shfl_up(sum, d, 64);
+        if (lane - d >= seg) sum += up;
+      }
+      bool tail = (lane == 63) || ((hb >> (lane + 1)) & 1ull);
+      if (tail && act) atomicAdd(&s_acc[owner], sum);
+    }
+    __syncthreads();
+    if (i < nrows) {
+      uint32_t r = rows[i];
+      if (PHASE == 2) acc[v_begin + r] += s_acc[threadIdx.x];
+      else acc[v_begin + r] = s_acc[threadIdx.x];
+    }
+    __syncthreads();
+  }
+}
+
 // thread per row
 template <int PHASE>
 __global__ void pr_pull_small_kernel(const uint64_t* __restrict__ off,
@@ -3187,7 +3279,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
     auto launch = [&](auto tag) {
       constexpr int P = decltype(tag)::value;
       if (g.n_small)
-        pr_pull_small_kernel<P><<<grid_for(g.n_small), kBlock, 0, s>>>(
+        pr_pull_small_cm_kernel<P><<<grid_for(g.n_small), kBlock, 0, s>>>(
             pull_off, pull_dst, contrib.data(), g.rows_small.data(),
             g.n_small, g.v_begin, g.v_begin, g.v_end, acc.data());
       if (g.n_mid)
