@@ -316,3 +316,35 @@ def test_memory_tracking(eng):
     assert mi["hip_alloc_current"] > 0
     assert mi["hip_alloc_peak"] >= mi["hip_alloc_current"]
     assert mi["VmHWM"] > 0
+
+
+def test_gpu_mutate_then_run(eng):
+    # in-place delta mutation must compact slack and re-upload: results on
+    # the mutated device graph equal a fresh build of the mutated edges
+    rng = np.random.default_rng(53)
+    nv = 3000
+    src = rng.integers(0, nv, 20000)
+    dst = rng.integers(0, nv, 20000)
+    k = src != dst
+    src, dst = src[k].astype(np.int64), dst[k].astype(np.int64)
+    g = eng.load_edges(src, dst, directed=False, num_vertices=nv)
+    ad_s = rng.integers(0, nv, 500).astype(np.int64)
+    ad_d = rng.integers(0, nv, 500).astype(np.int64)
+    ka = ad_s != ad_d
+    ad_s, ad_d = ad_s[ka], ad_d[ka]
+    empty = np.array([], dtype=np.int64)
+    g = eng.mutate_graph(g, add_src=ad_s, add_dst=ad_d,
+                         remove_src=src[:300].copy(),
+                         remove_dst=dst[:300].copy(),
+                         remove_vertices=empty)
+    pairs = set()
+    for a, b in zip(src[:300], dst[:300]):
+        pairs.add((a, b))
+        pairs.add((b, a))
+    keep = np.array([(a, b) not in pairs for a, b in zip(src, dst)])
+    e_s = np.concatenate([src[keep], ad_s])
+    e_d = np.concatenate([dst[keep], ad_d])
+    _, vals = by_oid(eng.bfs(g, 7))
+    assert np.array_equal(vals, bfs_oracle(nv, e_s, e_d, 7, directed=False))
+    _, w = by_oid(eng.wcc(g))
+    assert np.array_equal(w, wcc_oracle(nv, e_s, e_d))
