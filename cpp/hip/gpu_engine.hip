@@ -1027,6 +1027,59 @@ __global__ void pr_apply_kernel(double* __restrict__ rank,
   }
 }
 
+// Fused apply + NEXT iteration's contrib/dangling: one pass over rank
+// and the out-degree offsets instead of three (apply, then contrib, then
+// dangling each re-reading rank). Single-GPU pull path; rows beyond
+// owned_real (hashmap padding) get the rank update only.
+__global__ void pr_fused_apply_kernel(
+    double* __restrict__ rank, const double* __restrict__ acc,
+    const double* __restrict__ dangling_cur, double inv_n, double damping,
+    uint32_t v_begin, uint32_t owned, uint32_t owned_real,
+    const uint64_t* __restrict__ off, float* __restrict__ contrib,
+    double* __restrict__ dangling_next, double* __restrict__ l1_delta) {
+  __shared__ double s_wave[kBlock / kWave];
+  __shared__ double s_dang[kBlock / kWave];
+  const double base =
+      (1.0 - damping) * inv_n + damping * (*dangling_cur) * inv_n;
+  double d1 = 0, dang = 0;
+  size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (uint32_t r = blockIdx.x * blockDim.x + threadIdx.x; r < owned;
+       r += stride) {
+    double nv = base + damping * acc[v_begin + r];
+    if (l1_delta) d1 += fabs(nv - rank[v_begin + r]);
+    rank[v_begin + r] = nv;
+    if (r < owned_real) {
+      uint64_t deg = off[r + 1] - off[r];
+      if (deg) {
+        contrib[r] = static_cast<float>(nv / static_cast<double>(deg));
+      } else {
+        contrib[r] = 0.0f;
+        dang += nv;
+      }
+    }
+  }
+#pragma unroll
+  for (int d = 32; d > 0; d >>= 1) {
+    dang += __shfl_down(dang, d, 64);
+    if (l1_delta) d1 += __shfl_down(d1, d, 64);
+  }
+  if ((threadIdx.x & 63) == 0) {
+    s_dang[threadIdx.x >> 6] = dang;
+    s_wave[threadIdx.x >> 6] = d1;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double td = 0, t1 = 0;
+#pragma unroll
+    for (int w = 0; w < kBlock / kWave; ++w) {
+      td += s_dang[w];
+      t1 += s_wave[w];
+    }
+    if (td) unsafeAtomicAdd(dangling_next, td);
+    if (l1_delta && t1) unsafeAtomicAdd(l1_delta, t1);
+  }
+}
+
 __global__ void pr_dangling_kernel(const double* __restrict__ rank,
                                    const uint64_t* __restrict__ off,
                                    uint32_t owned, uint32_t v_begin,
@@ -3262,6 +3315,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
   DeviceBuffer<double>& acc = g.pr_acc;
   DeviceBuffer<float>& contrib = g.pr_contrib;
   DeviceBuffer<double>& d_dangling = g.pr_dangling;
+  if (d_dangling.size() < 2) d_dangling.resize(2);  // [0]=cur, [1]=next
   DeviceBuffer<double> d_l1(tol > 0 ? 1 : 0);
   DevGraphView view = make_view(g, rank_, world_);
 
@@ -3308,6 +3362,45 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
     I.h_scal[1] = dang;
     HIP_CHECK(hipMemcpyAsync(d_dangling.data(), I.h_scal + 1, 8,
                              hipMemcpyHostToDevice, s));
+  };
+
+  // Single-GPU pull path: the fused iteration folds apply + the NEXT
+  // iteration's contrib/dangling into one rank pass (bootstrap computes
+  // the first dangling/contrib outside the loop); the dangling scalar
+  // ping-pongs through a fixed [cur,next] pair so the sequence is
+  // hipGraph-capturable with stable pointers.
+  const bool fused = pull && !multi;
+  auto fused_bootstrap = [&]() {
+    d_dangling.zero(s);
+    pr_dangling_kernel<<<grid_for(g.owned_real), kBlock, 0, s>>>(
+        rank_arr.data(), g.oe_off.data(), g.owned_real, g.v_begin,
+        d_dangling.data());
+    pr_contrib_kernel<<<grid_for(g.owned_real), kBlock, 0, s>>>(
+        rank_arr.data(), g.oe_off.data(), g.owned_real, g.v_begin,
+        contrib.data() + g.v_begin);
+  };
+  auto fused_iteration = [&]() {
+    if (tiled) {
+      acc.zero(s);
+      static int tile_grid_f = [] {
+        const char* e = getenv("GRAPEHIP_PR_GRID");
+        return e ? atoi(e) : 8192;
+      }();
+      if (g.pr_total)
+        pr_tiled_pull_kernel<<<tile_grid_f, kBlock, 0, s>>>(
+            g.pr_tiles.data(), 0, g.pr_total, contrib.data(), acc.data());
+    } else {
+      pulls(0);
+    }
+    HIP_CHECK(hipMemsetAsync(d_dangling.data() + 1, 0, 8, s));
+    if (tol > 0) d_l1.zero(s);
+    pr_fused_apply_kernel<<<grid_for(owned), kBlock, 0, s>>>(
+        rank_arr.data(), acc.data(), d_dangling.data(), inv_n, damping,
+        g.v_begin, owned, g.owned_real, g.oe_off.data(),
+        contrib.data() + g.v_begin, d_dangling.data() + 1,
+        tol > 0 ? d_l1.data() : nullptr);
+    HIP_CHECK(hipMemcpyAsync(d_dangling.data(), d_dangling.data() + 1, 8,
+                             hipMemcpyDeviceToDevice, s));
   };
 
   // one iteration, recorded as a stream of kernels. Single-GPU fixed-iter
@@ -3393,6 +3486,7 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
   const bool capture = !multi && tol <= 0 && g.pr_calls++ > 0;
   hipGraphExec_t graph_exec =
       reinterpret_cast<hipGraphExec_t>(g.pr_graph_exec);
+  if (fused) fused_bootstrap();
   if (capture && (!graph_exec || g.pr_graph_damping != damping)) {
     if (graph_exec) {
       HIP_CHECK(hipGraphExecDestroy(graph_exec));
@@ -3400,7 +3494,8 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
     }
     hipGraph_t graph = nullptr;
     HIP_CHECK(hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal));
-    record_iteration();
+    if (fused) fused_iteration();
+    else record_iteration();
     HIP_CHECK(hipStreamEndCapture(s, &graph));
     HIP_CHECK(hipGraphInstantiate(&graph_exec, graph, nullptr, nullptr, 0));
     HIP_CHECK(hipGraphDestroy(graph));
@@ -3411,6 +3506,8 @@ GpuRunResult GpuContext::pagerank(DeviceGraph& g, double damping,
   for (int it = 0; it < iters; ++it) {
     if (capture)
       HIP_CHECK(hipGraphLaunch(graph_exec, s));
+    else if (fused)
+      fused_iteration();
     else
       record_iteration();
     ++rounds;
@@ -4330,6 +4427,7 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
                              s));
     HIP_CHECK(hipStreamSynchronize(s));
     uint64_t g_nch = multi ? comm_->allreduce_sum(nch) : nch;
+    if (g_nch == 0) break;  // fixpoint: nothing changed, nothing to sync
     if (multi) {
       // refresh referenced remote labels point-to-point (per-link volume
       // scales with boundary size, not V·world); once churn decays,
@@ -4344,7 +4442,6 @@ GpuRunResult GpuContext::cdlp(DeviceGraph& g, int iters, bool fetch) {
         mirror_sync_end(I, g, lab.data(), s);
       }
     }
-    if (g_nch == 0) break;  // fixpoint: further rounds are no-ops
     if (it + 1 == iters) break;
     // adaptive: building the dirty set costs an edge expansion over the
     // changed rows — only worth it once changes are sparse (converging
