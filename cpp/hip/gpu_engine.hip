@@ -3004,7 +3004,11 @@ GpuRunResult GpuContext::bfs(DeviceGraph& g, int64_t source,
   int rounds = 0;
   // Beamer-style switch: pull when the frontier's out-edges exceed a
   // fraction of the stored edges (cuda/bfs/bfs.h heuristic, retuned).
-  const uint64_t pull_edge_threshold = g.total_edges / 16;
+  static const int pull_div = [] {
+    const char* e = getenv("GRAPEHIP_BFS_PULL_DIV");
+    return e ? atoi(e) : 16;
+  }();
+  const uint64_t pull_edge_threshold = g.total_edges / pull_div;
   while (global_curr > 0) {
     bool use_pull = false;
     uint64_t fedges = 0;
