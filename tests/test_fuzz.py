@@ -84,3 +84,55 @@ def test_fuzz_cdlp_lcc_coreness(eng, g):
     rc = eng.core_decomposition(gr)
     assert np.array_equal(rc["values"][np.argsort(rc["oids"])],
                           coreness_oracle(nv, src, dst))
+
+
+def test_fuzz_mutation_sequences():
+    # random in-place delta sequences (duplicate edges, repeated removals,
+    # fresh outer vertices) must track a host-maintained edge multiset
+    import grapehip
+    from oracles import bfs_oracle, wcc_oracle
+    eng = grapehip.Engine(rank=0, world=1, master_port=29667)
+    rng = np.random.default_rng(271)
+    for trial in range(4):
+        nv = int(rng.integers(50, 400))
+        ne = int(rng.integers(nv, nv * 6))
+        src = rng.integers(0, nv, ne)
+        dst = rng.integers(0, nv, ne)
+        k = src != dst
+        src, dst = src[k].astype(np.int64), dst[k].astype(np.int64)
+        g = eng.load_edges(src, dst, directed=False, num_vertices=nv)
+        cur = list(zip(src.tolist(), dst.tolist()))
+        empty = np.array([], dtype=np.int64)
+        for step in range(3):
+            na = int(rng.integers(0, 60))
+            a_s = rng.integers(0, nv, na)
+            a_d = rng.integers(0, nv, na)
+            ka = a_s != a_d
+            a_s, a_d = a_s[ka].astype(np.int64), a_d[ka].astype(np.int64)
+            nr = int(rng.integers(0, min(20, len(cur)) + 1))
+            if nr and cur:
+                pick = rng.integers(0, len(cur), nr)
+                r_s = np.array([cur[i][0] for i in pick], dtype=np.int64)
+                r_d = np.array([cur[i][1] for i in pick], dtype=np.int64)
+            else:
+                r_s = r_d = empty
+            g = eng.mutate_graph(g, add_src=a_s, add_dst=a_d,
+                                 remove_src=r_s, remove_dst=r_d,
+                                 remove_vertices=empty)
+            rmset = set()
+            for a, b in zip(r_s, r_d):
+                rmset.add((a, b))
+                rmset.add((b, a))
+            cur = [e for e in cur if e not in rmset]
+            cur += list(zip(a_s.tolist(), a_d.tolist()))
+            e_s = np.array([e[0] for e in cur], dtype=np.int64)
+            e_d = np.array([e[1] for e in cur], dtype=np.int64)
+            r = eng.bfs(g, 1)
+            order = np.argsort(r["oids"])
+            assert np.array_equal(
+                r["values"][order],
+                bfs_oracle(nv, e_s, e_d, 1, directed=False)), (trial, step)
+            r = eng.wcc(g)
+            assert np.array_equal(
+                r["values"][np.argsort(r["oids"])],
+                wcc_oracle(nv, e_s, e_d)), (trial, step)
