@@ -726,6 +726,54 @@ __global__ void expand_strict_frontier(DevGraphView g,
   }
 }
 
+// WM: wave-granular CM (reference LBWARP, parallel_engine.h:775-845,
+// re-derived for 64-wide waves). Each wave claims 64 frontier entries and
+// stages their adjacency bases + degree prefix in REGISTERS (no LDS):
+// the prefix comes from a wave inclusive scan, owner search probes
+// lane-held prefix values with __shfl (6-step binary search per edge).
+template <typename EdgeOp>
+__global__ void expand_wm_frontier(DevGraphView g,
+                                   const uint32_t* __restrict__ q,
+                                   uint32_t qn, EdgeOp op) {
+  const int lane = threadIdx.x & 63;
+  const uint64_t wid =
+      (static_cast<uint64_t>(blockIdx.x) * blockDim.x + threadIdx.x) /
+      kWave;
+  const uint64_t waves =
+      (static_cast<uint64_t>(gridDim.x) * blockDim.x) / kWave;
+  for (uint64_t c = wid * kWave; c < qn; c += waves * kWave) {
+    uint32_t i = static_cast<uint32_t>(c) + lane;
+    bool act = i < qn;
+    uint32_t u = act ? q[i] : 0;
+    uint64_t b = 0;
+    uint32_t deg = 0;
+    if (act) {
+      uint32_t r = g.row(u);
+      b = g.oe_off[r];
+      deg = static_cast<uint32_t>(g.oe_off[r + 1] - b);
+    }
+    uint64_t incl = wave_incl_scan(deg);
+    uint64_t total = __shfl(static_cast<unsigned long long>(incl), 63, 64);
+    uint64_t excl = incl - deg;
+    for (uint64_t e = lane; e < total; e += kWave) {
+      int lo = 0, hi = 63;
+#pragma unroll
+      for (int step = 0; step < 6; ++step) {
+        int mid = (lo + hi + 1) >> 1;
+        uint64_t px =
+            __shfl(static_cast<unsigned long long>(excl), mid, 64);
+        if (px <= e) lo = mid;
+        else hi = mid - 1;
+      }
+      uint32_t uo = __shfl(u, lo, 64);
+      uint64_t bo = __shfl(static_cast<unsigned long long>(b), lo, 64);
+      uint64_t po = __shfl(static_cast<unsigned long long>(excl), lo, 64);
+      uint64_t eid = bo + (e - po);
+      op(uo, g.oe_dst[eid], g.oe_w ? g.oe_w[eid] : 1.0f);
+    }
+  }
+}
+
 // Thread-per-item baseline (LB=none), frontier form.
 template <typename EdgeOp>
 __global__ void expand_none_frontier(DevGraphView g,
@@ -2934,7 +2982,8 @@ void expand_frontier(GpuContext::Impl& I, const DevGraphView& view,
   uint64_t total = exclusive_scan(I.frontier_deg.data(),
                                   I.frontier_off.data(), qn, s, I.scan);
   // LB strategy (reference --lb flag, default cta/cm): cm = LDS-staged
-  // owner search; strict = perfectly edge-balanced; none = thread-per-row
+  // owner search; wm = wave-granular register-staged CM; strict =
+  // perfectly edge-balanced; none = thread-per-row
   static const char* lb = getenv("GRAPEHIP_LB");
   if (lb && lb[0] == 's') {
     expand_strict_frontier<Op><<<grid_for(total), kBlock, 0, s>>>(
@@ -2942,6 +2991,9 @@ void expand_frontier(GpuContext::Impl& I, const DevGraphView& view,
   } else if (lb && lb[0] == 'n') {
     expand_none_frontier<Op><<<grid_for(qn), kBlock, 0, s>>>(view, q, qn,
                                                              op);
+  } else if (lb && lb[0] == 'w') {
+    expand_wm_frontier<Op><<<grid_for(qn), kBlock, 0, s>>>(view, q, qn,
+                                                           op);
   } else {
     int nchunks = static_cast<int>((qn + kBlock - 1) / kBlock);
     expand_cm_frontier<Op><<<std::min(nchunks, kMaxGrid), kBlock, 0, s>>>(

@@ -60,7 +60,7 @@ def build_parser():
     ap.add_argument("--deserialize", action="store_true")
     ap.add_argument("--serialization_prefix", default="")
     # reference flags.cc parity
-    ap.add_argument("--lb", default="", choices=("", "cm", "strict", "none"),
+    ap.add_argument("--lb", default="", choices=("", "cm", "wm", "strict", "none"),
                     help="GPU edge-expansion scheduler (reference --lb)")
     ap.add_argument("--rebalance", action="store_true",
                     help="degree-balanced ownership (reference rebalancer)")

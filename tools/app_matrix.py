@@ -140,7 +140,7 @@ def main():
     apps = args.apps.split(",")
     worlds = [int(w) for w in args.worlds.split(",")]
     lbs = args.lbs.split(",") if args.lbs else (
-        ["cm", "strict", "none"] if args.gpu else [""])
+        ["cm", "wm", "strict", "none"] if args.gpu else [""])
 
     tmp = Path(tempfile.mkdtemp(prefix="grapehip_matrix_"))
     vfile, efile = gen_dataset(tmp, args.num_v, args.num_e)
