@@ -278,7 +278,7 @@ def test_force_terminate_mp(world, free_port, tmp_path):
     assert (vals == 1).all()  # 1 = raised with the right info on that rank
 
 
-@pytest.mark.parametrize("world", [2, 3])
+@pytest.mark.parametrize("world", [2, 3, 8])
 def test_large_messages_eager_flush(world, free_port, tmp_path):
     # boundary messages exceeding the flush block force the mid-round
     # flush (background sender thread); results must still merge exactly

@@ -156,3 +156,41 @@ def test_serialize_after_mutate(eng, tmp_path):
     order = np.argsort(r["oids"])
     assert np.array_equal(r["values"][order],
                           bfs_oracle(nv, e_src, e_dst, 5, directed=False))
+
+
+def test_mutate_after_deserialize(eng, tmp_path):
+    # in-place delta applied to a checkpoint-restored fragment (FromParts
+    # state must enter mutable mode cleanly), then re-checkpointed
+    src, dst = base_graph(seed=113)
+    nv = 600
+    g = eng.load_edges(src, dst, directed=False, num_vertices=nv)
+    ck1 = tmp_path / "ck1"
+    ck1.mkdir()
+    eng.save_graph(g, str(ck1))
+    g2 = eng.load_serialized(str(ck1))
+    rng = np.random.default_rng(127)
+    ad_s = rng.integers(0, nv, 120).astype(np.int64)
+    ad_d = rng.integers(0, nv, 120).astype(np.int64)
+    k = ad_s != ad_d
+    ad_s, ad_d = ad_s[k], ad_d[k]
+    empty = np.array([], dtype=np.int64)
+    g2 = eng.mutate_graph(g2, add_src=ad_s, add_dst=ad_d,
+                          remove_src=src[:50].copy(),
+                          remove_dst=dst[:50].copy(),
+                          remove_vertices=empty)
+    ck2 = tmp_path / "ck2"
+    ck2.mkdir()
+    eng.save_graph(g2, str(ck2))
+    g3 = eng.load_serialized(str(ck2))
+    pairs = set()
+    for a, b in zip(src[:50], dst[:50]):
+        pairs.add((a, b))
+        pairs.add((b, a))
+    keep = np.array([(a, b) not in pairs for a, b in zip(src, dst)])
+    e_s = np.concatenate([src[keep], ad_s])
+    e_d = np.concatenate([dst[keep], ad_d])
+    for gx in (g2, g3):
+        r = eng.bfs(gx, 5)
+        order = np.argsort(r["oids"])
+        assert np.array_equal(
+            r["values"][order], bfs_oracle(nv, e_s, e_d, 5, directed=False))
